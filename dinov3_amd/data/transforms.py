@@ -1,0 +1,180 @@
+"""Torch-native image transform primitives (no torchvision/PIL dependency).
+
+Images are float32 CHW tensors in [0,1] (datasets decode uint8 HWC -> this).
+Covers what DataAugmentationDINO needs (reference dinov3_jax/data/
+augmentations.py + transforms.py): random resized crop, hflip, color jitter,
+grayscale, gaussian blur, solarize, normalize, plus eval presets.
+"""
+
+from __future__ import annotations
+
+import math
+import random
+from typing import List, Optional, Sequence, Tuple
+
+import torch
+import torch.nn.functional as F
+
+
+def resize(img: torch.Tensor, size: int, mode: str = "bicubic") -> torch.Tensor:
+    return F.interpolate(img.unsqueeze(0), size=(size, size), mode=mode, align_corners=False,
+                         antialias=True).squeeze(0).clamp(0.0, 1.0)
+
+
+def center_crop(img: torch.Tensor, size: int) -> torch.Tensor:
+    _, h, w = img.shape
+    top = max((h - size) // 2, 0)
+    left = max((w - size) // 2, 0)
+    return img[:, top: top + size, left: left + size]
+
+
+def sample_rrc_box(h: int, w: int, scale: Tuple[float, float],
+                   ratio: Tuple[float, float] = (3.0 / 4.0, 4.0 / 3.0)) -> Tuple[int, int, int, int]:
+    area = h * w
+    for _ in range(10):
+        target_area = area * random.uniform(*scale)
+        log_ratio = (math.log(ratio[0]), math.log(ratio[1]))
+        aspect = math.exp(random.uniform(*log_ratio))
+        cw = int(round(math.sqrt(target_area * aspect)))
+        ch = int(round(math.sqrt(target_area / aspect)))
+        if 0 < cw <= w and 0 < ch <= h:
+            top = random.randint(0, h - ch)
+            left = random.randint(0, w - cw)
+            return top, left, ch, cw
+    # fallback: central crop at clamped aspect
+    in_ratio = w / h
+    if in_ratio < ratio[0]:
+        cw, ch = w, int(round(w / ratio[0]))
+    elif in_ratio > ratio[1]:
+        ch, cw = h, int(round(h * ratio[1]))
+    else:
+        cw, ch = w, h
+    return (h - ch) // 2, (w - cw) // 2, ch, cw
+
+
+def random_resized_crop(img: torch.Tensor, size: int, scale: Tuple[float, float],
+                        box: Optional[Tuple[int, int, int, int]] = None) -> torch.Tensor:
+    _, h, w = img.shape
+    if box is None:
+        box = sample_rrc_box(h, w, scale)
+    top, left, ch, cw = box
+    crop = img[:, top: top + ch, left: left + cw]
+    return resize(crop, size)
+
+
+def hflip(img: torch.Tensor) -> torch.Tensor:
+    return img.flip(-1)
+
+
+def _blend(a: torch.Tensor, b: torch.Tensor, alpha: float) -> torch.Tensor:
+    return (alpha * a + (1.0 - alpha) * b).clamp(0.0, 1.0)
+
+
+def rgb_to_grayscale(img: torch.Tensor) -> torch.Tensor:
+    g = (0.2989 * img[0] + 0.587 * img[1] + 0.114 * img[2]).unsqueeze(0)
+    return g.expand_as(img).contiguous()
+
+
+def adjust_brightness(img: torch.Tensor, factor: float) -> torch.Tensor:
+    return _blend(img, torch.zeros_like(img), factor)
+
+
+def adjust_contrast(img: torch.Tensor, factor: float) -> torch.Tensor:
+    mean = rgb_to_grayscale(img).mean()
+    return _blend(img, mean.expand_as(img), factor)
+
+
+def adjust_saturation(img: torch.Tensor, factor: float) -> torch.Tensor:
+    return _blend(img, rgb_to_grayscale(img), factor)
+
+
+def adjust_hue(img: torch.Tensor, hue_shift: float) -> torch.Tensor:
+    """hue_shift in [-0.5, 0.5] turns of the hue wheel."""
+    r, g, b = img[0], img[1], img[2]
+    maxc, _ = img.max(dim=0)
+    minc, _ = img.min(dim=0)
+    v = maxc
+    deltac = maxc - minc
+    s = torch.where(maxc > 0, deltac / maxc.clamp_min(1e-8), torch.zeros_like(maxc))
+    dz = deltac.clamp_min(1e-8)
+    rc = (maxc - r) / dz
+    gc = (maxc - g) / dz
+    bc = (maxc - b) / dz
+    h = torch.where(r == maxc, bc - gc, torch.where(g == maxc, 2.0 + rc - bc, 4.0 + gc - rc))
+    h = (h / 6.0) % 1.0
+    h = torch.where(deltac > 0, h, torch.zeros_like(h))
+    h = (h + hue_shift) % 1.0
+    # hsv -> rgb
+    i = (h * 6.0).floor()
+    f = h * 6.0 - i
+    p = v * (1.0 - s)
+    q = v * (1.0 - f * s)
+    t = v * (1.0 - (1.0 - f) * s)
+    i = i.long() % 6
+    out = torch.empty_like(img)
+    masks = [i == k for k in range(6)]
+    rgb_cases = [(v, t, p), (q, v, p), (p, v, t), (p, q, v), (t, p, v), (v, p, q)]
+    for ch in range(3):
+        acc = torch.zeros_like(v)
+        for k in range(6):
+            acc = torch.where(masks[k], rgb_cases[k][ch], acc)
+        out[ch] = acc
+    return out.clamp(0.0, 1.0)
+
+
+def color_jitter(img: torch.Tensor, brightness: float, contrast: float,
+                 saturation: float, hue: float) -> torch.Tensor:
+    ops = []
+    if brightness > 0:
+        ops.append(("b", random.uniform(max(0.0, 1 - brightness), 1 + brightness)))
+    if contrast > 0:
+        ops.append(("c", random.uniform(max(0.0, 1 - contrast), 1 + contrast)))
+    if saturation > 0:
+        ops.append(("s", random.uniform(max(0.0, 1 - saturation), 1 + saturation)))
+    if hue > 0:
+        ops.append(("h", random.uniform(-hue, hue)))
+    random.shuffle(ops)
+    for kind, val in ops:
+        if kind == "b":
+            img = adjust_brightness(img, val)
+        elif kind == "c":
+            img = adjust_contrast(img, val)
+        elif kind == "s":
+            img = adjust_saturation(img, val)
+        else:
+            img = adjust_hue(img, val)
+    return img
+
+
+def gaussian_blur(img: torch.Tensor, sigma: float) -> torch.Tensor:
+    ksize = max(int(2 * round(3.0 * sigma) + 1), 3)
+    x = torch.arange(ksize, dtype=torch.float32) - ksize // 2
+    kernel = torch.exp(-0.5 * (x / sigma) ** 2)
+    kernel = kernel / kernel.sum()
+    c = img.shape[0]
+    kx = kernel.view(1, 1, 1, ksize).expand(c, 1, 1, ksize)
+    ky = kernel.view(1, 1, ksize, 1).expand(c, 1, ksize, 1)
+    pad = ksize // 2
+    out = F.conv2d(img.unsqueeze(0), kx, padding=(0, pad), groups=c)
+    out = F.conv2d(out, ky, padding=(pad, 0), groups=c)
+    return out.squeeze(0)
+
+
+def solarize(img: torch.Tensor, threshold: float = 0.5) -> torch.Tensor:
+    return torch.where(img >= threshold, 1.0 - img, img)
+
+
+def normalize(img: torch.Tensor, mean: Sequence[float], std: Sequence[float]) -> torch.Tensor:
+    mean_t = torch.tensor(mean, dtype=img.dtype).view(-1, 1, 1)
+    std_t = torch.tensor(std, dtype=img.dtype).view(-1, 1, 1)
+    return (img - mean_t) / std_t
+
+
+def make_eval_transform(resize_size: int = 256, crop_size: int = 224,
+                        mean=(0.485, 0.456, 0.406), std=(0.229, 0.224, 0.225)):
+    def _t(img: torch.Tensor) -> torch.Tensor:
+        img = resize(img, resize_size)
+        img = center_crop(img, crop_size)
+        return normalize(img, mean, std)
+
+    return _t

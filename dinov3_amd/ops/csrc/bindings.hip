@@ -1,0 +1,396 @@
+// PyTorch bindings for the dinov3_amd CDNA4 kernel library.
+// Compiled by hipcc (PYTORCH_ROCM_ARCH=gfx950) into dinov3_amd/ops/_hip_ops.so.
+
+#include <torch/extension.h>
+#include <ATen/hip/HIPContext.h>
+#include <hip/hip_runtime.h>
+#include <hip/hip_bf16.h>
+
+#include <vector>
+
+#define CHECK_INPUT(x) \
+  TORCH_CHECK(x.is_cuda() && x.is_contiguous(), #x " must be a contiguous HIP tensor")
+
+// ---- extern launchers (defined in the .hip kernel TUs) ----
+template <typename T>
+void launch_layernorm_fwd(const T*, const T*, const T*, T*, float*, float*, long, int,
+                          float, hipStream_t);
+template <typename T>
+void launch_layernorm_bwd(const T*, const T*, const T*, const float*, const float*, T*,
+                          float*, float*, long, int, hipStream_t);
+template <typename T>
+void launch_rmsnorm_fwd(const T*, const T*, T*, float*, long, int, float, hipStream_t);
+template <typename T>
+void launch_rmsnorm_bwd(const T*, const T*, const T*, const float*, T*, float*, long, int,
+                        hipStream_t);
+template <typename T>
+void launch_l2norm_fwd(const T*, T*, float*, long, int, float, hipStream_t);
+template <typename T>
+void launch_l2norm_bwd(const T*, const T*, const float*, T*, long, int, float, hipStream_t);
+template <typename T>
+void launch_bias_gelu_fwd(const T*, const T*, T*, long, int, hipStream_t);
+template <typename T>
+void launch_bias_gelu_bwd(const T*, const T*, const T*, T*, float*, long, int, hipStream_t);
+template <typename T>
+void launch_swiglu_fwd(const T*, T*, long, int, hipStream_t);
+template <typename T>
+void launch_swiglu_bwd(const T*, const T*, T*, long, int, hipStream_t);
+template <typename T>
+void launch_rope_fwd(const T*, const float*, const float*, T*, long, int, int, int,
+                     hipStream_t);
+template <typename T>
+void launch_multi_tensor_ema(T* const*, const T* const*, const long*, const int*,
+                             const long*, int, float, hipStream_t);
+template <typename T>
+void launch_multi_tensor_adamw(T* const*, const T* const*, float* const*, float* const*,
+                               float* const*, const long*, const int*, const long*, int,
+                               float, float, float, float, float, float, float, float,
+                               bool, hipStream_t);
+template <typename T>
+void launch_multi_tensor_l2norm_sq(const T* const*, const long*, const int*, const long*,
+                                   int, float*, hipStream_t);
+
+namespace {
+
+hipStream_t current_stream() { return at::hip::getCurrentHIPStream().stream(); }
+
+#define DISPATCH_FLOAT_BF16(TYPE, NAME, ...)                                   \
+  [&] {                                                                        \
+    if (TYPE == at::ScalarType::Float) {                                       \
+      using scalar_t = float;                                                  \
+      return __VA_ARGS__();                                                    \
+    } else if (TYPE == at::ScalarType::BFloat16) {                             \
+      using scalar_t = __hip_bfloat16;                                         \
+      return __VA_ARGS__();                                                    \
+    } else {                                                                   \
+      TORCH_CHECK(false, NAME ": unsupported dtype ", TYPE);                   \
+    }                                                                          \
+  }()
+
+// ------------------------------- norms ---------------------------------
+
+std::vector<torch::Tensor> layernorm_fwd(torch::Tensor x, torch::Tensor w, torch::Tensor b,
+                                         double eps) {
+  CHECK_INPUT(x);
+  const int D = x.size(-1);
+  const long rows = x.numel() / D;
+  auto y = torch::empty_like(x);
+  auto mean = torch::empty({rows}, x.options().dtype(torch::kFloat));
+  auto rstd = torch::empty({rows}, x.options().dtype(torch::kFloat));
+  DISPATCH_FLOAT_BF16(x.scalar_type(), "layernorm_fwd", [&] {
+    launch_layernorm_fwd<scalar_t>(
+        (const scalar_t*)x.data_ptr(), (const scalar_t*)w.data_ptr(),
+        (const scalar_t*)b.data_ptr(), (scalar_t*)y.data_ptr(),
+        mean.data_ptr<float>(), rstd.data_ptr<float>(), rows, D, (float)eps,
+        current_stream());
+  });
+  return {y, mean, rstd};
+}
+
+std::vector<torch::Tensor> layernorm_bwd(torch::Tensor dy, torch::Tensor x, torch::Tensor w,
+                                         torch::Tensor mean, torch::Tensor rstd) {
+  CHECK_INPUT(dy);
+  CHECK_INPUT(x);
+  const int D = x.size(-1);
+  const long rows = x.numel() / D;
+  auto dx = torch::empty_like(x);
+  auto dw = torch::zeros({D}, x.options().dtype(torch::kFloat));
+  auto db = torch::zeros({D}, x.options().dtype(torch::kFloat));
+  DISPATCH_FLOAT_BF16(x.scalar_type(), "layernorm_bwd", [&] {
+    launch_layernorm_bwd<scalar_t>(
+        (const scalar_t*)dy.data_ptr(), (const scalar_t*)x.data_ptr(),
+        (const scalar_t*)w.data_ptr(), mean.data_ptr<float>(), rstd.data_ptr<float>(),
+        (scalar_t*)dx.data_ptr(), dw.data_ptr<float>(), db.data_ptr<float>(), rows, D,
+        current_stream());
+  });
+  return {dx, dw.to(x.scalar_type()), db.to(x.scalar_type())};
+}
+
+std::vector<torch::Tensor> rmsnorm_fwd(torch::Tensor x, torch::Tensor w, double eps) {
+  CHECK_INPUT(x);
+  const int D = x.size(-1);
+  const long rows = x.numel() / D;
+  auto y = torch::empty_like(x);
+  auto rstd = torch::empty({rows}, x.options().dtype(torch::kFloat));
+  DISPATCH_FLOAT_BF16(x.scalar_type(), "rmsnorm_fwd", [&] {
+    launch_rmsnorm_fwd<scalar_t>((const scalar_t*)x.data_ptr(), (const scalar_t*)w.data_ptr(),
+                                 (scalar_t*)y.data_ptr(), rstd.data_ptr<float>(), rows, D,
+                                 (float)eps, current_stream());
+  });
+  return {y, rstd};
+}
+
+std::vector<torch::Tensor> rmsnorm_bwd(torch::Tensor dy, torch::Tensor x, torch::Tensor w,
+                                       torch::Tensor rstd) {
+  CHECK_INPUT(dy);
+  const int D = x.size(-1);
+  const long rows = x.numel() / D;
+  auto dx = torch::empty_like(x);
+  auto dw = torch::zeros({D}, x.options().dtype(torch::kFloat));
+  DISPATCH_FLOAT_BF16(x.scalar_type(), "rmsnorm_bwd", [&] {
+    launch_rmsnorm_bwd<scalar_t>((const scalar_t*)dy.data_ptr(), (const scalar_t*)x.data_ptr(),
+                                 (const scalar_t*)w.data_ptr(), rstd.data_ptr<float>(),
+                                 (scalar_t*)dx.data_ptr(), dw.data_ptr<float>(), rows, D,
+                                 current_stream());
+  });
+  return {dx, dw.to(x.scalar_type())};
+}
+
+std::vector<torch::Tensor> l2norm_fwd(torch::Tensor x, double eps) {
+  CHECK_INPUT(x);
+  const int D = x.size(-1);
+  const long rows = x.numel() / D;
+  auto y = torch::empty_like(x);
+  auto s = torch::empty({rows}, x.options().dtype(torch::kFloat));
+  DISPATCH_FLOAT_BF16(x.scalar_type(), "l2norm_fwd", [&] {
+    launch_l2norm_fwd<scalar_t>((const scalar_t*)x.data_ptr(), (scalar_t*)y.data_ptr(),
+                                s.data_ptr<float>(), rows, D, (float)eps, current_stream());
+  });
+  return {y, s};
+}
+
+torch::Tensor l2norm_bwd(torch::Tensor dy, torch::Tensor y, torch::Tensor s, double eps) {
+  CHECK_INPUT(dy);
+  const int D = y.size(-1);
+  const long rows = y.numel() / D;
+  auto dx = torch::empty_like(y);
+  DISPATCH_FLOAT_BF16(y.scalar_type(), "l2norm_bwd", [&] {
+    launch_l2norm_bwd<scalar_t>((const scalar_t*)dy.data_ptr(), (const scalar_t*)y.data_ptr(),
+                                s.data_ptr<float>(), (scalar_t*)dx.data_ptr(), rows, D,
+                                (float)eps, current_stream());
+  });
+  return dx;
+}
+
+// ----------------------------- elementwise ------------------------------
+
+torch::Tensor bias_gelu_fwd(torch::Tensor x, torch::Tensor bias) {
+  CHECK_INPUT(x);
+  const int H = x.size(-1);
+  const long rows = x.numel() / H;
+  auto y = torch::empty_like(x);
+  DISPATCH_FLOAT_BF16(x.scalar_type(), "bias_gelu_fwd", [&] {
+    launch_bias_gelu_fwd<scalar_t>((const scalar_t*)x.data_ptr(),
+                                   (const scalar_t*)bias.data_ptr(), (scalar_t*)y.data_ptr(),
+                                   rows, H, current_stream());
+  });
+  return y;
+}
+
+std::vector<torch::Tensor> bias_gelu_bwd(torch::Tensor dy, torch::Tensor x,
+                                         torch::Tensor bias) {
+  CHECK_INPUT(dy);
+  const int H = x.size(-1);
+  const long rows = x.numel() / H;
+  auto dx = torch::empty_like(x);
+  auto dbias = torch::zeros({H}, x.options().dtype(torch::kFloat));
+  DISPATCH_FLOAT_BF16(x.scalar_type(), "bias_gelu_bwd", [&] {
+    launch_bias_gelu_bwd<scalar_t>((const scalar_t*)dy.data_ptr(),
+                                   (const scalar_t*)x.data_ptr(),
+                                   (const scalar_t*)bias.data_ptr(), (scalar_t*)dx.data_ptr(),
+                                   dbias.data_ptr<float>(), rows, H, current_stream());
+  });
+  return {dx, dbias.to(x.scalar_type())};
+}
+
+torch::Tensor swiglu_fwd(torch::Tensor x12) {
+  CHECK_INPUT(x12);
+  const int H2 = x12.size(-1);
+  const int H = H2 / 2;
+  const long rows = x12.numel() / H2;
+  auto sizes = x12.sizes().vec();
+  sizes.back() = H;
+  auto y = torch::empty(sizes, x12.options());
+  DISPATCH_FLOAT_BF16(x12.scalar_type(), "swiglu_fwd", [&] {
+    launch_swiglu_fwd<scalar_t>((const scalar_t*)x12.data_ptr(), (scalar_t*)y.data_ptr(),
+                                rows, H, current_stream());
+  });
+  return y;
+}
+
+torch::Tensor swiglu_bwd(torch::Tensor dy, torch::Tensor x12) {
+  CHECK_INPUT(dy);
+  const int H2 = x12.size(-1);
+  const int H = H2 / 2;
+  const long rows = x12.numel() / H2;
+  auto dx12 = torch::empty_like(x12);
+  DISPATCH_FLOAT_BF16(x12.scalar_type(), "swiglu_bwd", [&] {
+    launch_swiglu_bwd<scalar_t>((const scalar_t*)dy.data_ptr(),
+                                (const scalar_t*)x12.data_ptr(), (scalar_t*)dx12.data_ptr(),
+                                rows, H, current_stream());
+  });
+  return dx12;
+}
+
+torch::Tensor rope_fwd(torch::Tensor x, torch::Tensor sin_t, torch::Tensor cos_t,
+                       int64_t prefix) {
+  CHECK_INPUT(x);
+  TORCH_CHECK(x.dim() == 4, "rope_fwd expects [B, H, N, hd]");
+  TORCH_CHECK(sin_t.scalar_type() == at::ScalarType::Float, "rope tables must be fp32");
+  const long BH = x.size(0) * x.size(1);
+  const int N = x.size(2);
+  const int hd = x.size(3);
+  const int P = N - (int)prefix;
+  TORCH_CHECK(sin_t.size(0) == P && sin_t.size(1) == hd, "rope table shape mismatch");
+  auto y = torch::empty_like(x);
+  DISPATCH_FLOAT_BF16(x.scalar_type(), "rope_fwd", [&] {
+    launch_rope_fwd<scalar_t>((const scalar_t*)x.data_ptr(), sin_t.data_ptr<float>(),
+                              cos_t.data_ptr<float>(), (scalar_t*)y.data_ptr(), BH, N, P, hd,
+                              current_stream());
+  });
+  return y;
+}
+
+// ---------------------------- multi-tensor ------------------------------
+
+struct MTTables {
+  torch::Tensor device_buf;  // holds sizes + chunk tables + ptr arrays
+  long* sizes;
+  int* chunk_tensor;
+  long* chunk_offset;
+  long* ptr_arrays;  // base of pointer storage (n_lists * n_tensors int64)
+  int n_chunks;
+  int n_tensors;
+};
+
+constexpr long kMTChunk = 65536;
+
+MTTables build_tables(const std::vector<std::vector<torch::Tensor>>& lists,
+                      torch::Device device) {
+  const int n_tensors = (int)lists[0].size();
+  const int n_lists = (int)lists.size();
+  std::vector<long> sizes(n_tensors);
+  std::vector<int> chunk_tensor;
+  std::vector<long> chunk_offset;
+  for (int t = 0; t < n_tensors; ++t) {
+    sizes[t] = lists[0][t].numel();
+    for (long off = 0; off < sizes[t]; off += kMTChunk) {
+      chunk_tensor.push_back(t);
+      chunk_offset.push_back(off);
+    }
+  }
+  const int n_chunks = (int)chunk_tensor.size();
+  // layout (int64 slots): [sizes n_tensors][chunk_tensor n_chunks (as i64)]
+  //                       [chunk_offset n_chunks][ptrs n_lists*n_tensors]
+  const long total = n_tensors + n_chunks + n_chunks + (long)n_lists * n_tensors;
+  auto host = torch::empty({total}, torch::dtype(torch::kLong).pinned_memory(true));
+  long* hp = host.data_ptr<long>();
+  long* p_sizes = hp;
+  long* p_ct = hp + n_tensors;
+  long* p_co = p_ct + n_chunks;
+  long* p_ptr = p_co + n_chunks;
+  for (int t = 0; t < n_tensors; ++t) p_sizes[t] = sizes[t];
+  for (int c = 0; c < n_chunks; ++c) {
+    p_ct[c] = chunk_tensor[c];
+    p_co[c] = chunk_offset[c];
+  }
+  for (int l = 0; l < n_lists; ++l)
+    for (int t = 0; t < n_tensors; ++t)
+      p_ptr[(long)l * n_tensors + t] = (long)(uintptr_t)lists[l][t].data_ptr();
+  auto dev = host.to(device, /*non_blocking=*/true);
+  MTTables out;
+  out.device_buf = dev;
+  long* dp = dev.data_ptr<long>();
+  out.sizes = dp;
+  out.chunk_tensor = nullptr;  // stored as i64; kernel wants int — handled below
+  out.chunk_offset = dp + n_tensors + n_chunks;
+  out.ptr_arrays = dp + n_tensors + 2 * (long)n_chunks;
+  out.n_chunks = n_chunks;
+  out.n_tensors = n_tensors;
+  return out;
+}
+
+// The chunk_tensor table must be int32 for the kernels; build a separate one.
+torch::Tensor build_chunk_tensor_i32(const std::vector<long>& sizes, torch::Device device,
+                                     int* n_chunks_out) {
+  std::vector<int> ct;
+  for (int t = 0; t < (int)sizes.size(); ++t)
+    for (long off = 0; off < sizes[t]; off += kMTChunk) ct.push_back(t);
+  *n_chunks_out = (int)ct.size();
+  auto host = torch::from_blob(ct.data(), {(long)ct.size()}, torch::kInt).clone();
+  return host.to(device);
+}
+
+void multi_tensor_ema(std::vector<torch::Tensor> t_list, std::vector<torch::Tensor> s_list,
+                      double m) {
+  TORCH_CHECK(!t_list.empty() && t_list.size() == s_list.size());
+  auto device = t_list[0].device();
+  auto tables = build_tables({t_list, s_list}, device);
+  std::vector<long> sizes(tables.n_tensors);
+  for (int t = 0; t < tables.n_tensors; ++t) sizes[t] = t_list[t].numel();
+  int n_chunks;
+  auto ct32 = build_chunk_tensor_i32(sizes, device, &n_chunks);
+  auto stream = current_stream();
+  DISPATCH_FLOAT_BF16(t_list[0].scalar_type(), "multi_tensor_ema", [&] {
+    launch_multi_tensor_ema<scalar_t>(
+        (scalar_t* const*)(tables.ptr_arrays),
+        (const scalar_t* const*)(tables.ptr_arrays + tables.n_tensors),
+        tables.sizes, ct32.data_ptr<int>(), tables.chunk_offset, n_chunks, (float)m, stream);
+  });
+}
+
+void multi_tensor_adamw(std::vector<torch::Tensor> p, std::vector<torch::Tensor> g,
+                        std::vector<torch::Tensor> m, std::vector<torch::Tensor> v,
+                        std::vector<torch::Tensor> w, double lr, double beta1, double beta2,
+                        double eps, double weight_decay, double bc1, double bc2,
+                        double grad_scale) {
+  TORCH_CHECK(!p.empty());
+  const bool has_master = !w.empty();
+  auto device = p[0].device();
+  std::vector<std::vector<torch::Tensor>> lists = {p, g, m, v};
+  if (has_master) lists.push_back(w);
+  auto tables = build_tables(lists, device);
+  std::vector<long> sizes(tables.n_tensors);
+  for (int t = 0; t < tables.n_tensors; ++t) sizes[t] = p[t].numel();
+  int n_chunks;
+  auto ct32 = build_chunk_tensor_i32(sizes, device, &n_chunks);
+  auto stream = current_stream();
+  const int nt = tables.n_tensors;
+  DISPATCH_FLOAT_BF16(p[0].scalar_type(), "multi_tensor_adamw", [&] {
+    launch_multi_tensor_adamw<scalar_t>(
+        (scalar_t* const*)(tables.ptr_arrays),
+        (const scalar_t* const*)(tables.ptr_arrays + nt),
+        (float* const*)(tables.ptr_arrays + 2L * nt),
+        (float* const*)(tables.ptr_arrays + 3L * nt),
+        has_master ? (float* const*)(tables.ptr_arrays + 4L * nt) : nullptr,
+        tables.sizes, ct32.data_ptr<int>(), tables.chunk_offset, n_chunks, (float)lr,
+        (float)beta1, (float)beta2, (float)eps, (float)weight_decay, (float)bc1, (float)bc2,
+        (float)grad_scale, has_master, stream);
+  });
+}
+
+torch::Tensor multi_tensor_l2norm_sq(std::vector<torch::Tensor> g) {
+  TORCH_CHECK(!g.empty());
+  auto device = g[0].device();
+  auto tables = build_tables({g}, device);
+  std::vector<long> sizes(tables.n_tensors);
+  for (int t = 0; t < tables.n_tensors; ++t) sizes[t] = g[t].numel();
+  int n_chunks;
+  auto ct32 = build_chunk_tensor_i32(sizes, device, &n_chunks);
+  auto out = torch::zeros({}, torch::dtype(torch::kFloat).device(device));
+  DISPATCH_FLOAT_BF16(g[0].scalar_type(), "multi_tensor_l2norm_sq", [&] {
+    launch_multi_tensor_l2norm_sq<scalar_t>(
+        (const scalar_t* const*)(tables.ptr_arrays), tables.sizes, ct32.data_ptr<int>(),
+        tables.chunk_offset, n_chunks, out.data_ptr<float>(), current_stream());
+  });
+  return out;
+}
+
+}  // namespace
+
+PYBIND11_MODULE(TORCH_EXTENSION_NAME, mod) {
+  mod.def("layernorm_fwd", &layernorm_fwd);
+  mod.def("layernorm_bwd", &layernorm_bwd);
+  mod.def("rmsnorm_fwd", &rmsnorm_fwd);
+  mod.def("rmsnorm_bwd", &rmsnorm_bwd);
+  mod.def("l2norm_fwd", &l2norm_fwd);
+  mod.def("l2norm_bwd", &l2norm_bwd);
+  mod.def("bias_gelu_fwd", &bias_gelu_fwd);
+  mod.def("bias_gelu_bwd", &bias_gelu_bwd);
+  mod.def("swiglu_fwd", &swiglu_fwd);
+  mod.def("swiglu_bwd", &swiglu_bwd);
+  mod.def("rope_fwd", &rope_fwd);
+  mod.def("multi_tensor_ema", &multi_tensor_ema);
+  mod.def("multi_tensor_adamw", &multi_tensor_adamw);
+  mod.def("multi_tensor_l2norm_sq", &multi_tensor_l2norm_sq);
+}

@@ -1,0 +1,194 @@
+// Fused elementwise kernels: bias+GELU (K8 epilogue), SwiGLU gate (K9),
+// RoPE rotate-half application (K5).
+
+#include "common.h"
+
+#define EW_BLOCK 256
+
+DEV_INLINE float gelu_tanh(float x) {
+  // tanh approximation (matches torch F.gelu(approximate="tanh"))
+  const float k0 = 0.7978845608028654f;  // sqrt(2/pi)
+  const float k1 = 0.044715f;
+  float inner = k0 * (x + k1 * x * x * x);
+  return 0.5f * x * (1.0f + tanhf(inner));
+}
+
+DEV_INLINE float gelu_tanh_grad(float x) {
+  const float k0 = 0.7978845608028654f;
+  const float k1 = 0.044715f;
+  float x2 = x * x;
+  float inner = k0 * (x + k1 * x * x2);
+  float t = tanhf(inner);
+  float sech2 = 1.0f - t * t;
+  return 0.5f * (1.0f + t) + 0.5f * x * sech2 * k0 * (1.0f + 3.0f * k1 * x2);
+}
+
+// ---------------------------- bias + gelu ------------------------------
+
+template <typename T>
+__global__ void bias_gelu_fwd_kernel(const T* __restrict__ x, const T* __restrict__ bias,
+                                     T* __restrict__ y, long rows, int H) {
+  long total = rows * (long)H;
+  for (long idx = blockIdx.x * (long)blockDim.x + threadIdx.x; idx < total;
+       idx += (long)gridDim.x * blockDim.x) {
+    int col = (int)(idx % H);
+    float v = ScalarOps<T>::load(x + idx) + ScalarOps<T>::load(bias + col);
+    ScalarOps<T>::store(y + idx, gelu_tanh(v));
+  }
+}
+
+template <typename T>
+__global__ void bias_gelu_bwd_kernel(const T* __restrict__ dy, const T* __restrict__ x,
+                                     const T* __restrict__ bias, T* __restrict__ dx,
+                                     float* __restrict__ dbias, long rows, int H) {
+  extern __shared__ float db_acc[];  // [H]
+  for (int i = threadIdx.x; i < H; i += blockDim.x) db_acc[i] = 0.f;
+  __syncthreads();
+  long total = rows * (long)H;
+  for (long idx = blockIdx.x * (long)blockDim.x + threadIdx.x; idx < total;
+       idx += (long)gridDim.x * blockDim.x) {
+    int col = (int)(idx % H);
+    float v = ScalarOps<T>::load(x + idx) + ScalarOps<T>::load(bias + col);
+    float g = ScalarOps<T>::load(dy + idx) * gelu_tanh_grad(v);
+    ScalarOps<T>::store(dx + idx, g);
+    atomicAdd(db_acc + col, g);  // LDS atomic (cross-wave same-column safety)
+  }
+  __syncthreads();
+  for (int i = threadIdx.x; i < H; i += blockDim.x) atomicAdd(dbias + i, db_acc[i]);
+}
+
+// ------------------------------ swiglu ---------------------------------
+// x12 = [rows, 2H] as [x1 | x2]; y = silu(x1) * x2.
+
+template <typename T>
+__global__ void swiglu_fwd_kernel(const T* __restrict__ x12, T* __restrict__ y,
+                                  long rows, int H) {
+  long total = rows * (long)H;
+  for (long idx = blockIdx.x * (long)blockDim.x + threadIdx.x; idx < total;
+       idx += (long)gridDim.x * blockDim.x) {
+    long row = idx / H;
+    int col = (int)(idx % H);
+    const T* xr = x12 + row * (long)(2 * H);
+    float x1 = ScalarOps<T>::load(xr + col);
+    float x2 = ScalarOps<T>::load(xr + H + col);
+    float sig = 1.0f / (1.0f + expf(-x1));
+    ScalarOps<T>::store(y + idx, x1 * sig * x2);
+  }
+}
+
+template <typename T>
+__global__ void swiglu_bwd_kernel(const T* __restrict__ dy, const T* __restrict__ x12,
+                                  T* __restrict__ dx12, long rows, int H) {
+  long total = rows * (long)H;
+  for (long idx = blockIdx.x * (long)blockDim.x + threadIdx.x; idx < total;
+       idx += (long)gridDim.x * blockDim.x) {
+    long row = idx / H;
+    int col = (int)(idx % H);
+    const T* xr = x12 + row * (long)(2 * H);
+    T* dxr = dx12 + row * (long)(2 * H);
+    float x1 = ScalarOps<T>::load(xr + col);
+    float x2 = ScalarOps<T>::load(xr + H + col);
+    float g = ScalarOps<T>::load(dy + idx);
+    float sig = 1.0f / (1.0f + expf(-x1));
+    float silu = x1 * sig;
+    float dsilu = sig * (1.0f + x1 * (1.0f - sig));
+    ScalarOps<T>::store(dxr + col, g * x2 * dsilu);
+    ScalarOps<T>::store(dxr + H + col, g * silu);
+  }
+}
+
+// ------------------------------- RoPE ----------------------------------
+// x: [B, Hh, N, hd] contiguous; sin/cos: [P, hd] fp32 with P = N - prefix.
+// rotate-half: out[j] = x[j]*cos[j] - x[j+hd/2]*sin[j]          (j < hd/2)
+//              out[j] = x[j]*cos[j] + x[j-hd/2]*sin[j]          (j >= hd/2)
+// Tokens [0, prefix) pass through unchanged (cls + storage tokens).
+// One thread per pair (j < hd/2) handles both halves: 2 loads, 2 stores.
+
+template <typename T>
+__global__ void rope_fwd_kernel(const T* __restrict__ x, const float* __restrict__ sin_t,
+                                const float* __restrict__ cos_t, T* __restrict__ y,
+                                long BH, int N, int P, int hd) {
+  const int half = hd / 2;
+  long total = BH * (long)N * half;
+  for (long idx = blockIdx.x * (long)blockDim.x + threadIdx.x; idx < total;
+       idx += (long)gridDim.x * blockDim.x) {
+    int j = (int)(idx % half);
+    long t = idx / half;          // BH*N token index
+    int n = (int)(t % N);
+    long base = t * (long)hd;
+    int p = n - (N - P);          // patch index (negative for prefix tokens)
+    float lo = ScalarOps<T>::load(x + base + j);
+    float hi = ScalarOps<T>::load(x + base + half + j);
+    if (p < 0) {
+      ScalarOps<T>::store(y + base + j, lo);
+      ScalarOps<T>::store(y + base + half + j, hi);
+    } else {
+      const float* srow = sin_t + (long)p * hd;
+      const float* crow = cos_t + (long)p * hd;
+      float s_lo = srow[j], c_lo = crow[j];
+      float s_hi = srow[half + j], c_hi = crow[half + j];
+      ScalarOps<T>::store(y + base + j, lo * c_lo - hi * s_lo);
+      ScalarOps<T>::store(y + base + half + j, hi * c_hi + lo * s_hi);
+    }
+  }
+}
+
+// ---------------------------- C wrappers -------------------------------
+
+template <typename T>
+void launch_bias_gelu_fwd(const T* x, const T* bias, T* y, long rows, int H,
+                          hipStream_t stream) {
+  long total = rows * (long)H;
+  int grid = (int)min((total + EW_BLOCK - 1) / EW_BLOCK, (long)2048);
+  hipLaunchKernelGGL((bias_gelu_fwd_kernel<T>), dim3(grid), dim3(EW_BLOCK), 0, stream,
+                     x, bias, y, rows, H);
+}
+
+template <typename T>
+void launch_bias_gelu_bwd(const T* dy, const T* x, const T* bias, T* dx, float* dbias,
+                          long rows, int H, hipStream_t stream) {
+  long total = rows * (long)H;
+  int grid = (int)min((total + EW_BLOCK - 1) / EW_BLOCK, (long)1024);
+  size_t shmem = (size_t)H * sizeof(float);
+  hipLaunchKernelGGL((bias_gelu_bwd_kernel<T>), dim3(grid), dim3(EW_BLOCK), shmem, stream,
+                     dy, x, bias, dx, dbias, rows, H);
+}
+
+template <typename T>
+void launch_swiglu_fwd(const T* x12, T* y, long rows, int H, hipStream_t stream) {
+  long total = rows * (long)H;
+  int grid = (int)min((total + EW_BLOCK - 1) / EW_BLOCK, (long)2048);
+  hipLaunchKernelGGL((swiglu_fwd_kernel<T>), dim3(grid), dim3(EW_BLOCK), 0, stream,
+                     x12, y, rows, H);
+}
+
+template <typename T>
+void launch_swiglu_bwd(const T* dy, const T* x12, T* dx12, long rows, int H,
+                       hipStream_t stream) {
+  long total = rows * (long)H;
+  int grid = (int)min((total + EW_BLOCK - 1) / EW_BLOCK, (long)2048);
+  hipLaunchKernelGGL((swiglu_bwd_kernel<T>), dim3(grid), dim3(EW_BLOCK), 0, stream,
+                     dy, x12, dx12, rows, H);
+}
+
+template <typename T>
+void launch_rope_fwd(const T* x, const float* sin_t, const float* cos_t, T* y, long BH,
+                     int N, int P, int hd, hipStream_t stream) {
+  long total = BH * (long)N * (hd / 2);
+  int grid = (int)min((total + EW_BLOCK - 1) / EW_BLOCK, (long)4096);
+  hipLaunchKernelGGL((rope_fwd_kernel<T>), dim3(grid), dim3(EW_BLOCK), 0, stream,
+                     x, sin_t, cos_t, y, BH, N, P, hd);
+}
+
+#define INSTANTIATE_EW(T)                                                            \
+  template void launch_bias_gelu_fwd<T>(const T*, const T*, T*, long, int,           \
+                                        hipStream_t);                                \
+  template void launch_bias_gelu_bwd<T>(const T*, const T*, const T*, T*, float*,    \
+                                        long, int, hipStream_t);                     \
+  template void launch_swiglu_fwd<T>(const T*, T*, long, int, hipStream_t);          \
+  template void launch_swiglu_bwd<T>(const T*, const T*, T*, long, int, hipStream_t);\
+  template void launch_rope_fwd<T>(const T*, const float*, const float*, T*, long,   \
+                                   int, int, int, hipStream_t);
+
+INSTANTIATE_EW(float)
+INSTANTIATE_EW(__hip_bfloat16)

@@ -1,0 +1,374 @@
+"""SSL meta-architecture: DINOv3 student/teacher orchestration.
+
+Parity: dinov3_jax/train/ssl_meta_arch.py:32-659, with the reference's wiring
+bugs fixed per SURVEY §8:
+- B2: update_ema blends the LIVE teacher params from the student;
+- B7: the teacher forward runs under torch.no_grad() (halves activation
+  memory; the reference relied on grads only being taken wrt student leaves).
+"""
+
+from __future__ import annotations
+
+import logging
+from typing import Dict, Optional, Tuple
+
+import torch
+import torch.nn as nn
+
+from ..layers.dino_head import DINOHead
+from ..loss import DINOLoss, GramLoss, KoLeoLoss, KoLeoLossDistributed, iBOTPatchLoss
+from ..models import build_model_from_cfg
+from ..ops import ema_update_
+from ..utils.utils import count_parameters
+from .param_groups import get_params_groups_with_decay
+
+logger = logging.getLogger("dinov3")
+
+
+class SSLMetaArch(nn.Module):
+    def __init__(self, config):
+        super().__init__()
+        self.config = config
+        assert config.crops.local_crops_number > 0
+        assert config.ibot.separate_head is True
+        assert config.train.centering == "sinkhorn_knopp"
+
+        student_backbone, teacher_backbone, embed_dim = build_model_from_cfg(config)
+        self.student_backbone = student_backbone
+        self.teacher_backbone = teacher_backbone
+        self.embed_dim = embed_dim
+        self.dino_out_dim = config.dino.head_n_prototypes
+        self.n_local_crops = config.crops.local_crops_number
+        logger.info("student params: %.1fM", count_parameters(student_backbone) / 1e6)
+
+        def dino_head():
+            return DINOHead(
+                in_dim=embed_dim,
+                out_dim=config.dino.head_n_prototypes,
+                hidden_dim=config.dino.head_hidden_dim,
+                bottleneck_dim=config.dino.head_bottleneck_dim,
+                nlayers=config.dino.head_nlayers,
+            )
+
+        def ibot_head():
+            return DINOHead(
+                in_dim=embed_dim,
+                out_dim=config.ibot.head_n_prototypes,
+                hidden_dim=config.ibot.head_hidden_dim,
+                bottleneck_dim=config.ibot.head_bottleneck_dim,
+                nlayers=config.ibot.head_nlayers,
+            )
+
+        self.student_dino_head = dino_head()
+        self.teacher_dino_head = dino_head()
+        self.student_ibot_head = ibot_head()
+        self.teacher_ibot_head = ibot_head()
+
+        self.dino_loss = DINOLoss(self.dino_out_dim)
+        self.ibot_patch_loss = iBOTPatchLoss(config.ibot.head_n_prototypes)
+        if config.dino.koleo_loss_distributed:
+            assert config.dino.koleo_distributed_replicas == 0
+            self.koleo_loss = KoLeoLossDistributed(
+                topk=config.dino.koleo_topk,
+                loss_group_size=config.dino.koleo_distributed_loss_group_size,
+            )
+        else:
+            assert config.dino.koleo_topk == 1
+            self.koleo_loss = KoLeoLoss()
+
+        self.dino_loss_weight = config.dino.loss_weight
+        self.dino_global_ignore_diagonal = config.dino.global_ignore_diagonal
+        self.dino_koleo_loss_weight = config.dino.koleo_loss_weight
+        self.ibot_loss_weight = config.ibot.loss_weight
+
+        # gram anchoring
+        self.gram_use_loss = config.gram.use_loss
+        self.gram_img_level = config.gram.img_level
+        self.gram_loss_weight = config.gram.loss_weight
+        self.gram_compute_stats = config.gram.compute_stats
+        self.has_gram_teacher = self.gram_use_loss
+        if self.gram_use_loss:
+            gram_backbone, _ = build_model_from_cfg(config, only_teacher=True)
+            self.gram_backbone = gram_backbone
+            self.gram_loss = GramLoss(
+                apply_norm=config.gram.normalized,
+                img_level=config.gram.img_level,
+                remove_neg=config.gram.remove_neg,
+                remove_only_teacher_neg=config.gram.remove_only_teacher_neg,
+            )
+        else:
+            self.gram_backbone = None
+
+        # the teacher tower never takes gradients and starts as a copy of the student
+        self._sync_teacher_from_student()
+        for module in (self.teacher_backbone, self.teacher_dino_head, self.teacher_ibot_head):
+            module.requires_grad_(False)
+        if self.gram_backbone is not None:
+            self.gram_backbone.requires_grad_(False)
+
+    # ------------------------------------------------------------------
+    @torch.no_grad()
+    def _sync_teacher_from_student(self) -> None:
+        self.teacher_backbone.load_state_dict(self.student_backbone.state_dict())
+        self.teacher_dino_head.load_state_dict(self.student_dino_head.state_dict())
+        self.teacher_ibot_head.load_state_dict(self.student_ibot_head.state_dict())
+
+    def _teacher_student_param_pairs(self):
+        pairs = []
+        for t_mod, s_mod in (
+            (self.teacher_backbone, self.student_backbone),
+            (self.teacher_dino_head, self.student_dino_head),
+            (self.teacher_ibot_head, self.student_ibot_head),
+        ):
+            t_params = dict(t_mod.named_parameters())
+            for name, s_param in s_mod.named_parameters():
+                pairs.append((t_params[name], s_param))
+        return pairs
+
+    @torch.no_grad()
+    def update_ema(self, momentum: float) -> None:
+        """teacher <- m*teacher + (1-m)*student, one fused multi-tensor kernel."""
+        pairs = self._teacher_student_param_pairs()
+        ema_update_([t for t, _ in pairs], [s.detach() for _, s in pairs], momentum)
+
+    # ------------------------------------------------------------------
+    def forward(self, data: Dict[str, torch.Tensor], *, teacher_temp: float,
+                iteration: int = 0) -> Tuple[torch.Tensor, Dict[str, torch.Tensor]]:
+        metrics: Dict[str, torch.Tensor] = {}
+        n_global_crops = 2
+        n_local_crops = self.n_local_crops
+        B = data["collated_local_crops"].shape[0] // n_local_crops
+        metrics["local_batch_size"] = B
+
+        global_crops = data["collated_global_crops"]
+        local_crops = data["collated_local_crops"]
+        masks = data["collated_masks"]
+        mask_indices_list = data["mask_indices_list"]
+        masks_weight = data["masks_weight"]
+        n_masked_patches_tensor = data["n_masked_patches"]
+
+        teacher_global = self.get_teacher_output(
+            global_crops, n_global_crops=n_global_crops, B=B,
+            teacher_temp=teacher_temp,
+            n_masked_patches_tensor=n_masked_patches_tensor,
+            mask_indices_list=mask_indices_list,
+        )
+        student_global, student_local = self.get_student_output(
+            global_crops=global_crops, local_crops=local_crops,
+            n_global_crops=n_global_crops, n_local_crops=n_local_crops, B=B,
+            masks=masks, mask_indices_list=mask_indices_list,
+        )
+
+        gram_global: Dict[str, torch.Tensor] = {}
+        if self.gram_use_loss:
+            gram_global = self.get_gram_teacher_output(
+                data.get("collated_gram_teacher_crops"),
+                n_global_crops=n_global_crops, B=B,
+                teacher_global=teacher_global, student_global=student_global,
+            )
+
+        loss, loss_dict = self.compute_losses(
+            teacher_global=teacher_global,
+            student_global=student_global,
+            student_local=student_local,
+            gram_global=gram_global,
+            masks=masks,
+            mask_indices_list=mask_indices_list,
+            masks_weight=masks_weight,
+            iteration=iteration,
+        )
+        # (the cross-rank loss pmean for logging happens in the trainer; grads
+        # are averaged by the gradient reduction, matching C8 semantics)
+        metrics.update(loss_dict)
+        return loss, metrics
+
+    # ------------------------------------------------------------------
+    @torch.no_grad()
+    def get_teacher_output(self, global_crops: torch.Tensor, *, n_global_crops: int, B: int,
+                           teacher_temp: float, n_masked_patches_tensor: torch.Tensor,
+                           mask_indices_list: torch.Tensor) -> Dict[str, torch.Tensor]:
+        out = self.teacher_backbone(global_crops, is_training=True)
+        cls = out["x_norm_clstoken"]          # [2B, D]
+        reg = out["x_storage_tokens"]         # [2B, R, D]
+        ibot_patch = out["x_norm_patchtokens"]  # [2B, P, D]
+
+        buffer = ibot_patch.reshape(-1, ibot_patch.shape[-1])[mask_indices_list]
+        masked_patch_after_head = self.teacher_ibot_head(buffer)
+        cls_after_head = self.teacher_dino_head(cls)
+
+        cls_centered = self.dino_loss.sinkhorn_knopp_teacher(
+            cls_after_head, teacher_temp=teacher_temp,
+        ).reshape(n_global_crops, B, -1)
+        masked_patch_centered = self.ibot_patch_loss.sinkhorn_knopp_teacher(
+            masked_patch_after_head, teacher_temp=teacher_temp,
+            n_masked_patches_tensor=n_masked_patches_tensor,
+        )
+        D = cls.shape[-1]
+        return {
+            "cls_pre_head": cls.reshape(n_global_crops, B, D),
+            "reg_pre_head": reg.reshape(n_global_crops, B, *reg.shape[1:]),
+            "patch_pre_head": ibot_patch.reshape(n_global_crops, B, *ibot_patch.shape[1:]),
+            "cls_after_head": cls_after_head.reshape(n_global_crops, B, -1),
+            "cls_centered": cls_centered,
+            "masked_patch_centered": masked_patch_centered,
+        }
+
+    def get_student_output(self, *, global_crops: torch.Tensor, local_crops: torch.Tensor,
+                           n_global_crops: int, n_local_crops: int, B: int,
+                           masks: torch.Tensor, mask_indices_list: torch.Tensor):
+        global_out, local_out = self.student_backbone(
+            [global_crops, local_crops], masks=[masks, None], is_training=True,
+        )
+        g_cls = global_out["x_norm_clstoken"]
+        g_reg = global_out["x_storage_tokens"]
+        g_patch = global_out["x_norm_patchtokens"]
+        l_cls = local_out["x_norm_clstoken"]
+        l_reg = local_out["x_storage_tokens"]
+        l_patch = local_out["x_norm_patchtokens"]
+
+        masked_patches_pre_head = g_patch.reshape(-1, g_patch.shape[-1])[mask_indices_list]
+        global_masked_patch_after_head = self.student_ibot_head(masked_patches_pre_head)
+
+        # one DINO-head pass over [global cls | local cls]
+        split = g_cls.shape[0]
+        buffer = torch.cat([g_cls, l_cls], dim=0)
+        buffer = self.student_dino_head(buffer)
+        g_after, l_after = buffer[:split], buffer[split:]
+
+        D = g_cls.shape[-1]
+        student_global = {
+            "cls_pre_head": g_cls.reshape(n_global_crops, B, D),
+            "reg_pre_head": g_reg.reshape(n_global_crops, B, *g_reg.shape[1:]),
+            "patch_pre_head": g_patch.reshape(n_global_crops, B, *g_patch.shape[1:]),
+            "cls_after_head": g_after.reshape(n_global_crops, B, -1),
+            "masked_patch_after_head": global_masked_patch_after_head,
+            "masked_patch_pre_head": masked_patches_pre_head,
+        }
+        student_local = {
+            "cls_pre_head": l_cls.reshape(n_local_crops, B, D),
+            "reg_pre_head": l_reg.reshape(n_local_crops, B, *l_reg.shape[1:]),
+            "patch_pre_head": l_patch.reshape(n_local_crops, B, *l_patch.shape[1:]),
+            "cls_after_head": l_after.reshape(n_local_crops, B, -1),
+        }
+        return student_global, student_local
+
+    @torch.no_grad()
+    def get_gram_teacher_output(self, gram_teacher_crops: Optional[torch.Tensor], *,
+                                n_global_crops: int, B: int,
+                                teacher_global: Dict[str, torch.Tensor],
+                                student_global: Dict[str, torch.Tensor]) -> Dict[str, torch.Tensor]:
+        """Gram-teacher patch features for the gram anchoring loss."""
+        if gram_teacher_crops is not None and self.gram_backbone is not None:
+            out = self.gram_backbone(gram_teacher_crops, is_training=True)
+            teacher_patches = out["x_norm_patchtokens"]
+        else:
+            teacher_patches = teacher_global["patch_pre_head"].reshape(
+                -1, *teacher_global["patch_pre_head"].shape[2:]
+            )
+        student_patches = student_global["patch_pre_head"].reshape(
+            -1, *student_global["patch_pre_head"].shape[2:]
+        )
+        if teacher_patches.shape[1] != student_patches.shape[1]:
+            # resize teacher token grid to the student's (bicubic over the 2D grid)
+            import math
+
+            import torch.nn.functional as F
+
+            t = teacher_patches
+            n_t = int(math.sqrt(t.shape[1]))
+            n_s = int(math.sqrt(student_patches.shape[1]))
+            t = t.reshape(t.shape[0], n_t, n_t, -1).permute(0, 3, 1, 2)
+            t = F.interpolate(t.float(), size=(n_s, n_s), mode=self.config.gram.global_teacher_resize_method,
+                              antialias=self.config.gram.global_teacher_resize_antialias)
+            teacher_patches = t.permute(0, 2, 3, 1).reshape(t.shape[0], n_s * n_s, -1).to(student_patches.dtype)
+        return {
+            "teacher_patches": teacher_patches.detach(),
+            "student_patches": student_patches,
+            "orig_student_patches": student_patches,
+            "orig_teacher_patches": teacher_patches.detach(),
+        }
+
+    @torch.no_grad()
+    def update_gram_teacher(self) -> None:
+        """Refresh the gram teacher from the EMA teacher (gram.rep_update cadence)."""
+        if self.gram_backbone is not None:
+            self.gram_backbone.load_state_dict(self.teacher_backbone.state_dict())
+
+    # ------------------------------------------------------------------
+    def compute_losses(self, *, teacher_global, student_global, student_local, gram_global,
+                       masks, mask_indices_list, masks_weight, iteration):
+        n_global_crops = student_global["cls_after_head"].shape[0]
+        n_local_crops = student_local["cls_after_head"].shape[0]
+        loss_dict: Dict[str, torch.Tensor] = {}
+        loss_accumulator = torch.zeros((), device=masks.device, dtype=torch.float32)
+
+        dino_global_terms = (
+            n_global_crops * (n_global_crops - 1) if self.dino_global_ignore_diagonal else n_global_crops**2
+        )
+        dino_local_terms = n_global_crops * n_local_crops
+        dino_global_scale = dino_global_terms / (dino_global_terms + dino_local_terms)
+        dino_local_scale = dino_local_terms / (dino_global_terms + dino_local_terms)
+        koleo_scale = n_global_crops
+
+        dino_local_crops_loss = self.dino_loss(
+            student_logits=student_local["cls_after_head"],
+            teacher_probs=teacher_global["cls_centered"],
+        )
+        loss_dict["dino_local_crops_loss"] = dino_local_crops_loss
+        local_weight = 1.0
+        if self.config.dino.reweight_dino_local_loss:
+            from .cosine_lr_scheduler import linear_warmup_cosine_decay  # noqa: F401
+
+            sched = self.config.dino.local_loss_weight_schedule
+            local_weight = float(sched.peak)
+        loss_dict["dino_local_loss_weight"] = local_weight
+        loss_accumulator = loss_accumulator + self.dino_loss_weight * dino_local_scale * local_weight * dino_local_crops_loss
+
+        dino_global_crops_loss = self.dino_loss(
+            student_logits=student_global["cls_after_head"],
+            teacher_probs=teacher_global["cls_centered"],
+            ignore_diagonal=self.dino_global_ignore_diagonal,
+        )
+        loss_dict["dino_global_crops_loss"] = dino_global_crops_loss
+        loss_accumulator = loss_accumulator + self.dino_loss_weight * dino_global_scale * dino_global_crops_loss
+
+        koleo_loss = sum(self.koleo_loss(x) for x in student_global["cls_pre_head"]) / n_global_crops
+        loss_dict["koleo_loss"] = koleo_loss
+        loss_accumulator = loss_accumulator + self.dino_koleo_loss_weight * koleo_scale * koleo_loss
+
+        ibot_loss = self.ibot_patch_loss.forward_masked(
+            student_global["masked_patch_after_head"],
+            teacher_global["masked_patch_centered"],
+            student_masks_flat=masks,
+            n_masked_patches=mask_indices_list.shape[0],
+            masks_weight=masks_weight,
+        )
+        loss_dict["ibot_loss"] = ibot_loss
+        loss_accumulator = loss_accumulator + self.ibot_loss_weight * ibot_loss
+
+        if self.gram_use_loss and gram_global:
+            gram_loss = self.gram_loss(
+                gram_global["student_patches"], gram_global["teacher_patches"], img_level=self.gram_img_level,
+            )
+            loss_dict["gram_loss"] = gram_loss
+            loss_dict["gram_loss_weight"] = self.gram_loss_weight
+            loss_accumulator = loss_accumulator + self.gram_loss_weight * gram_loss
+
+        loss_dict["total_loss"] = loss_accumulator
+        return loss_accumulator, loss_dict
+
+    # ------------------------------------------------------------------
+    def student_submodels(self) -> Dict[str, nn.Module]:
+        return {
+            "backbone": self.student_backbone,
+            "dino_head": self.student_dino_head,
+            "ibot_head": self.student_ibot_head,
+        }
+
+    def get_params_groups(self):
+        return get_params_groups_with_decay(
+            self.student_submodels(),
+            lr_decay_rate=self.config.optim.layerwise_decay,
+            patch_embed_lr_mult=self.config.optim.patch_embed_lr_mult,
+            dino_head_wd_multiplier=self.config.optim.dino_head_wd_multiplier,
+        )

@@ -1,0 +1,313 @@
+"""Training entrypoint + hot loop.
+
+Parity: dinov3_jax/train/train.py (CLI :51-72, schedulers :127-208, optimizer
+:75-122, do_train :319-712) with the reference's defects fixed (SURVEY §8 B1:
+optimizer updates APPLY; B2: EMA updates the live teacher).
+
+MI355X process model: torchrun one rank per GPU over RCCL; batches move
+host->device with pinned memory + non-blocking copies; gradient reduction
+overlaps backward (parallel/ddp.py or the sharding engine in parallel/fsdp.py).
+"""
+
+from __future__ import annotations
+
+import argparse
+import logging
+import math
+import sys
+import time
+from functools import partial
+from typing import Dict, Optional
+
+import torch
+
+from .. import parallel
+from ..checkpointer import find_latest_checkpoint, load_checkpoint, save_checkpoint
+from ..configs import setup_config, setup_job
+from ..data import (
+    DataAugmentationDINO,
+    MaskingGenerator,
+    SamplerType,
+    collate_data_and_cast,
+    make_data_loader,
+    make_dataset,
+)
+from ..logging import MetricLogger, setup_logging
+from ..parallel.ddp import GradReducer, all_reduce_scalar_sums
+from .cosine_lr_scheduler import CosineScheduler, linear_warmup_cosine_decay
+from .optim import FusedAdamW
+from .ssl_meta_arch import SSLMetaArch
+
+logger = logging.getLogger("dinov3")
+
+DTYPE_MAP = {"bf16": torch.bfloat16, "fp16": torch.float16, "fp32": torch.float32,
+             "float32": torch.float32, "float16": torch.float16, "bfloat16": torch.bfloat16}
+
+
+def get_args_parser(add_help: bool = True) -> argparse.ArgumentParser:
+    parser = argparse.ArgumentParser("DINOv3 MI355X training", add_help=add_help)
+    parser.add_argument("--config-file", default="", metavar="FILE", help="path to config file")
+    parser.add_argument("--no-resume", action="store_true", help="do not attempt to resume")
+    parser.add_argument("--eval-only", action="store_true")
+    parser.add_argument("--eval", type=str, default="", help="eval type")
+    parser.add_argument("--profiling", action="store_true", help="emit roctx/profiler ranges")
+    parser.add_argument("--multi-distillation", action="store_true")
+    parser.add_argument("--seed", type=int, default=0)
+    parser.add_argument("--output-dir", default="", help="output directory")
+    parser.add_argument("--max-iterations", type=int, default=-1, help="cap iterations (debug/bench)")
+    parser.add_argument("opts", nargs=argparse.REMAINDER,
+                        help="config overrides as key=value dotlist")
+    return parser
+
+
+def build_schedulers(cfg) -> Dict[str, object]:
+    epoch_len = cfg.train.OFFICIAL_EPOCH_LENGTH
+    total = cfg.optim.epochs * epoch_len
+    schedules_v2 = cfg.get("schedules")
+    if schedules_v2:
+        def mk(s, total_iters):
+            return linear_warmup_cosine_decay(
+                start=s["start"], peak=s["peak"], end=s["end"],
+                warmup_iterations=int(s.get("warmup_epochs", 0) * epoch_len),
+                total_iterations=total_iters,
+                cosine_iterations=int(s["cosine_epochs"] * epoch_len) if "cosine_epochs" in s else None,
+            )
+
+        lr = mk(schedules_v2["lr"], total)
+        wd = mk(schedules_v2["weight_decay"], total)
+        momentum = mk(schedules_v2["momentum"], total)
+        teacher_temp = mk(schedules_v2["teacher_temp"], total)
+        freeze_iters = int(schedules_v2["lr"].get("freeze_last_layer_epochs", 0) * epoch_len)
+    else:
+        lr = CosineScheduler(
+            base_value=cfg.optim.lr, final_value=cfg.optim.min_lr, total_iters=total,
+            warmup_iters=int(cfg.optim.warmup_epochs * epoch_len),
+            start_warmup_value=0.0, trunc_extra=cfg.optim.schedule_trunc_extra,
+        )
+        wd = CosineScheduler(
+            base_value=cfg.optim.weight_decay, final_value=cfg.optim.weight_decay_end, total_iters=total,
+        )
+        momentum = CosineScheduler(
+            base_value=cfg.teacher.momentum_teacher, final_value=cfg.teacher.final_momentum_teacher,
+            total_iters=total,
+        )
+        teacher_temp = CosineScheduler(
+            base_value=cfg.teacher.teacher_temp, final_value=cfg.teacher.teacher_temp, total_iters=total,
+            warmup_iters=int(cfg.teacher.warmup_teacher_temp_epochs * epoch_len),
+            start_warmup_value=cfg.teacher.warmup_teacher_temp,
+        )
+        freeze_iters = int(cfg.optim.freeze_last_layer_epochs * epoch_len)
+    return {
+        "lr": lr,
+        "wd": wd,
+        "momentum": momentum,
+        "teacher_temp": teacher_temp,
+        "freeze_last_layer_iterations": freeze_iters,
+        "total_iterations": total,
+    }
+
+
+def build_optimizer(cfg, params_groups) -> FusedAdamW:
+    return FusedAdamW(
+        params_groups,
+        beta1=cfg.optim.adamw_beta1,
+        beta2=cfg.optim.adamw_beta2,
+        use_master_weights=True,
+    )
+
+
+def build_data_loader_from_cfg(cfg, model: SSLMetaArch, sampler_advance: int = 0):
+    img_size = cfg.crops.global_crops_size
+    patch_size = cfg.student.patch_size
+    n_tokens = (img_size // patch_size) ** 2
+    mask_generator = MaskingGenerator(
+        input_size=(img_size // patch_size, img_size // patch_size),
+        max_num_patches=int(0.5 * n_tokens),
+    )
+    transform = DataAugmentationDINO(
+        cfg.crops.global_crops_scale,
+        cfg.crops.local_crops_scale,
+        cfg.crops.local_crops_number,
+        global_crops_size=cfg.crops.global_crops_size,
+        local_crops_size=cfg.crops.local_crops_size,
+        gram_teacher_crops_size=cfg.crops.gram_teacher_crops_size,
+        gram_teacher_no_distortions=cfg.crops.gram_teacher_no_distortions,
+        local_crops_subset_of_global_crops=cfg.crops.localcrops_subset_of_globalcrops,
+        patch_size=patch_size,
+        share_color_jitter=cfg.crops.share_color_jitter,
+        horizontal_flips=cfg.crops.horizontal_flips,
+        mean=cfg.crops.rgb_mean,
+        std=cfg.crops.rgb_std,
+    )
+    collate_fn = partial(
+        collate_data_and_cast,
+        mask_ratio_tuple=tuple(cfg.ibot.mask_ratio_min_max),
+        mask_probability=cfg.ibot.mask_sample_probability,
+        n_tokens=n_tokens,
+        mask_generator=mask_generator,
+        random_circular_shift=cfg.ibot.mask_random_circular_shift,
+        dtype=DTYPE_MAP.get(cfg.compute_precision.param_dtype, torch.float32),
+    )
+    dataset = make_dataset(
+        dataset_str=cfg.train.dataset_path,
+        transform=transform,
+        target_transform=lambda _: (),
+    )
+    return make_data_loader(
+        dataset=dataset,
+        batch_size=cfg.train.batch_size_per_gpu,
+        num_workers=cfg.train.num_workers,
+        shuffle=True,
+        seed=cfg.train.seed,
+        sampler_type=SamplerType.EPOCH,
+        sampler_advance=sampler_advance,
+        drop_last=True,
+        collate_fn=collate_fn,
+    )
+
+
+def batch_to_device(data: dict, device: torch.device) -> dict:
+    return {
+        k: (v.to(device, non_blocking=True) if isinstance(v, torch.Tensor) else v)
+        for k, v in data.items()
+    }
+
+
+def do_train(cfg, model: SSLMetaArch, resume: bool = True, max_iterations: int = -1):
+    device = parallel.device()
+    param_dtype = DTYPE_MAP.get(cfg.compute_precision.param_dtype, torch.float32)
+    if device.type == "cuda":
+        model = model.to(device=device, dtype=param_dtype)
+    else:
+        model = model.to(device)  # fp32 reference path on CPU
+    model.train()
+
+    schedulers = build_schedulers(cfg)
+    params_groups = model.get_params_groups()
+    optimizer = build_optimizer(cfg, params_groups)
+    student_params = [p for g in params_groups for p in g["params"]]
+    reducer = GradReducer(student_params, reduce_dtype=torch.float32
+                          if cfg.compute_precision.reduce_dtype == "fp32" else None)
+
+    start_iter = 0
+    output_dir = cfg.train.output_dir
+    if resume and output_dir:
+        latest = find_latest_checkpoint(output_dir)
+        if latest is not None:
+            payload = load_checkpoint(latest, model, optimizer, strict=True)
+            start_iter = payload["iteration"] + 1
+            logger.info("resumed at iteration %d", start_iter)
+
+    data_loader = build_data_loader_from_cfg(cfg, model)
+    total_iterations = schedulers["total_iterations"]
+    if max_iterations > 0:
+        total_iterations = min(total_iterations, start_iter + max_iterations)
+    ckpt_period = cfg.checkpointing.period
+
+    metrics_file = None
+    if output_dir:
+        import os
+
+        metrics_file = os.path.join(output_dir, "training_metrics.json")
+    metric_logger = MetricLogger(delimiter="  ", output_file=metrics_file)
+    header = "Train"
+
+    nan_count = 0
+    iteration = start_iter
+
+    def infinite_batches():
+        epoch = 0
+        while True:
+            if hasattr(data_loader.sampler, "set_epoch"):
+                data_loader.sampler.set_epoch(epoch)
+            yield from data_loader
+            epoch += 1
+
+    clip = cfg.optim.clip_grad
+    for data in metric_logger.log_every(
+        infinite_batches(), 10, header, n_iterations=total_iterations, start_iteration=start_iter
+    ):
+        if iteration >= total_iterations:
+            break
+        it = iteration
+        lr = schedulers["lr"][it]
+        wd = schedulers["wd"][it]
+        mom = schedulers["momentum"][it]
+        teacher_temp = schedulers["teacher_temp"][it]
+        last_layer_lr = 0.0 if it < schedulers["freeze_last_layer_iterations"] else lr
+
+        data = batch_to_device(data, device)
+        loss, loss_dict = model(data, teacher_temp=teacher_temp, iteration=it)
+
+        if not torch.isfinite(loss):
+            nan_count += 1
+            nan_logger = logging.getLogger("nan")
+            nan_logger.error("NaN/Inf loss at iteration %d (%d consecutive)", it, nan_count)
+            if nan_count > 2:
+                raise FloatingPointError(f"aborting: >{nan_count - 1} consecutive non-finite losses")
+            optimizer.zero_grad()
+            iteration += 1
+            continue
+        nan_count = 0
+
+        loss.backward()
+        reducer.finalize()
+
+        clip_scales: Optional[Dict[str, float]] = None
+        if clip is not None and clip > 0:
+            sums = optimizer.grad_norms_per_submodel()
+            sums = all_reduce_scalar_sums(sums) if parallel.get_world_size() > 1 else sums
+            clip_scales = {}
+            for name, s in sums.items():
+                norm = float(s) ** 0.5
+                clip_scales[name] = min(1.0, clip / (norm + 1e-6))
+                loss_dict[f"grad_norm_{name}"] = norm
+
+        optimizer.step(lr=lr, weight_decay=wd, last_layer_lr=last_layer_lr, clip_scales=clip_scales)
+        optimizer.zero_grad()
+        model.update_ema(mom)
+
+        if model.gram_use_loss and cfg.gram.rep_update and it >= cfg.gram.it_first_update:
+            if (it - cfg.gram.it_first_update) % cfg.gram.update_frequency == 0:
+                model.update_gram_teacher()
+
+        metric_logger.update(
+            lr=lr, wd=wd, mom=mom, last_layer_lr=last_layer_lr, teacher_temp=teacher_temp,
+            total_loss=loss.item(),
+            **{k: (v.item() if isinstance(v, torch.Tensor) else v) for k, v in loss_dict.items()
+               if k != "total_loss"},
+        )
+
+        if output_dir and ckpt_period > 0 and (it + 1) % ckpt_period == 0:
+            save_checkpoint(
+                output_dir, it, model, optimizer,
+                max_to_keep=cfg.checkpointing.max_to_keep, keep_every=cfg.checkpointing.keep_every,
+            )
+        iteration += 1
+
+    if output_dir:
+        save_checkpoint(
+            output_dir, iteration - 1, model, optimizer,
+            max_to_keep=cfg.checkpointing.max_to_keep, keep_every=cfg.checkpointing.keep_every,
+        )
+    metric_logger.synchronize_between_processes()
+    logger.info("training done at iteration %d", iteration)
+    return {k: meter.global_avg for k, meter in metric_logger.meters.items()}
+
+
+def do_test(cfg, model, iteration: int):
+    raise NotImplementedError("eval harness lands with the eval subsystem")
+
+
+def main(argv=None):
+    args = get_args_parser().parse_args(argv)
+    setup_job(output_dir=args.output_dir or None, seed=args.seed)
+    cfg = setup_config(args)
+    model = SSLMetaArch(cfg)
+    if args.eval_only:
+        return do_test(cfg, model, 0)
+    return do_train(cfg, model, resume=not args.no_resume, max_iterations=args.max_iterations)
+
+
+if __name__ == "__main__":
+    main(sys.argv[1:])

@@ -1,0 +1,104 @@
+import math
+
+import pytest
+import torch
+import torch.nn.functional as F
+
+from dinov3_amd.loss import DINOLoss, GramLoss, KoLeoLoss, iBOTPatchLoss
+from dinov3_amd.ops.proto_scores import sinkhorn_knopp
+
+
+def test_sinkhorn_doubly_stochastic_invariants():
+    torch.manual_seed(0)
+    M, K = 16, 64
+    logits = torch.randn(M, K)
+    Q = sinkhorn_knopp(logits, teacher_temp=1.0, n_iterations=20)
+    assert Q.shape == (M, K)
+    # rows (samples): sum to 1 after the final B rescale (col-norm is last)
+    assert torch.allclose(Q.sum(dim=1), torch.ones(M), atol=1e-3)
+    # columns (prototypes): converge to uniform mass B/K
+    assert torch.allclose(Q.sum(dim=0), torch.full((K,), M / K), rtol=0.1)
+    assert (Q >= 0).all()
+    # 3 iterations at low temp (training setting) still yields valid rows
+    Q3 = sinkhorn_knopp(logits, teacher_temp=0.07)
+    assert torch.allclose(Q3.sum(dim=1), torch.ones(M), atol=1e-3)
+    assert (Q3 >= 0).all()
+
+
+def test_dino_loss_matches_handwritten():
+    torch.manual_seed(0)
+    S, T, B, K = 4, 2, 3, 16
+    loss_mod = DINOLoss(out_dim=K)
+    student = torch.randn(S, B, K)
+    teacher = torch.softmax(torch.randn(T, B, K), dim=-1)
+    got = loss_mod(student, teacher)
+    logp = F.log_softmax(student / 0.1, dim=-1)
+    want = -(torch.einsum("sbk,tbk->", logp, teacher)) / (B * S * T)
+    assert torch.allclose(got, want, atol=1e-5)
+
+
+def test_dino_loss_ignore_diagonal():
+    torch.manual_seed(0)
+    S = T = 2
+    B, K = 3, 16
+    loss_mod = DINOLoss(out_dim=K)
+    student = torch.randn(S, B, K)
+    teacher = torch.softmax(torch.randn(T, B, K), dim=-1)
+    got = loss_mod(student, teacher, ignore_diagonal=True)
+    logp = F.log_softmax(student / 0.1, dim=-1)
+    total = 0.0
+    for s in range(S):
+        for t in range(T):
+            if s == t:
+                continue
+            total += -(logp[s] * teacher[t]).sum()
+    want = total / (B * S * T - B * min(S, T))
+    assert torch.allclose(got, want, atol=1e-5)
+
+
+def test_ibot_loss_masked_weighting():
+    torch.manual_seed(0)
+    B, N, K = 4, 8, 16
+    loss_mod = iBOTPatchLoss(patch_out_dim=K)
+    masks = torch.zeros(B, N, dtype=torch.bool)
+    masks[0, :3] = True
+    masks[1, :1] = True
+    idx = masks.flatten().nonzero().flatten()
+    n_masked = idx.numel()
+    student = torch.randn(n_masked, K)
+    teacher = torch.softmax(torch.randn(n_masked, K), dim=-1)
+    weights = (1 / masks.sum(-1).clamp(min=1.0)).unsqueeze(-1).expand_as(masks)[masks]
+    got = loss_mod.forward_masked(student, teacher, masks, n_masked_patches=n_masked,
+                                 masks_weight=weights)
+    logp = F.log_softmax(student / 0.1, dim=-1)
+    per_row = (teacher * logp).sum(-1) * weights
+    want = -per_row.sum() / B
+    assert torch.allclose(got, want, atol=1e-5)
+
+
+def test_koleo_loss_basic():
+    torch.manual_seed(0)
+    loss_mod = KoLeoLoss()
+    x = torch.randn(16, 8)
+    out = loss_mod(x)
+    assert out.ndim == 0 and torch.isfinite(out)
+    # a degenerate batch (identical rows) must blow the loss up (-log small)
+    x2 = torch.ones(8, 4) + 1e-4 * torch.randn(8, 4)
+    assert loss_mod(x2) > out
+
+
+def test_gram_loss_zero_for_identical():
+    torch.manual_seed(0)
+    g = GramLoss(apply_norm=True, remove_neg=True)
+    x = torch.randn(2, 5, 8)
+    assert g(x, x.clone()).abs() < 1e-10
+    y = torch.randn(2, 5, 8)
+    assert g(x, y) > 0
+
+
+def test_gram_loss_batch_level():
+    g = GramLoss(apply_norm=True, remove_neg=True)
+    x = torch.randn(2, 5, 8)
+    y = torch.randn(2, 5, 8)
+    out = g(x, y, img_level=False)
+    assert out.ndim == 0 and out > 0

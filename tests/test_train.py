@@ -1,0 +1,109 @@
+import os
+
+import numpy as np
+import pytest
+import torch
+
+from dinov3_amd.train.cosine_lr_scheduler import CosineScheduler, linear_warmup_cosine_decay
+from dinov3_amd.train.param_groups import get_params_groups_with_decay, get_vit_lr_decay_rate
+from dinov3_amd.train.optim import FusedAdamW
+
+
+def test_cosine_scheduler_shape():
+    s = CosineScheduler(base_value=1.0, final_value=0.1, total_iters=100, warmup_iters=10,
+                        start_warmup_value=0.0, freeze_iters=5)
+    assert len(s.schedule) == 100
+    assert s[0] == 0.0  # freeze
+    assert s[5] == 0.0  # warmup start
+    assert abs(s[14] - 1.0) < 1e-6  # warmup end ~ base
+    assert abs(s[99] - 0.1) < 0.01
+    assert s[1000] == 0.1  # past end -> final
+
+
+def test_cosine_scheduler_trunc_extra():
+    s = CosineScheduler(base_value=1.0, final_value=0.1, total_iters=100, trunc_extra=0.25)
+    assert len(s.schedule) == 100
+    assert abs(s[0] - 1.0) < 1e-6
+    assert abs(s[99] - 0.1) < 1e-6
+
+
+def test_linear_warmup_cosine_decay():
+    s = linear_warmup_cosine_decay(start=0.0, peak=1.0, end=0.01, warmup_iterations=10,
+                                   total_iterations=50)
+    assert len(s.schedule) == 50
+    assert s[0] == 0.0
+    assert abs(s[10] - 1.0) < 1e-6
+    assert abs(s[49] - 0.01) < 0.05
+
+
+def test_layerwise_decay_rates():
+    # patch_embed -> layer 0, block i -> i+1, norm -> n+1
+    assert get_vit_lr_decay_rate("patch_embed.proj.weight", 0.9, 12) == pytest.approx(0.9**13)
+    assert get_vit_lr_decay_rate("blocks.0.attn.qkv.weight", 0.9, 12) == pytest.approx(0.9**12)
+    assert get_vit_lr_decay_rate("blocks.11.mlp.fc1.weight", 0.9, 12) == pytest.approx(0.9**1)
+    assert get_vit_lr_decay_rate("norm.weight", 0.9, 12) == pytest.approx(1.0)
+
+
+def test_param_groups_fuse_and_flags():
+    from dinov3_amd.models.vision_transformer import vit_small
+    from dinov3_amd.layers.dino_head import DINOHead
+
+    backbone = vit_small(img_size=32, layerscale_init=1e-5)
+    head = DINOHead(in_dim=384, out_dim=64, nlayers=2, hidden_dim=32, bottleneck_dim=16)
+    groups = get_params_groups_with_decay(
+        {"backbone": backbone, "dino_head": head},
+        lr_decay_rate=0.9, patch_embed_lr_mult=0.2, dino_head_wd_multiplier=0.5,
+    )
+    total = sum(len(g["params"]) for g in groups)
+    want = sum(1 for _ in backbone.parameters()) + sum(1 for _ in head.parameters())
+    assert total == want
+    last_layer_groups = [g for g in groups if g["is_last_layer"]]
+    assert last_layer_groups and all(g["submodel"] == "dino_head" for g in last_layer_groups)
+    # biases/norms have wd 0
+    for g in groups:
+        for name in g["names"]:
+            if name.endswith(".bias") or "norm" in name:
+                assert g["wd_multiplier"] == 0.0
+
+
+def test_fused_adamw_matches_torch_adamw():
+    torch.manual_seed(0)
+    p1 = torch.nn.Parameter(torch.randn(10, 10))
+    p2 = torch.nn.Parameter(torch.randn(7))
+    groups = [{"params": [p1, p2], "names": ["a", "b"], "submodel": "backbone",
+               "lr_multiplier": 1.0, "wd_multiplier": 1.0, "is_last_layer": False}]
+    opt = FusedAdamW(groups, beta1=0.9, beta2=0.999, use_master_weights=False)
+
+    ref_p1 = p1.detach().clone().requires_grad_(True)
+    ref_p2 = p2.detach().clone().requires_grad_(True)
+    ref_opt = torch.optim.AdamW([ref_p1, ref_p2], lr=0.01, betas=(0.9, 0.999),
+                                eps=1e-8, weight_decay=0.05)
+    for step in range(3):
+        g1, g2 = torch.randn_like(p1), torch.randn_like(p2)
+        p1.grad, p2.grad = g1.clone(), g2.clone()
+        ref_p1.grad, ref_p2.grad = g1.clone(), g2.clone()
+        opt.step(lr=0.01, weight_decay=0.05)
+        ref_opt.step()
+    assert torch.allclose(p1, ref_p1, atol=1e-5)
+    assert torch.allclose(p2, ref_p2, atol=1e-5)
+
+
+def test_fused_adamw_last_layer_freeze():
+    p = torch.nn.Parameter(torch.randn(5, 5))
+    before = p.detach().clone()
+    groups = [{"params": [p], "names": ["last_layer.weight"], "submodel": "dino_head",
+               "lr_multiplier": 1.0, "wd_multiplier": 0.0, "is_last_layer": True}]
+    opt = FusedAdamW(groups, use_master_weights=False)
+    p.grad = torch.randn_like(p)
+    opt.step(lr=0.01, weight_decay=0.0, last_layer_lr=0.0)
+    assert torch.allclose(p, before)
+
+
+def test_ema_update_moves_teacher():
+    from dinov3_amd.ops import ema_update_
+
+    t = [torch.ones(4), torch.zeros(3)]
+    s = [torch.zeros(4), torch.ones(3)]
+    ema_update_(t, s, momentum=0.9)
+    assert torch.allclose(t[0], torch.full((4,), 0.9))
+    assert torch.allclose(t[1], torch.full((3,), 0.1))
