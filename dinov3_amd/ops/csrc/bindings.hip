@@ -38,6 +38,18 @@ void launch_swiglu_bwd(const T*, const T*, T*, long, int, hipStream_t);
 template <typename T>
 void launch_rope_fwd(const T*, const float*, const float*, T*, long, int, int, int,
                      hipStream_t);
+void launch_probe_mfma(const __hip_bfloat16*, const __hip_bfloat16*, float*, hipStream_t);
+void launch_fmha_fwd(const __hip_bfloat16*, const __hip_bfloat16*, const __hip_bfloat16*,
+                     __hip_bfloat16*, float*, int, int, int, float, hipStream_t);
+void launch_fmha_bwd_pre(const __hip_bfloat16*, const __hip_bfloat16*, float*, long, int,
+                         hipStream_t);
+void launch_fmha_bwd_dq(const __hip_bfloat16*, const __hip_bfloat16*, const __hip_bfloat16*,
+                        const __hip_bfloat16*, const float*, const float*, __hip_bfloat16*,
+                        int, int, int, float, hipStream_t);
+void launch_fmha_bwd_dkv(const __hip_bfloat16*, const __hip_bfloat16*,
+                         const __hip_bfloat16*, const __hip_bfloat16*, const float*,
+                         const float*, __hip_bfloat16*, __hip_bfloat16*, int, int, int,
+                         float, hipStream_t);
 template <typename T>
 void launch_multi_tensor_ema(T* const*, const T* const*, const long*, const int*,
                              const long*, int, float, hipStream_t);
@@ -241,6 +253,61 @@ torch::Tensor rope_fwd(torch::Tensor x, torch::Tensor sin_t, torch::Tensor cos_t
   return y;
 }
 
+// -------------------------------- fmha ----------------------------------
+
+torch::Tensor probe_mfma(torch::Tensor A, torch::Tensor B) {
+  CHECK_INPUT(A);
+  CHECK_INPUT(B);
+  TORCH_CHECK(A.scalar_type() == at::ScalarType::BFloat16);
+  auto C = torch::empty({32, 32}, A.options().dtype(torch::kFloat));
+  launch_probe_mfma((const __hip_bfloat16*)A.data_ptr(), (const __hip_bfloat16*)B.data_ptr(),
+                    C.data_ptr<float>(), current_stream());
+  return C;
+}
+
+std::vector<torch::Tensor> fmha_fwd(torch::Tensor q, torch::Tensor k, torch::Tensor v) {
+  CHECK_INPUT(q);
+  CHECK_INPUT(k);
+  CHECK_INPUT(v);
+  TORCH_CHECK(q.scalar_type() == at::ScalarType::BFloat16, "fmha: bf16 only");
+  TORCH_CHECK(q.dim() == 4, "fmha expects [B, H, N, hd]");
+  const int B = q.size(0), H = q.size(1), N = q.size(2), HD = q.size(3);
+  TORCH_CHECK(HD == 64 || HD == 128, "fmha: head_dim must be 64 or 128, got ", HD);
+  auto o = torch::empty_like(q);
+  auto lse = torch::empty({B, H, N}, q.options().dtype(torch::kFloat));
+  const float scale = 1.0f / std::sqrt((float)HD);
+  launch_fmha_fwd((const __hip_bfloat16*)q.data_ptr(), (const __hip_bfloat16*)k.data_ptr(),
+                  (const __hip_bfloat16*)v.data_ptr(), (__hip_bfloat16*)o.data_ptr(),
+                  lse.data_ptr<float>(), B * H, N, HD, scale, current_stream());
+  return {o, lse};
+}
+
+std::vector<torch::Tensor> fmha_bwd(torch::Tensor dout, torch::Tensor q, torch::Tensor k,
+                                    torch::Tensor v, torch::Tensor o, torch::Tensor lse) {
+  CHECK_INPUT(dout);
+  const int B = q.size(0), H = q.size(1), N = q.size(2), HD = q.size(3);
+  const float scale = 1.0f / std::sqrt((float)HD);
+  auto D = torch::empty({B, H, N}, q.options().dtype(torch::kFloat));
+  auto dq = torch::empty_like(q);
+  auto dk = torch::empty_like(k);
+  auto dv = torch::empty_like(v);
+  auto stream = current_stream();
+  launch_fmha_bwd_pre((const __hip_bfloat16*)dout.data_ptr(),
+                      (const __hip_bfloat16*)o.data_ptr(), D.data_ptr<float>(),
+                      (long)B * H * N, HD, stream);
+  launch_fmha_bwd_dq((const __hip_bfloat16*)q.data_ptr(), (const __hip_bfloat16*)k.data_ptr(),
+                     (const __hip_bfloat16*)v.data_ptr(),
+                     (const __hip_bfloat16*)dout.data_ptr(), lse.data_ptr<float>(),
+                     D.data_ptr<float>(), (__hip_bfloat16*)dq.data_ptr(), B * H, N, HD,
+                     scale, stream);
+  launch_fmha_bwd_dkv((const __hip_bfloat16*)q.data_ptr(), (const __hip_bfloat16*)k.data_ptr(),
+                      (const __hip_bfloat16*)v.data_ptr(),
+                      (const __hip_bfloat16*)dout.data_ptr(), lse.data_ptr<float>(),
+                      D.data_ptr<float>(), (__hip_bfloat16*)dk.data_ptr(),
+                      (__hip_bfloat16*)dv.data_ptr(), B * H, N, HD, scale, stream);
+  return {dq, dk, dv};
+}
+
 // ---------------------------- multi-tensor ------------------------------
 
 struct MTTables {
@@ -390,6 +457,9 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, mod) {
   mod.def("swiglu_fwd", &swiglu_fwd);
   mod.def("swiglu_bwd", &swiglu_bwd);
   mod.def("rope_fwd", &rope_fwd);
+  mod.def("probe_mfma", &probe_mfma);
+  mod.def("fmha_fwd", &fmha_fwd);
+  mod.def("fmha_bwd", &fmha_bwd);
   mod.def("multi_tensor_ema", &multi_tensor_ema);
   mod.def("multi_tensor_adamw", &multi_tensor_adamw);
   mod.def("multi_tensor_l2norm_sq", &multi_tensor_l2norm_sq);

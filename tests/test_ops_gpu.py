@@ -1,0 +1,234 @@
+"""GPU kernel numerics: each HIP kernel vs a plain PyTorch fp32 reference."""
+
+import math
+
+import pytest
+import torch
+
+pytestmark = pytest.mark.gpu
+
+DEV = "cuda:0"
+
+
+def _assert_close(got, want, atol, rtol=1e-2, what=""):
+    got = got.float()
+    want = want.float()
+    err = (got - want).abs().max().item()
+    scale = want.abs().max().item()
+    assert err <= atol + rtol * scale, f"{what}: max err {err} (scale {scale})"
+
+
+@pytest.fixture(scope="module")
+def ops():
+    from dinov3_amd.ops import hip_ops
+
+    return hip_ops()
+
+
+def test_mfma_layout_probe(ops):
+    """C[32,32] = A[32,16] @ B[16,32]; asymmetric inputs (transpose-detecting)."""
+    torch.manual_seed(0)
+    A = torch.randn(32, 16, device=DEV).bfloat16()
+    B = torch.randn(16, 32, device=DEV).bfloat16()
+    C = ops.probe_mfma(A.contiguous(), B.contiguous())
+    ref = A.float() @ B.float()
+    _assert_close(C, ref, atol=0.05, what="mfma probe")
+
+
+@pytest.mark.parametrize("rows,D", [(64, 1024), (197, 384), (1000, 4096), (33, 256)])
+def test_layernorm_fwd_bwd(ops, rows, D):
+    torch.manual_seed(0)
+    x = torch.randn(rows, D, device=DEV).bfloat16().requires_grad_(True)
+    w = torch.randn(D, device=DEV).bfloat16().requires_grad_(True)
+    b = torch.randn(D, device=DEV).bfloat16().requires_grad_(True)
+    from dinov3_amd.ops import layer_norm
+
+    y = layer_norm(x, w, b, eps=1e-6)
+    xr = x.detach().float().requires_grad_(True)
+    wr = w.detach().float().requires_grad_(True)
+    br = b.detach().float().requires_grad_(True)
+    yr = torch.nn.functional.layer_norm(xr, (D,), wr, br, 1e-6)
+    _assert_close(y, yr, atol=0.05, what="ln fwd")
+    dy = torch.randn_like(y)
+    y.backward(dy)
+    yr.backward(dy.float())
+    _assert_close(x.grad, xr.grad, atol=0.08, what="ln dx")
+    _assert_close(w.grad, wr.grad, atol=0.3, rtol=2e-2, what="ln dw")
+    _assert_close(b.grad, br.grad, atol=0.3, rtol=2e-2, what="ln db")
+
+
+def test_rmsnorm_fwd_bwd(ops):
+    torch.manual_seed(1)
+    rows, D = 128, 512
+    x = torch.randn(rows, D, device=DEV).bfloat16().requires_grad_(True)
+    w = torch.randn(D, device=DEV).bfloat16().requires_grad_(True)
+    from dinov3_amd.ops import rms_norm
+
+    y = rms_norm(x, w, eps=1e-6)
+    xr = x.detach().float().requires_grad_(True)
+    wr = w.detach().float().requires_grad_(True)
+    yr = xr * torch.rsqrt(xr.pow(2).mean(-1, keepdim=True) + 1e-6) * wr
+    _assert_close(y, yr, atol=0.05, what="rms fwd")
+    dy = torch.randn_like(y)
+    y.backward(dy)
+    yr.backward(dy.float())
+    _assert_close(x.grad, xr.grad, atol=0.08, what="rms dx")
+    _assert_close(w.grad, wr.grad, atol=0.3, rtol=2e-2, what="rms dw")
+
+
+def test_l2norm_fwd_bwd(ops):
+    torch.manual_seed(2)
+    rows, D = 200, 256
+    x = torch.randn(rows, D, device=DEV).bfloat16().requires_grad_(True)
+    from dinov3_amd.ops import l2_normalize
+
+    y = l2_normalize(x, eps=1e-12)
+    xr = x.detach().float().requires_grad_(True)
+    yr = xr / (xr.norm(dim=-1, keepdim=True) + 1e-12)
+    _assert_close(y, yr, atol=0.02, what="l2 fwd")
+    dy = torch.randn_like(y)
+    y.backward(dy)
+    yr.backward(dy.float())
+    _assert_close(x.grad, xr.grad, atol=0.02, what="l2 dx")
+
+
+def test_bias_gelu_fwd_bwd(ops):
+    torch.manual_seed(3)
+    rows, H = 256, 1024
+    x = torch.randn(rows, H, device=DEV).bfloat16().requires_grad_(True)
+    b = torch.randn(H, device=DEV).bfloat16().requires_grad_(True)
+    from dinov3_amd.ops import bias_gelu
+
+    y = bias_gelu(x, b)
+    xr = x.detach().float().requires_grad_(True)
+    br = b.detach().float().requires_grad_(True)
+    yr = torch.nn.functional.gelu(xr + br, approximate="tanh")
+    _assert_close(y, yr, atol=0.03, what="bias_gelu fwd")
+    dy = torch.randn_like(y)
+    y.backward(dy)
+    yr.backward(dy.float())
+    _assert_close(x.grad, xr.grad, atol=0.05, what="bias_gelu dx")
+    _assert_close(b.grad, br.grad, atol=0.5, rtol=2e-2, what="bias_gelu db")
+
+
+def test_swiglu_fwd_bwd(ops):
+    torch.manual_seed(4)
+    rows, H = 128, 344
+    x = torch.randn(rows, 2 * H, device=DEV).bfloat16().requires_grad_(True)
+    from dinov3_amd.ops.bias_act import swiglu_gate
+
+    y = swiglu_gate(x)
+    xr = x.detach().float().requires_grad_(True)
+    yr = torch.nn.functional.silu(xr[..., :H]) * xr[..., H:]
+    _assert_close(y, yr, atol=0.05, what="swiglu fwd")
+    dy = torch.randn_like(y)
+    y.backward(dy)
+    yr.backward(dy.float())
+    _assert_close(x.grad, xr.grad, atol=0.05, what="swiglu dx")
+
+
+@pytest.mark.parametrize("prefix", [0, 1, 5])
+def test_rope_fwd_bwd(ops, prefix):
+    torch.manual_seed(5)
+    B, H, N, hd = 2, 4, 37, 64
+    P = N - prefix
+    x = torch.randn(B, H, N, hd, device=DEV).bfloat16().requires_grad_(True)
+    angles = torch.rand(P, hd // 2, device=DEV) * 6.28
+    angles = torch.cat([angles, angles], dim=-1)
+    sin, cos = angles.sin(), angles.cos()
+    from dinov3_amd.ops import rope_apply
+
+    y = rope_apply(x, sin, cos, prefix)
+    xr = x.detach().float().requires_grad_(True)
+    h = hd // 2
+    tail = xr[..., prefix:, :]
+    rot = torch.cat([-tail[..., h:], tail[..., :h]], dim=-1)
+    yr = torch.cat([xr[..., :prefix, :], tail * cos + rot * sin], dim=-2)
+    _assert_close(y, yr, atol=0.03, what="rope fwd")
+    dy = torch.randn_like(y)
+    y.backward(dy)
+    yr.backward(dy.float())
+    _assert_close(x.grad, xr.grad, atol=0.03, what="rope dx")
+
+
+@pytest.mark.parametrize("N,hd", [(37, 64), (197, 64), (201, 64), (128, 64), (197, 128), (530, 64)])
+def test_fmha_fwd_matches_reference(ops, N, hd):
+    torch.manual_seed(6)
+    B, H = 3, 4
+    q = torch.randn(B, H, N, hd, device=DEV).bfloat16()
+    k = torch.randn(B, H, N, hd, device=DEV).bfloat16()
+    v = torch.randn(B, H, N, hd, device=DEV).bfloat16()
+    from dinov3_amd.ops.fmha import fmha_ref
+
+    o, lse = ops.fmha_fwd(q, k, v)
+    ref = fmha_ref(q.float(), k.float(), v.float())
+    _assert_close(o, ref, atol=0.03, what=f"fmha fwd N={N}")
+    # lse sanity: softmax denominators positive/finite
+    assert torch.isfinite(lse).all()
+
+
+@pytest.mark.parametrize("N,hd", [(37, 64), (197, 64), (197, 128)])
+def test_fmha_bwd_matches_reference(ops, N, hd):
+    torch.manual_seed(7)
+    B, H = 2, 3
+    q = torch.randn(B, H, N, hd, device=DEV).bfloat16().requires_grad_(True)
+    k = torch.randn(B, H, N, hd, device=DEV).bfloat16().requires_grad_(True)
+    v = torch.randn(B, H, N, hd, device=DEV).bfloat16().requires_grad_(True)
+    from dinov3_amd.ops import fmha
+
+    o = fmha(q, k, v)
+    dy = torch.randn_like(o)
+    o.backward(dy)
+
+    qr = q.detach().float().requires_grad_(True)
+    kr = k.detach().float().requires_grad_(True)
+    vr = v.detach().float().requires_grad_(True)
+    s = torch.einsum("bhqd,bhkd->bhqk", qr, kr) / math.sqrt(hd)
+    p = torch.softmax(s, dim=-1)
+    oref = torch.einsum("bhqk,bhkd->bhqd", p, vr)
+    oref.backward(dy.float())
+    _assert_close(q.grad, qr.grad, atol=0.05, what=f"fmha dq N={N}")
+    _assert_close(k.grad, kr.grad, atol=0.05, what=f"fmha dk N={N}")
+    _assert_close(v.grad, vr.grad, atol=0.05, what=f"fmha dv N={N}")
+
+
+def test_multi_tensor_ema_gpu(ops):
+    torch.manual_seed(8)
+    t = [torch.randn(1000, device=DEV).bfloat16(), torch.randn(3, 7, device=DEV).bfloat16()]
+    s = [torch.randn_like(x) for x in t]
+    t_ref = [x.float().clone() for x in t]
+    ops.multi_tensor_ema(t, s, 0.9)
+    for tr, sv, tv in zip(t_ref, s, t):
+        want = 0.9 * tr + 0.1 * sv.float()
+        _assert_close(tv, want, atol=0.02, what="ema")
+
+
+def test_multi_tensor_adamw_gpu(ops):
+    torch.manual_seed(9)
+    shapes = [(64, 64), (130,), (1000, 3)]
+    p32 = [torch.randn(s, device=DEV) for s in shapes]
+    p = [x.bfloat16() for x in p32]
+    master = [x.clone() for x in p32]
+    g = [torch.randn(s, device=DEV).bfloat16() for s in shapes]
+    m = [torch.zeros(s, device=DEV) for s in shapes]
+    v = [torch.zeros(s, device=DEV) for s in shapes]
+    ops.multi_tensor_adamw(p, g, m, v, master, 0.01, 0.9, 0.999, 1e-8, 0.05,
+                           1 - 0.9, 1 - 0.999, 1.0)
+    # torch reference on fp32 copies
+    for i, s in enumerate(shapes):
+        ref = p32[i].clone()
+        gm = g[i].float()
+        mm = 0.1 * gm
+        vv = 0.001 * gm * gm
+        mh = mm / (1 - 0.9)
+        vh = vv / (1 - 0.999)
+        ref = ref * (1 - 0.01 * 0.05) - 0.01 * mh / (vh.sqrt() + 1e-8)
+        _assert_close(master[i], ref, atol=1e-4, what=f"adamw master {i}")
+        _assert_close(p[i], ref, atol=0.01, what=f"adamw param {i}")
+
+
+def test_multi_tensor_l2norm_gpu(ops):
+    g = [torch.randn(100, device=DEV).bfloat16(), torch.randn(55, 3, device=DEV).bfloat16()]
+    got = ops.multi_tensor_l2norm_sq(g)
+    want = sum(x.float().pow(2).sum() for x in g)
+    _assert_close(got, want, atol=0.5, rtol=1e-2, what="l2norm_sq")
