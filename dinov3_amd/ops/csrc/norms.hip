@@ -204,11 +204,12 @@ __global__ void l2norm_bwd_kernel(
       dot += ScalarOps<T>::load(dyr + i) * ScalarOps<T>::load(yr + i);
     }
     dot = block_reduce_sum(dot, red);
+    // y = x * s, s = 1/(n+eps)  =>  dx = s*dy - (dot(dy,y)/n) * y
     const float c = dot / fmaxf(n, 1e-20f);
     for (int i = threadIdx.x; i < D; i += blockDim.x) {
       float g = ScalarOps<T>::load(dyr + i);
       float yv = ScalarOps<T>::load(yr + i);
-      ScalarOps<T>::store(dxr + i, s * (g - c * yv));
+      ScalarOps<T>::store(dxr + i, s * g - c * yv);
     }
     __syncthreads();
   }
