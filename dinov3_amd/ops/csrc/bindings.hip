@@ -38,6 +38,20 @@ void launch_swiglu_bwd(const T*, const T*, T*, long, int, hipStream_t);
 template <typename T>
 void launch_rope_fwd(const T*, const float*, const float*, T*, long, int, int, int,
                      hipStream_t);
+void launch_dino_ce_fwd(const __hip_bfloat16*, const float*, float*, float*, float*, int,
+                        int, int, long, float, bool, hipStream_t);
+void launch_dino_ce_bwd(const __hip_bfloat16*, const float*, const float*, const float*,
+                        const float*, __hip_bfloat16*, int, int, int, long, float, bool,
+                        hipStream_t);
+void launch_ibot_ce_fwd(const __hip_bfloat16*, const float*, const float*, float*, float*,
+                        float*, int, long, float, hipStream_t);
+void launch_ibot_ce_bwd(const __hip_bfloat16*, const float*, const float*, const float*,
+                        const float*, const float*, __hip_bfloat16*, int, long, float,
+                        hipStream_t);
+void launch_sinkhorn_exp(const __hip_bfloat16*, float*, float*, long, float, hipStream_t);
+void launch_sinkhorn_colsum(const float*, float*, int, long, const float*, hipStream_t);
+void launch_sinkhorn_div_row(float*, const float*, int, long, float, const float*, bool,
+                             hipStream_t);
 void launch_probe_mfma(const __hip_bfloat16*, const __hip_bfloat16*, float*, hipStream_t);
 void launch_fmha_fwd(const __hip_bfloat16*, const __hip_bfloat16*, const __hip_bfloat16*,
                      __hip_bfloat16*, float*, int, int, int, float, hipStream_t);
@@ -179,6 +193,7 @@ torch::Tensor l2norm_bwd(torch::Tensor dy, torch::Tensor y, torch::Tensor s, dou
 torch::Tensor bias_gelu_fwd(torch::Tensor x, torch::Tensor bias) {
   CHECK_INPUT(x);
   const int H = x.size(-1);
+  TORCH_CHECK(H % 8 == 0, "bias_gelu: H must be a multiple of 8");
   const long rows = x.numel() / H;
   auto y = torch::empty_like(x);
   DISPATCH_FLOAT_BF16(x.scalar_type(), "bias_gelu_fwd", [&] {
@@ -251,6 +266,96 @@ torch::Tensor rope_fwd(torch::Tensor x, torch::Tensor sin_t, torch::Tensor cos_t
                               current_stream());
   });
   return y;
+}
+
+// ------------------------------ proto CE --------------------------------
+
+std::vector<torch::Tensor> dino_ce_fwd(torch::Tensor x, torch::Tensor t, double temp,
+                                       bool ignore_diag) {
+  CHECK_INPUT(x);
+  CHECK_INPUT(t);
+  TORCH_CHECK(x.dim() == 3 && t.dim() == 3, "dino_ce expects [S,B,K] and [T,B,K]");
+  TORCH_CHECK(x.scalar_type() == at::ScalarType::BFloat16);
+  TORCH_CHECK(t.scalar_type() == at::ScalarType::Float);
+  const int S = x.size(0), B = x.size(1);
+  const int T = t.size(0);
+  const long K = x.size(2);
+  auto lse = torch::empty({S * B}, x.options().dtype(torch::kFloat));
+  auto st = torch::empty({S * B}, x.options().dtype(torch::kFloat));
+  auto loss = torch::zeros({}, x.options().dtype(torch::kFloat));
+  launch_dino_ce_fwd((const __hip_bfloat16*)x.data_ptr(), t.data_ptr<float>(),
+                     lse.data_ptr<float>(), st.data_ptr<float>(), loss.data_ptr<float>(),
+                     S, T, B, K, (float)(1.0 / temp), ignore_diag, current_stream());
+  return {loss, lse, st};
+}
+
+torch::Tensor dino_ce_bwd(torch::Tensor g, torch::Tensor x, torch::Tensor t,
+                          torch::Tensor lse, torch::Tensor st, double temp,
+                          bool ignore_diag) {
+  const int S = x.size(0), B = x.size(1);
+  const int T = t.size(0);
+  const long K = x.size(2);
+  auto dx = torch::empty_like(x);
+  launch_dino_ce_bwd((const __hip_bfloat16*)x.data_ptr(), t.data_ptr<float>(),
+                     lse.data_ptr<float>(), st.data_ptr<float>(), g.data_ptr<float>(),
+                     (__hip_bfloat16*)dx.data_ptr(), S, T, B, K, (float)(1.0 / temp),
+                     ignore_diag, current_stream());
+  return dx;
+}
+
+std::vector<torch::Tensor> ibot_ce_fwd(torch::Tensor x, torch::Tensor t, torch::Tensor w,
+                                       double temp) {
+  CHECK_INPUT(x);
+  CHECK_INPUT(t);
+  const int M = x.size(0);
+  const long K = x.size(1);
+  auto lse = torch::empty({M}, x.options().dtype(torch::kFloat));
+  auto st = torch::empty({M}, x.options().dtype(torch::kFloat));
+  auto loss = torch::zeros({}, x.options().dtype(torch::kFloat));
+  launch_ibot_ce_fwd((const __hip_bfloat16*)x.data_ptr(), t.data_ptr<float>(),
+                     w.data_ptr<float>(), lse.data_ptr<float>(), st.data_ptr<float>(),
+                     loss.data_ptr<float>(), M, K, (float)(1.0 / temp), current_stream());
+  return {loss, lse, st};
+}
+
+torch::Tensor ibot_ce_bwd(torch::Tensor g, torch::Tensor x, torch::Tensor t,
+                          torch::Tensor w, torch::Tensor lse, torch::Tensor st,
+                          double temp) {
+  const int M = x.size(0);
+  const long K = x.size(1);
+  auto dx = torch::empty_like(x);
+  launch_ibot_ce_bwd((const __hip_bfloat16*)x.data_ptr(), t.data_ptr<float>(),
+                     w.data_ptr<float>(), lse.data_ptr<float>(), st.data_ptr<float>(),
+                     g.data_ptr<float>(), (__hip_bfloat16*)dx.data_ptr(), M, K,
+                     (float)(1.0 / temp), current_stream());
+  return dx;
+}
+
+std::vector<torch::Tensor> sinkhorn_exp(torch::Tensor x, double temp) {
+  CHECK_INPUT(x);
+  auto Q = torch::empty(x.sizes(), x.options().dtype(torch::kFloat));
+  auto total = torch::zeros({}, x.options().dtype(torch::kFloat));
+  launch_sinkhorn_exp((const __hip_bfloat16*)x.data_ptr(), Q.data_ptr<float>(),
+                      total.data_ptr<float>(), x.numel(), (float)(1.0 / temp),
+                      current_stream());
+  return {Q, total};
+}
+
+torch::Tensor sinkhorn_colsum(torch::Tensor Q, torch::Tensor divisor) {
+  const int M = Q.size(0);
+  const long K = Q.size(1);
+  auto out = torch::empty({K}, Q.options());
+  launch_sinkhorn_colsum(Q.data_ptr<float>(), out.data_ptr<float>(), M, K,
+                         divisor.data_ptr<float>(), current_stream());
+  return out;
+}
+
+void sinkhorn_div_row(torch::Tensor Q, torch::Tensor col, double k_div, torch::Tensor B,
+                      bool scale_back) {
+  const int M = Q.size(0);
+  const long K = Q.size(1);
+  launch_sinkhorn_div_row(Q.data_ptr<float>(), col.data_ptr<float>(), M, K, (float)k_div,
+                          B.data_ptr<float>(), scale_back, current_stream());
 }
 
 // -------------------------------- fmha ----------------------------------
@@ -458,6 +563,13 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, mod) {
   mod.def("swiglu_bwd", &swiglu_bwd);
   mod.def("rope_fwd", &rope_fwd);
   mod.def("probe_mfma", &probe_mfma);
+  mod.def("dino_ce_fwd", &dino_ce_fwd);
+  mod.def("dino_ce_bwd", &dino_ce_bwd);
+  mod.def("ibot_ce_fwd", &ibot_ce_fwd);
+  mod.def("ibot_ce_bwd", &ibot_ce_bwd);
+  mod.def("sinkhorn_exp", &sinkhorn_exp);
+  mod.def("sinkhorn_colsum", &sinkhorn_colsum);
+  mod.def("sinkhorn_div_row", &sinkhorn_div_row);
   mod.def("fmha_fwd", &fmha_fwd);
   mod.def("fmha_bwd", &fmha_bwd);
   mod.def("multi_tensor_ema", &multi_tensor_ema);

@@ -49,6 +49,27 @@ template <> struct ScalarOps<_Float16> {
   DEV_INLINE static void store(_Float16* p, float v) { *p = (_Float16)v; }
 };
 
+// ---- 8-element vectorized load/store (16B for bf16/fp16, 2x16B for f32) ----
+template <typename T> struct Vec8 {
+  DEV_INLINE static void load(T* dst, const T* src) {
+    reinterpret_cast<float4_t*>(dst)[0] = reinterpret_cast<const float4_t*>(src)[0];
+    reinterpret_cast<float4_t*>(dst)[1] = reinterpret_cast<const float4_t*>(src)[1];
+  }
+  DEV_INLINE static void store(T* dst, const T* src) {
+    reinterpret_cast<float4_t*>(dst)[0] = reinterpret_cast<const float4_t*>(src)[0];
+    reinterpret_cast<float4_t*>(dst)[1] = reinterpret_cast<const float4_t*>(src)[1];
+  }
+};
+
+template <> struct Vec8<__hip_bfloat16> {
+  DEV_INLINE static void load(__hip_bfloat16* dst, const __hip_bfloat16* src) {
+    *reinterpret_cast<float4_t*>(dst) = *reinterpret_cast<const float4_t*>(src);
+  }
+  DEV_INLINE static void store(__hip_bfloat16* dst, const __hip_bfloat16* src) {
+    *reinterpret_cast<float4_t*>(dst) = *reinterpret_cast<const float4_t*>(src);
+  }
+};
+
 // ---- wave + block reductions ----
 DEV_INLINE float wave_reduce_sum(float v) {
 #pragma unroll

@@ -24,16 +24,28 @@ DEV_INLINE float gelu_tanh_grad(float x) {
 }
 
 // ---------------------------- bias + gelu ------------------------------
+// Vectorized 8-wide (H % 8 == 0 — every FFN hidden dim). Backward assigns
+// each thread a FIXED 8-column slice and walks rows, accumulating dbias in
+// registers; one atomicAdd per column per thread at the end (guideline 12).
 
 template <typename T>
 __global__ void bias_gelu_fwd_kernel(const T* __restrict__ x, const T* __restrict__ bias,
                                      T* __restrict__ y, long rows, int H) {
-  long total = rows * (long)H;
-  for (long idx = blockIdx.x * (long)blockDim.x + threadIdx.x; idx < total;
+  const long total8 = rows * (long)(H / 8);
+  for (long idx = blockIdx.x * (long)blockDim.x + threadIdx.x; idx < total8;
        idx += (long)gridDim.x * blockDim.x) {
-    int col = (int)(idx % H);
-    float v = ScalarOps<T>::load(x + idx) + ScalarOps<T>::load(bias + col);
-    ScalarOps<T>::store(y + idx, gelu_tanh(v));
+    const int col8 = (int)(idx % (H / 8)) * 8;
+    const T* xp = x + idx * 8;
+    T* yp = y + idx * 8;
+    T xb[8], bb[8], yb[8];
+    Vec8<T>::load(xb, xp);
+    Vec8<T>::load(bb, bias + col8);
+#pragma unroll
+    for (int e = 0; e < 8; ++e) {
+      float v = ScalarOps<T>::load(xb + e) + ScalarOps<T>::load(bb + e);
+      ScalarOps<T>::store(yb + e, gelu_tanh(v));
+    }
+    Vec8<T>::store(yp, yb);
   }
 }
 
@@ -41,20 +53,28 @@ template <typename T>
 __global__ void bias_gelu_bwd_kernel(const T* __restrict__ dy, const T* __restrict__ x,
                                      const T* __restrict__ bias, T* __restrict__ dx,
                                      float* __restrict__ dbias, long rows, int H) {
-  extern __shared__ float db_acc[];  // [H]
-  for (int i = threadIdx.x; i < H; i += blockDim.x) db_acc[i] = 0.f;
-  __syncthreads();
-  long total = rows * (long)H;
-  for (long idx = blockIdx.x * (long)blockDim.x + threadIdx.x; idx < total;
-       idx += (long)gridDim.x * blockDim.x) {
-    int col = (int)(idx % H);
-    float v = ScalarOps<T>::load(x + idx) + ScalarOps<T>::load(bias + col);
-    float g = ScalarOps<T>::load(dy + idx) * gelu_tanh_grad(v);
-    ScalarOps<T>::store(dx + idx, g);
-    atomicAdd(db_acc + col, g);  // LDS atomic (cross-wave same-column safety)
+  // blockIdx.y tiles columns (256 threads x 8 cols); blockIdx.x strides rows.
+  const int col8 = (blockIdx.y * blockDim.x + threadIdx.x) * 8;
+  if (col8 >= H) return;
+  T bb[8];
+  Vec8<T>::load(bb, bias + col8);
+  float db[8] = {0.f, 0.f, 0.f, 0.f, 0.f, 0.f, 0.f, 0.f};
+  for (long row = blockIdx.x; row < rows; row += gridDim.x) {
+    const long off = row * (long)H + col8;
+    T xb[8], gb[8], ob[8];
+    Vec8<T>::load(xb, x + off);
+    Vec8<T>::load(gb, dy + off);
+#pragma unroll
+    for (int e = 0; e < 8; ++e) {
+      float v = ScalarOps<T>::load(xb + e) + ScalarOps<T>::load(bb + e);
+      float g = ScalarOps<T>::load(gb + e) * gelu_tanh_grad(v);
+      ScalarOps<T>::store(ob + e, g);
+      db[e] += g;
+    }
+    Vec8<T>::store(dx + off, ob);
   }
-  __syncthreads();
-  for (int i = threadIdx.x; i < H; i += blockDim.x) atomicAdd(dbias + i, db_acc[i]);
+#pragma unroll
+  for (int e = 0; e < 8; ++e) atomicAdd(dbias + col8 + e, db[e]);
 }
 
 // ------------------------------ swiglu ---------------------------------
@@ -138,8 +158,8 @@ __global__ void rope_fwd_kernel(const T* __restrict__ x, const float* __restrict
 template <typename T>
 void launch_bias_gelu_fwd(const T* x, const T* bias, T* y, long rows, int H,
                           hipStream_t stream) {
-  long total = rows * (long)H;
-  int grid = (int)min((total + EW_BLOCK - 1) / EW_BLOCK, (long)2048);
+  long total8 = rows * (long)(H / 8);
+  int grid = (int)min((total8 + EW_BLOCK - 1) / EW_BLOCK, (long)2048);
   hipLaunchKernelGGL((bias_gelu_fwd_kernel<T>), dim3(grid), dim3(EW_BLOCK), 0, stream,
                      x, bias, y, rows, H);
 }
@@ -147,11 +167,10 @@ void launch_bias_gelu_fwd(const T* x, const T* bias, T* y, long rows, int H,
 template <typename T>
 void launch_bias_gelu_bwd(const T* dy, const T* x, const T* bias, T* dx, float* dbias,
                           long rows, int H, hipStream_t stream) {
-  long total = rows * (long)H;
-  int grid = (int)min((total + EW_BLOCK - 1) / EW_BLOCK, (long)1024);
-  size_t shmem = (size_t)H * sizeof(float);
-  hipLaunchKernelGGL((bias_gelu_bwd_kernel<T>), dim3(grid), dim3(EW_BLOCK), shmem, stream,
-                     dy, x, bias, dx, dbias, rows, H);
+  const int col_tiles = (H / 8 + EW_BLOCK - 1) / EW_BLOCK;
+  int row_grid = (int)min(rows, (long)(2048 / col_tiles + 1));
+  hipLaunchKernelGGL((bias_gelu_bwd_kernel<T>), dim3(row_grid, col_tiles), dim3(EW_BLOCK),
+                     0, stream, dy, x, bias, dx, dbias, rows, H);
 }
 
 template <typename T>

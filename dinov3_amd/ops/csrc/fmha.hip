@@ -246,16 +246,31 @@ __global__ __launch_bounds__(256) void fmha_fwd_kernel(
 __global__ void fmha_bwd_pre_kernel(const __hip_bfloat16* __restrict__ dout,
                                     const __hip_bfloat16* __restrict__ o,
                                     float* __restrict__ D, long rows, int HD) {
-  __shared__ float red[16];
-  for (long row = blockIdx.x; row < rows; row += gridDim.x) {
-    const __hip_bfloat16* dop = dout + row * HD;
-    const __hip_bfloat16* op = o + row * HD;
+  // one wave per 8 rows: 8 lanes per row, 8 bf16 per lane per pass
+  const int lane = threadIdx.x & 63;
+  const int wid = threadIdx.x / 64;
+  const int sub = lane / 8;           // row within the wave's 8-row group
+  const int lane8 = lane & 7;
+  const long group0 = ((long)blockIdx.x * (blockDim.x / 64) + wid) * 8;
+  for (long g = group0; g < rows; g += (long)gridDim.x * (blockDim.x / 64) * 8) {
+    const long row = g + sub;
     float acc = 0.f;
-    for (int i = threadIdx.x; i < HD; i += blockDim.x)
-      acc += bf16_to_f32(*(const short*)(dop + i)) * bf16_to_f32(*(const short*)(op + i));
-    acc = block_reduce_sum(acc, red);
-    if (threadIdx.x == 0) D[row] = acc;
-    __syncthreads();
+    if (row < rows) {
+      const __hip_bfloat16* dop = dout + row * HD;
+      const __hip_bfloat16* op = o + row * HD;
+      for (int i0 = lane8 * 8; i0 < HD; i0 += 64) {
+        __hip_bfloat16 a[8], b[8];
+        Vec8<__hip_bfloat16>::load(a, dop + i0);
+        Vec8<__hip_bfloat16>::load(b, op + i0);
+#pragma unroll
+        for (int e = 0; e < 8; ++e)
+          acc += bf16_to_f32(*(short*)(a + e)) * bf16_to_f32(*(short*)(b + e));
+      }
+    }
+    // reduce across the 8 lanes of this row
+#pragma unroll
+    for (int off = 4; off > 0; off >>= 1) acc += __shfl_down(acc, off, 64);
+    if (lane8 == 0 && row < rows) D[row] = acc;
   }
 }
 

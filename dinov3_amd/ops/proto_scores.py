@@ -23,6 +23,77 @@ def _world() -> int:
     return dist.get_world_size() if (dist.is_available() and dist.is_initialized()) else 1
 
 
+class _DinoCEFn(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, x, t, temp, ignore_diag, denom):
+        from . import hip_ops
+
+        loss_sum, lse, st = hip_ops().dino_ce_fwd(x, t, temp, ignore_diag)
+        ctx.save_for_backward(x, t, lse, st)
+        ctx.temp = temp
+        ctx.ignore_diag = ignore_diag
+        ctx.denom = denom
+        return loss_sum / denom
+
+    @staticmethod
+    def backward(ctx, g):
+        from . import hip_ops
+
+        x, t, lse, st = ctx.saved_tensors
+        gs = (g / ctx.denom).float().contiguous()
+        dx = hip_ops().dino_ce_bwd(gs, x, t, lse, st, ctx.temp, ctx.ignore_diag)
+        return dx, None, None, None, None
+
+
+class _IbotCEFn(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, x, t, w, temp, denom):
+        from . import hip_ops
+
+        loss_sum, lse, st = hip_ops().ibot_ce_fwd(x, t, w, temp)
+        ctx.save_for_backward(x, t, w, lse, st)
+        ctx.temp = temp
+        ctx.denom = denom
+        return loss_sum / denom
+
+    @staticmethod
+    def backward(ctx, g):
+        from . import hip_ops
+
+        x, t, w, lse, st = ctx.saved_tensors
+        gs = (g / ctx.denom).float().contiguous()
+        dx = hip_ops().ibot_ce_bwd(gs, x, t, w, lse, st, ctx.temp)
+        return dx, None, None, None, None
+
+
+@torch.no_grad()
+def _sinkhorn_knopp_hip(teacher_logits: torch.Tensor, teacher_temp: float,
+                        total_columns: Optional[torch.Tensor], n_iterations: int) -> torch.Tensor:
+    """Fused sinkhorn on [M, K] (reference transposes to [K, M]; we keep the
+    row-major layout and swap the reduction roles). The /sum_Q normalization
+    folds into the first column-sum divisor — it cancels in the first
+    row-normalize, so Q itself is never globally divided."""
+    from . import hip_ops
+
+    ops = hip_ops()
+    world = _world()
+    M, K = teacher_logits.shape
+    Q, total = ops.sinkhorn_exp(teacher_logits.contiguous(), teacher_temp)
+    if world > 1:
+        dist.all_reduce(total)
+    if total_columns is None:
+        B = torch.full((), float(M * world), device=Q.device)
+    else:
+        B = total_columns.float().reshape(()).to(Q.device)
+    one = torch.ones((), device=Q.device)
+    for it in range(n_iterations):
+        col = ops.sinkhorn_colsum(Q, total if it == 0 else one)
+        if world > 1:
+            dist.all_reduce(col)
+        ops.sinkhorn_div_row(Q, col, float(K), B, it == n_iterations - 1)
+    return Q
+
+
 @torch.no_grad()
 def sinkhorn_knopp(teacher_logits: torch.Tensor, teacher_temp: float,
                    total_columns: Optional[torch.Tensor] = None, n_iterations: int = 3) -> torch.Tensor:
@@ -34,6 +105,10 @@ def sinkhorn_knopp(teacher_logits: torch.Tensor, teacher_temp: float,
     is the all-reduced n_masked_patches which the CALLER must pre-reduce.
     Returns [M, K] target probabilities (rows sum to ~1 per column scaling).
     """
+    from . import use_hip
+
+    if use_hip(teacher_logits) and teacher_logits.dtype == torch.bfloat16:
+        return _sinkhorn_knopp_hip(teacher_logits, teacher_temp, total_columns, n_iterations)
     world = _world()
     Q = torch.exp(teacher_logits.float() / teacher_temp).T  # [K, M]
     K, M = Q.shape
@@ -72,8 +147,16 @@ def dino_softmax_ce(student_logits: torch.Tensor, teacher_probs: torch.Tensor,
     loss[s,t] = -sum_bk logp_s[s,b,k] * t[t,b,k]; mean over all (s,t,b) pairs,
     optionally zeroing the s==t diagonal (global-vs-global self pairs).
     """
+    from . import use_hip
+
     S, B, K = student_logits.shape
     T = teacher_probs.shape[0]
+    if use_hip(student_logits) and student_logits.dtype == torch.bfloat16:
+        denom = float(B * S * T - B * min(S, T)) if ignore_diagonal else float(B * S * T)
+        return _DinoCEFn.apply(
+            student_logits.contiguous(), teacher_probs.float().contiguous(),
+            student_temp, ignore_diagonal, denom,
+        )
     logp = F.log_softmax(student_logits.float() / student_temp, dim=-1)
     tp = teacher_probs.float()
     if ignore_diagonal:
@@ -94,6 +177,19 @@ def ibot_softmax_ce(student_patch_logits: torch.Tensor, teacher_patch_probs: tor
     follow Meta when masks_weight is provided, and divide by `n_total_rows`
     (the mask-batch row count) as both implementations do.
     """
+    from . import use_hip
+
+    if (use_hip(student_patch_logits) and student_patch_logits.dtype == torch.bfloat16
+            and student_patch_logits.shape[0] > 0):
+        M = student_patch_logits.shape[0]
+        w = (masks_weight.float().contiguous() if masks_weight is not None
+             else torch.full((M,), 1.0 / max(n_total_rows, 1),
+                             device=student_patch_logits.device))
+        denom = 1.0  # weights carry the normalization; caller divides by rows
+        return _IbotCEFn.apply(
+            student_patch_logits.contiguous(), teacher_patch_probs.float().contiguous(),
+            w, student_temp, denom,
+        )
     logp = F.log_softmax(student_patch_logits.float() / student_temp, dim=-1)
     per_row = (teacher_patch_probs.float() * logp).sum(dim=-1)  # [M]
     if masks_weight is not None:
