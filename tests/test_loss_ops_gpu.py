@@ -37,7 +37,8 @@ def test_dino_ce_fused_matches_reference(ignore_diag):
         ref = -torch.einsum("sbk,tbk->", logp, t) / (B * S * T)
     ref.backward()
     _close(loss, ref, 5e-3, "dino ce loss")
-    _close(x.grad, xr.grad, 2e-4, "dino ce dx")
+    # grads are returned in the input dtype (bf16): compare vs quantized ref
+    _close(x.grad, xr.grad.bfloat16(), 1e-4 + 0.01 * xr.grad.abs().max().item(), "dino ce dx")
 
 
 def test_ibot_ce_fused_matches_reference():
@@ -56,7 +57,7 @@ def test_ibot_ce_fused_matches_reference():
     ref = -((t * logp).sum(-1) * w).sum()
     ref.backward()
     _close(loss, ref, max(5e-3, 1e-4 * ref.abs().item()), "ibot ce loss")
-    _close(x.grad, xr.grad, 2e-3, "ibot ce dx")
+    _close(x.grad, xr.grad.bfloat16(), 1e-4 + 0.01 * xr.grad.abs().max().item(), "ibot ce dx")
 
 
 def test_sinkhorn_fused_matches_torch():
