@@ -130,10 +130,12 @@ def main():
         reducer.finalize()
         clip_scales = None
         if clip:
-            sums = optimizer.grad_norms_per_submodel()
+            sums = optimizer.grad_norm_sums()
             if world > 1:
-                sums = all_reduce_scalar_sums(sums)
-            clip_scales = {k: min(1.0, clip / (float(s) ** 0.5 + 1e-6)) for k, s in sums.items()}
+                import torch.distributed as dist
+
+                dist.all_reduce(sums)
+            clip_scales = optimizer.clip_factors(sums, clip)
         optimizer.step(lr=1e-4, weight_decay=0.04, last_layer_lr=0.0, clip_scales=clip_scales)
         optimizer.zero_grad()
         model.update_ema(0.992)

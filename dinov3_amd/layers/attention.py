@@ -1,8 +1,12 @@
-"""Self-attention: fused qkv GEMM -> RoPE apply -> FMHA -> out projection.
+"""Self-attention on the flat multi-crop token buffer.
 
-Reference: dinov3_jax/layers/attention.py:49-133. mask_k_bias (vit7b configs)
-means the k slice of the qkv bias is held at zero (the reference expresses
-this with a NaN-filled buffer times bias; SURVEY K4 intended semantics).
+qkv GEMM and out-projection run once on the concatenated [R, D] buffer of all
+crop groups; the fused rope+FMHA kernel consumes contiguous per-group slices
+of the qkv output directly (ops/flat_attention.py) — no permutes or concats
+on the hot path.
+
+Reference semantics: dinov3_jax/layers/attention.py:49-133. mask_k_bias
+(vit7b configs) keeps the k third of the qkv bias at zero.
 """
 
 from __future__ import annotations
@@ -13,7 +17,7 @@ import torch
 import torch.nn as nn
 import torch.nn.functional as F
 
-from ..ops import fmha, rope_apply
+from ..ops.flat_attention import GroupMeta, flat_multi_fmha
 from ..utils.utils import cat_keep_shapes, uncat_with_shapes
 
 RopeSinCos = Tuple[torch.Tensor, torch.Tensor]
@@ -56,35 +60,37 @@ class SelfAttention(nn.Module):
         self.proj = nn.Linear(dim, dim, bias=proj_bias)
         self.proj_drop = nn.Dropout(proj_drop) if proj_drop > 0 else nn.Identity()
 
-    def _attend(self, qkv: torch.Tensor, rope: Optional[RopeSinCos]) -> torch.Tensor:
-        """qkv: [B, N, 3*dim] -> context [B, N, dim]."""
-        B, N, _ = qkv.shape
-        qkv = qkv.reshape(B, N, 3, self.num_heads, self.head_dim)
-        q, k, v = qkv.permute(2, 0, 3, 1, 4).unbind(0)  # each [B, H, N, hd]
-        if rope is not None:
-            sin, cos = rope
-            prefix = N - sin.shape[0]
-            assert prefix >= 0
-            q = rope_apply(q, sin, cos, prefix)
-            k = rope_apply(k, sin, cos, prefix)
-        x = fmha(q.contiguous(), k.contiguous(), v.contiguous())  # [B, H, N, hd]
-        return x.permute(0, 2, 1, 3).reshape(B, N, self.dim)
+    def forward_flat(self, flat: torch.Tensor, metas: List[GroupMeta]) -> torch.Tensor:
+        """flat: [R, D] concatenated tokens of all crop groups."""
+        qkv_flat = self.qkv(flat)
+        ctx_flat = flat_multi_fmha(qkv_flat, self.num_heads, metas)
+        return self.proj_drop(self.proj(ctx_flat))
+
+    @staticmethod
+    def _meta_for(x: torch.Tensor, rope: Optional[RopeSinCos], offset: int) -> GroupMeta:
+        B, N, _ = x.shape
+        if rope is None:
+            return (offset, B, N, None, None, 0)
+        sin, cos = rope
+        prefix = N - sin.shape[0]
+        assert prefix >= 0
+        return (offset, B, N, sin.float().contiguous(), cos.float().contiguous(), prefix)
 
     def forward(self, x: torch.Tensor, rope: Optional[RopeSinCos] = None) -> torch.Tensor:
-        qkv = self.qkv(x)
-        ctx = self._attend(qkv, rope)
-        return self.proj_drop(self.proj(ctx))
+        B, N, D = x.shape
+        meta = self._meta_for(x, rope, 0)
+        out = self.forward_flat(x.reshape(B * N, D), [meta])
+        return out.reshape(B, N, D)
 
     def forward_list(self, x_list: List[torch.Tensor],
                      rope_list: Optional[List[Optional[RopeSinCos]]] = None) -> List[torch.Tensor]:
-        """Multi-crop forward: qkv + out-proj GEMMs run on the concatenated
-        token batch; FMHA runs per crop group (different seqlens)."""
         if rope_list is None:
             rope_list = [None] * len(x_list)
         flat, shapes, counts = cat_keep_shapes(x_list)
-        qkv_flat = self.qkv(flat)
-        qkv_list = uncat_with_shapes(qkv_flat, shapes, counts)
-        ctx_list = [self._attend(qkv, rope) for qkv, rope in zip(qkv_list, rope_list)]
-        ctx_flat, shapes2, counts2 = cat_keep_shapes(ctx_list)
-        out_flat = self.proj_drop(self.proj(ctx_flat))
-        return uncat_with_shapes(out_flat, shapes2, counts2)
+        metas = []
+        off = 0
+        for x, rope in zip(x_list, rope_list):
+            metas.append(self._meta_for(x, rope, off))
+            off += x.shape[0] * x.shape[1]
+        out_flat = self.forward_flat(flat, metas)
+        return uncat_with_shapes(out_flat, shapes, counts)

@@ -1,28 +1,42 @@
-"""Pre-norm transformer block with LayerScale and stochastic depth as
-batch-subset compute.
+"""Pre-norm transformer block on the flat multi-crop token buffer, with
+LayerScale and stochastic depth as batch-subset compute.
 
-Reference: dinov3_jax/layers/block.py:22-208. Stochastic depth runs
-attn/ffn only on a random ~b*(1-p) row subset and scatter-adds the rescaled
-residual — a throughput feature we keep. Norms and FFN run on the
-concatenated multi-crop token batch (cat_keep_shapes), attention per crop
-group.
+The whole residual stream lives in one flat [R, D] tensor; norms, qkv/proj
+GEMMs and the FFN run on it directly. Stochastic depth gathers the kept
+samples' rows (per crop group), computes on the smaller flat buffer, and
+index_add's the rescaled residual back — the reference's throughput feature
+(dinov3_jax/layers/block.py:88-194) without any per-block concat/split.
 """
 
 from __future__ import annotations
 
-from typing import List, Optional
+from typing import List, Optional, Tuple
 
 import torch
 import torch.nn as nn
 
+from ..ops.flat_attention import GroupMeta
 from ..utils.utils import cat_keep_shapes, uncat_with_shapes
 from .attention import RopeSinCos, SelfAttention
 from .ffn_layers import Mlp
 from .norms import LayerScale
 
 
-def _subset_indices(b: int, keep: int, device) -> torch.Tensor:
-    return torch.randperm(b, device=device)[:keep]
+def _subset_rows(metas: List[GroupMeta], keep_ratio: float, device) -> Tuple[torch.Tensor, List[GroupMeta], torch.Tensor]:
+    """Random per-group sample subsets -> (row indices, new metas, per-row scale)."""
+    rows = []
+    new_metas: List[GroupMeta] = []
+    scales = []
+    new_off = 0
+    for (off, B, N, sin, cos, prefix) in metas:
+        keep = max(int(B * keep_ratio), 1)
+        idx = torch.randperm(B, device=device)[:keep]
+        r = (off + idx.unsqueeze(1) * N + torch.arange(N, device=device).unsqueeze(0)).reshape(-1)
+        rows.append(r)
+        new_metas.append((new_off, keep, N, sin, cos, prefix))
+        scales.append(torch.full((keep * N,), B / keep, device=device))
+        new_off += keep * N
+    return torch.cat(rows), new_metas, torch.cat(scales)
 
 
 class SelfAttentionBlock(nn.Module):
@@ -58,15 +72,25 @@ class SelfAttentionBlock(nn.Module):
         self.ls2 = LayerScale(dim, init_values) if init_values is not None else nn.Identity()
         self.sample_drop_ratio = drop_path
 
-    @staticmethod
-    def _index_rope(rope: Optional[RopeSinCos], indices: torch.Tensor) -> Optional[RopeSinCos]:
-        if rope is None:
-            return None
-        sin, cos = rope
-        if sin.ndim == 4:  # per-sample tables (coord augments per sample) — not used currently
-            return sin[indices], cos[indices]
-        return rope  # shared [P, hd] table: subset along batch leaves it unchanged
+    # ------------------------------------------------------------------
+    def forward_flat(self, flat: torch.Tensor, metas: List[GroupMeta]) -> torch.Tensor:
+        if not (self.training and self.sample_drop_ratio > 0.0):
+            flat = flat + self.ls1(self.attn.forward_flat(self.norm1(flat), metas))
+            flat = flat + self.ls2(self.mlp(self.norm2(flat)))
+            return flat
 
+        keep_ratio = 1.0 - self.sample_drop_ratio
+        rows1, metas1, scale1 = _subset_rows(metas, keep_ratio, flat.device)
+        sub = flat.index_select(0, rows1)
+        res = self.ls1(self.attn.forward_flat(self.norm1(sub), metas1))
+        flat = flat.index_add(0, rows1, (res * scale1.unsqueeze(1)).to(flat.dtype))
+
+        rows2, _, scale2 = _subset_rows(metas, keep_ratio, flat.device)
+        sub = flat.index_select(0, rows2)
+        res = self.ls2(self.mlp(self.norm2(sub)))
+        return flat.index_add(0, rows2, (res * scale2.unsqueeze(1)).to(flat.dtype))
+
+    # ------------------------------------------------------------------
     def forward(self, x: torch.Tensor, rope: Optional[RopeSinCos] = None) -> torch.Tensor:
         return self.forward_list([x], [rope])[0]
 
@@ -74,39 +98,11 @@ class SelfAttentionBlock(nn.Module):
                      rope_list: Optional[List[Optional[RopeSinCos]]] = None) -> List[torch.Tensor]:
         if rope_list is None:
             rope_list = [None] * len(x_list)
-        use_droppath = self.training and self.sample_drop_ratio > 0.0
-        if not use_droppath:
-            flat, shapes, counts = cat_keep_shapes(x_list)
-            n1 = uncat_with_shapes(self.norm1(flat), shapes, counts)
-            attn_res = self.attn.forward_list(n1, rope_list)
-            x_list = [x + self.ls1(r) for x, r in zip(x_list, attn_res)]
-            flat, shapes, counts = cat_keep_shapes(x_list)
-            mlp_res = self.mlp(self.norm2(flat))
-            mlp_res = self.ls2(mlp_res)
-            return [x + r for x, r in zip(x_list, uncat_with_shapes(mlp_res, shapes, counts))]
-
-        # stochastic depth: subset-compute with scatter-add rescaled residual
-        b_list = [x.shape[0] for x in x_list]
-        keeps = [max(int(b * (1.0 - self.sample_drop_ratio)), 1) for b in b_list]
-        scales = [b / k for b, k in zip(b_list, keeps)]
-
-        idx1 = [_subset_indices(b, k, x.device) for b, k, x in zip(b_list, keeps, x_list)]
-        sub1 = [x[i] for x, i in zip(x_list, idx1)]
-        rope_sub = [self._index_rope(r, i) for r, i in zip(rope_list, idx1)]
-        flat, shapes, counts = cat_keep_shapes(sub1)
-        n1 = uncat_with_shapes(self.norm1(flat), shapes, counts)
-        attn_res = self.attn.forward_list(n1, rope_sub)
-        x_list = [
-            x.index_add(0, i, (self.ls1(r) * s).to(x.dtype))
-            for x, i, r, s in zip(x_list, idx1, attn_res, scales)
-        ]
-
-        idx2 = [_subset_indices(b, k, x.device) for b, k, x in zip(b_list, keeps, x_list)]
-        sub2 = [x[i] for x, i in zip(x_list, idx2)]
-        flat, shapes, counts = cat_keep_shapes(sub2)
-        mlp_res = self.ls2(self.mlp(self.norm2(flat)))
-        mlp_list = uncat_with_shapes(mlp_res, shapes, counts)
-        return [
-            x.index_add(0, i, (r * s).to(x.dtype))
-            for x, i, r, s in zip(x_list, idx2, mlp_list, scales)
-        ]
+        flat, shapes, counts = cat_keep_shapes(x_list)
+        metas = []
+        off = 0
+        for x, rope in zip(x_list, rope_list):
+            metas.append(SelfAttention._meta_for(x, rope, off))
+            off += x.shape[0] * x.shape[1]
+        out = self.forward_flat(flat, metas)
+        return uncat_with_shapes(out, shapes, counts)

@@ -166,15 +166,26 @@ class DinoVisionTransformer(nn.Module):
         rope_tables = {}
         for (H, W) in hw:
             if (H, W) not in rope_tables:
-                # tables stay fp32: the HIP rope kernel applies them in fp32
-                # and writes back bf16 (reference round-trips through bf16
-                # tables; fp32 here is a strict accuracy improvement)
+                # tables stay fp32: the HIP rope+fmha kernel applies them in
+                # fp32 and writes back bf16 (reference round-trips through
+                # bf16 tables; fp32 here is a strict accuracy improvement)
                 sin, cos = self.rope_embed(H=H, W=W, training=self.training, device=tokens[0].device)
-                rope_tables[(H, W)] = (sin, cos)
-        rope_list = [rope_tables[k] for k in hw]
+                rope_tables[(H, W)] = (sin.contiguous(), cos.contiguous())
+
+        # flatten all crop groups into ONE [R, D] buffer for the whole depth
+        from ..layers.attention import SelfAttention
+        from ..utils.utils import cat_keep_shapes, uncat_with_shapes
+
+        flat, shapes, counts = cat_keep_shapes(tokens)
+        metas = []
+        off = 0
+        for t, key in zip(tokens, hw):
+            metas.append(SelfAttention._meta_for(t, rope_tables[key], off))
+            off += t.shape[0] * t.shape[1]
 
         for block in self.blocks:
-            tokens = block.forward_list(tokens, rope_list)
+            flat = block.forward_flat(flat, metas)
+        tokens = uncat_with_shapes(flat, shapes, counts)
 
         output = []
         for idx, (x, masks) in enumerate(zip(tokens, masks_list)):

@@ -52,6 +52,17 @@ void launch_sinkhorn_exp(const __hip_bfloat16*, float*, float*, long, float, hip
 void launch_sinkhorn_colsum(const float*, float*, int, long, const float*, hipStream_t);
 void launch_sinkhorn_div_row(float*, const float*, int, long, float, const float*, bool,
                              hipStream_t);
+void launch_fmha_rope_fwd(const __hip_bfloat16*, const float*, const float*,
+                          __hip_bfloat16*, float*, int, int, int, int, int, float,
+                          hipStream_t);
+void launch_fmha_rope_bwd_pre(const __hip_bfloat16*, const __hip_bfloat16*, float*, int,
+                              int, int, int, hipStream_t);
+void launch_fmha_rope_bwd_dq(const __hip_bfloat16*, const __hip_bfloat16*, const float*,
+                             const float*, const float*, const float*, __hip_bfloat16*,
+                             int, int, int, int, int, float, hipStream_t);
+void launch_fmha_rope_bwd_dkv(const __hip_bfloat16*, const __hip_bfloat16*, const float*,
+                              const float*, const float*, const float*, __hip_bfloat16*,
+                              int, int, int, int, int, float, hipStream_t);
 void launch_probe_mfma(const __hip_bfloat16*, const __hip_bfloat16*, float*, hipStream_t);
 void launch_fmha_fwd(const __hip_bfloat16*, const __hip_bfloat16*, const __hip_bfloat16*,
                      __hip_bfloat16*, float*, int, int, int, float, hipStream_t);
@@ -75,6 +86,17 @@ void launch_multi_tensor_adamw(T* const*, const T* const*, float* const*, float*
 template <typename T>
 void launch_multi_tensor_l2norm_sq(const T* const*, const long*, const int*, const long*,
                                    int, float*, hipStream_t);
+template <typename T>
+void launch_multi_tensor_adamw_planned(T* const*, const T* const*, float* const*,
+                                       float* const*, float* const*, const long*,
+                                       const int*, const long*, int, const float*,
+                                       const float*, const float*, const int*,
+                                       const float*, float, float, float, float, float,
+                                       float, float, float, bool, hipStream_t);
+template <typename T>
+void launch_multi_tensor_l2norm_planned(const T* const*, const long*, const int*,
+                                        const long*, int, const int*, float*,
+                                        hipStream_t);
 
 namespace {
 
@@ -413,6 +435,140 @@ std::vector<torch::Tensor> fmha_bwd(torch::Tensor dout, torch::Tensor q, torch::
   return {dq, dk, dv};
 }
 
+// ----------------------------- fmha + rope ------------------------------
+
+std::vector<torch::Tensor> fmha_rope_fwd_out(torch::Tensor qkv, torch::Tensor sin_t,
+                                             torch::Tensor cos_t, int64_t prefix,
+                                             torch::Tensor out_o) {
+  CHECK_INPUT(qkv);
+  TORCH_CHECK(qkv.dim() == 5 && qkv.size(2) == 3, "qkv must be [B, N, 3, H, hd]");
+  TORCH_CHECK(qkv.scalar_type() == at::ScalarType::BFloat16);
+  const int B = qkv.size(0), N = qkv.size(1), H = qkv.size(3), HD = qkv.size(4);
+  TORCH_CHECK(HD == 64 || HD == 128, "head_dim must be 64 or 128");
+  const bool has_rope = sin_t.defined() && sin_t.numel() > 0;
+  const int P = has_rope ? (int)sin_t.size(0) : N - (int)prefix;
+  if (has_rope) {
+    TORCH_CHECK(sin_t.scalar_type() == at::ScalarType::Float && sin_t.is_contiguous());
+    TORCH_CHECK(P == N - prefix, "rope table rows must equal N - prefix");
+  }
+  torch::Tensor o = out_o.defined() && out_o.numel() > 0 ? out_o
+                                                        : torch::empty({B, N, H, HD}, qkv.options());
+  TORCH_CHECK(o.is_contiguous() && o.numel() == (long)B * N * H * HD);
+  auto lse = torch::empty({B, H, N}, qkv.options().dtype(torch::kFloat));
+  const float scale = 1.0f / std::sqrt((float)HD);
+  launch_fmha_rope_fwd((const __hip_bfloat16*)qkv.data_ptr(),
+                       has_rope ? sin_t.data_ptr<float>() : nullptr,
+                       has_rope ? cos_t.data_ptr<float>() : nullptr,
+                       (__hip_bfloat16*)o.data_ptr(), lse.data_ptr<float>(), B, H, N, P,
+                       HD, scale, current_stream());
+  return {o, lse};
+}
+
+torch::Tensor fmha_rope_bwd_out(torch::Tensor dout, torch::Tensor qkv, torch::Tensor o,
+                                torch::Tensor lse, torch::Tensor sin_t,
+                                torch::Tensor cos_t, int64_t prefix,
+                                torch::Tensor out_dqkv) {
+  CHECK_INPUT(dout);
+  const int B = qkv.size(0), N = qkv.size(1), H = qkv.size(3), HD = qkv.size(4);
+  const bool has_rope = sin_t.defined() && sin_t.numel() > 0;
+  const int P = N - (int)prefix;
+  const float scale = 1.0f / std::sqrt((float)HD);
+  auto D = torch::empty({B, H, N}, qkv.options().dtype(torch::kFloat));
+  torch::Tensor dqkv = out_dqkv.defined() && out_dqkv.numel() > 0 ? out_dqkv
+                                                                  : torch::empty_like(qkv);
+  auto stream = current_stream();
+  launch_fmha_rope_bwd_pre((const __hip_bfloat16*)dout.data_ptr(),
+                           (const __hip_bfloat16*)o.data_ptr(), D.data_ptr<float>(), B, H,
+                           N, HD, stream);
+  launch_fmha_rope_bwd_dq((const __hip_bfloat16*)qkv.data_ptr(),
+                          (const __hip_bfloat16*)dout.data_ptr(),
+                          has_rope ? sin_t.data_ptr<float>() : nullptr,
+                          has_rope ? cos_t.data_ptr<float>() : nullptr,
+                          lse.data_ptr<float>(), D.data_ptr<float>(),
+                          (__hip_bfloat16*)dqkv.data_ptr(), B, H, N, P, HD, scale, stream);
+  launch_fmha_rope_bwd_dkv((const __hip_bfloat16*)qkv.data_ptr(),
+                           (const __hip_bfloat16*)dout.data_ptr(),
+                           has_rope ? sin_t.data_ptr<float>() : nullptr,
+                           has_rope ? cos_t.data_ptr<float>() : nullptr,
+                           lse.data_ptr<float>(), D.data_ptr<float>(),
+                           (__hip_bfloat16*)dqkv.data_ptr(), B, H, N, P, HD, scale,
+                           stream);
+  return dqkv;
+}
+
+
+// planned single-launch variants: tables are prebuilt device tensors owned by
+// the Python-side MultiTensorPlan (no per-call table construction/upload).
+
+void multi_tensor_ema_planned(torch::Tensor ptrs, torch::Tensor sizes, torch::Tensor ct,
+                              torch::Tensor co, int64_t n_tensors, double m,
+                              bool is_bf16) {
+  const int n_chunks = (int)ct.numel();
+  long* pp = ptrs.data_ptr<long>();
+  if (is_bf16) {
+    launch_multi_tensor_ema<__hip_bfloat16>(
+        (__hip_bfloat16* const*)pp, (const __hip_bfloat16* const*)(pp + n_tensors),
+        sizes.data_ptr<long>(), ct.data_ptr<int>(), co.data_ptr<long>(), n_chunks,
+        (float)m, current_stream());
+  } else {
+    launch_multi_tensor_ema<float>(
+        (float* const*)pp, (const float* const*)(pp + n_tensors), sizes.data_ptr<long>(),
+        ct.data_ptr<int>(), co.data_ptr<long>(), n_chunks, (float)m, current_stream());
+  }
+}
+
+void multi_tensor_adamw_planned(torch::Tensor ptrs, torch::Tensor sizes, torch::Tensor ct,
+                                torch::Tensor co, int64_t n_tensors, torch::Tensor lr_mult,
+                                torch::Tensor wd_mult, torch::Tensor is_last,
+                                torch::Tensor sub_id, torch::Tensor clip, double lr,
+                                double last_lr, double wd, double beta1, double beta2,
+                                double eps, double bc1, double bc2, bool has_master,
+                                bool is_bf16) {
+  const int n_chunks = (int)ct.numel();
+  long* pp = ptrs.data_ptr<long>();
+  const long n = n_tensors;
+  if (is_bf16) {
+    launch_multi_tensor_adamw_planned<__hip_bfloat16>(
+        (__hip_bfloat16* const*)pp, (const __hip_bfloat16* const*)(pp + n),
+        (float* const*)(pp + 2 * n), (float* const*)(pp + 3 * n),
+        has_master ? (float* const*)(pp + 4 * n) : nullptr, sizes.data_ptr<long>(),
+        ct.data_ptr<int>(), co.data_ptr<long>(), n_chunks, lr_mult.data_ptr<float>(),
+        wd_mult.data_ptr<float>(), is_last.data_ptr<float>(), sub_id.data_ptr<int>(),
+        clip.data_ptr<float>(), (float)lr, (float)last_lr, (float)wd, (float)beta1,
+        (float)beta2, (float)eps, (float)bc1, (float)bc2, has_master, current_stream());
+  } else {
+    launch_multi_tensor_adamw_planned<float>(
+        (float* const*)pp, (const float* const*)(pp + n), (float* const*)(pp + 2 * n),
+        (float* const*)(pp + 3 * n), has_master ? (float* const*)(pp + 4 * n) : nullptr,
+        sizes.data_ptr<long>(), ct.data_ptr<int>(), co.data_ptr<long>(), n_chunks,
+        lr_mult.data_ptr<float>(), wd_mult.data_ptr<float>(), is_last.data_ptr<float>(),
+        sub_id.data_ptr<int>(), clip.data_ptr<float>(), (float)lr, (float)last_lr,
+        (float)wd, (float)beta1, (float)beta2, (float)eps, (float)bc1, (float)bc2,
+        has_master, current_stream());
+  }
+}
+
+torch::Tensor multi_tensor_l2norm_planned(torch::Tensor ptrs, torch::Tensor sizes,
+                                          torch::Tensor ct, torch::Tensor co,
+                                          torch::Tensor sub_id, int64_t n_submodels,
+                                          bool is_bf16) {
+  const int n_chunks = (int)ct.numel();
+  auto out = torch::zeros({n_submodels}, torch::dtype(torch::kFloat).device(ptrs.device()));
+  long* pp = ptrs.data_ptr<long>();
+  if (is_bf16) {
+    launch_multi_tensor_l2norm_planned<__hip_bfloat16>(
+        (const __hip_bfloat16* const*)pp, sizes.data_ptr<long>(), ct.data_ptr<int>(),
+        co.data_ptr<long>(), n_chunks, sub_id.data_ptr<int>(), out.data_ptr<float>(),
+        current_stream());
+  } else {
+    launch_multi_tensor_l2norm_planned<float>(
+        (const float* const*)pp, sizes.data_ptr<long>(), ct.data_ptr<int>(),
+        co.data_ptr<long>(), n_chunks, sub_id.data_ptr<int>(), out.data_ptr<float>(),
+        current_stream());
+  }
+  return out;
+}
+
 // ---------------------------- multi-tensor ------------------------------
 
 struct MTTables {
@@ -571,8 +727,22 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, mod) {
   mod.def("sinkhorn_colsum", &sinkhorn_colsum);
   mod.def("sinkhorn_div_row", &sinkhorn_div_row);
   mod.def("fmha_fwd", &fmha_fwd);
+  mod.def("fmha_rope_fwd", [](torch::Tensor qkv, torch::Tensor sin_t, torch::Tensor cos_t,
+                              int64_t prefix) {
+    return fmha_rope_fwd_out(qkv, sin_t, cos_t, prefix, torch::Tensor());
+  });
+  mod.def("fmha_rope_fwd_out", &fmha_rope_fwd_out);
+  mod.def("fmha_rope_bwd", [](torch::Tensor dout, torch::Tensor qkv, torch::Tensor o,
+                              torch::Tensor lse, torch::Tensor sin_t, torch::Tensor cos_t,
+                              int64_t prefix) {
+    return fmha_rope_bwd_out(dout, qkv, o, lse, sin_t, cos_t, prefix, torch::Tensor());
+  });
+  mod.def("fmha_rope_bwd_out", &fmha_rope_bwd_out);
   mod.def("fmha_bwd", &fmha_bwd);
   mod.def("multi_tensor_ema", &multi_tensor_ema);
+  mod.def("multi_tensor_ema_planned", &multi_tensor_ema_planned);
+  mod.def("multi_tensor_adamw_planned", &multi_tensor_adamw_planned);
+  mod.def("multi_tensor_l2norm_planned", &multi_tensor_l2norm_planned);
   mod.def("multi_tensor_adamw", &multi_tensor_adamw);
   mod.def("multi_tensor_l2norm_sq", &multi_tensor_l2norm_sq);
 }

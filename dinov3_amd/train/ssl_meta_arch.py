@@ -127,8 +127,18 @@ class SSLMetaArch(nn.Module):
 
     @torch.no_grad()
     def update_ema(self, momentum: float) -> None:
-        """teacher <- m*teacher + (1-m)*student, one fused multi-tensor kernel."""
+        """teacher <- m*teacher + (1-m)*student, one fused in-place kernel
+        launch over a cached device-side plan."""
         pairs = self._teacher_student_param_pairs()
+        if pairs and pairs[0][0].is_cuda:
+            from ..ops.mt_plan import MultiTensorPlan, ema_planned
+
+            if getattr(self, "_ema_plan", None) is None:
+                self._ema_plan = MultiTensorPlan(
+                    [[t for t, _ in pairs], [s for _, s in pairs]]
+                )
+            ema_planned(self._ema_plan, momentum)
+            return
         ema_update_([t for t, _ in pairs], [s.detach() for _, s in pairs], momentum)
 
     # ------------------------------------------------------------------

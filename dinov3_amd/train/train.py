@@ -253,15 +253,17 @@ def do_train(cfg, model: SSLMetaArch, resume: bool = True, max_iterations: int =
         loss.backward()
         reducer.finalize()
 
-        clip_scales: Optional[Dict[str, float]] = None
+        clip_scales = None
         if clip is not None and clip > 0:
-            sums = optimizer.grad_norms_per_submodel()
-            sums = all_reduce_scalar_sums(sums) if parallel.get_world_size() > 1 else sums
-            clip_scales = {}
-            for name, s in sums.items():
-                norm = float(s) ** 0.5
-                clip_scales[name] = min(1.0, clip / (norm + 1e-6))
-                loss_dict[f"grad_norm_{name}"] = norm
+            sums = optimizer.grad_norm_sums()  # [n_submodels], stays on device
+            if parallel.get_world_size() > 1:
+                import torch.distributed as dist
+
+                dist.all_reduce(sums)
+            clip_scales = optimizer.clip_factors(sums, clip)
+            if it % 10 == 0:  # avoid a host sync every step
+                for name, s in zip(optimizer.submodels, sums.tolist()):
+                    loss_dict[f"grad_norm_{name}"] = s ** 0.5
 
         optimizer.step(lr=lr, weight_decay=wd, last_layer_lr=last_layer_lr, clip_scales=clip_scales)
         optimizer.zero_grad()
