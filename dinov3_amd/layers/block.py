@@ -72,11 +72,19 @@ class SelfAttentionBlock(nn.Module):
         self.ls2 = LayerScale(dim, init_values) if init_values is not None else nn.Identity()
         self.sample_drop_ratio = drop_path
 
+    def _add_scaled(self, flat: torch.Tensor, res: torch.Tensor, ls) -> torch.Tensor:
+        """flat + LayerScale(res), fused on GPU when a gamma exists."""
+        from ..ops.ls_axpy import ls_axpy
+
+        if isinstance(ls, LayerScale):
+            return ls_axpy(flat, res, ls.gamma)
+        return flat + res
+
     # ------------------------------------------------------------------
     def forward_flat(self, flat: torch.Tensor, metas: List[GroupMeta]) -> torch.Tensor:
         if not (self.training and self.sample_drop_ratio > 0.0):
-            flat = flat + self.ls1(self.attn.forward_flat(self.norm1(flat), metas))
-            flat = flat + self.ls2(self.mlp(self.norm2(flat)))
+            flat = self._add_scaled(flat, self.attn.forward_flat(self.norm1(flat), metas), self.ls1)
+            flat = self._add_scaled(flat, self.mlp(self.norm2(flat)), self.ls2)
             return flat
 
         keep_ratio = 1.0 - self.sample_drop_ratio

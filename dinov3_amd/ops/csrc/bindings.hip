@@ -32,6 +32,10 @@ void launch_bias_gelu_fwd(const T*, const T*, T*, long, int, hipStream_t);
 template <typename T>
 void launch_bias_gelu_bwd(const T*, const T*, const T*, T*, float*, long, int, hipStream_t);
 template <typename T>
+void launch_ls_axpy_fwd(const T*, const T*, const T*, T*, long, int, hipStream_t);
+template <typename T>
+void launch_ls_axpy_bwd(const T*, const T*, const T*, T*, float*, long, int, hipStream_t);
+template <typename T>
 void launch_swiglu_fwd(const T*, T*, long, int, hipStream_t);
 template <typename T>
 void launch_swiglu_bwd(const T*, const T*, T*, long, int, hipStream_t);
@@ -240,6 +244,39 @@ std::vector<torch::Tensor> bias_gelu_bwd(torch::Tensor dy, torch::Tensor x,
                                    dbias.data_ptr<float>(), rows, H, current_stream());
   });
   return {dx, dbias.to(x.scalar_type())};
+}
+
+torch::Tensor ls_axpy_fwd(torch::Tensor x, torch::Tensor res, torch::Tensor gamma) {
+  CHECK_INPUT(x);
+  CHECK_INPUT(res);
+  const int D = x.size(-1);
+  TORCH_CHECK(D % 8 == 0, "ls_axpy: D must be a multiple of 8");
+  const long rows = x.numel() / D;
+  auto out = torch::empty_like(x);
+  DISPATCH_FLOAT_BF16(x.scalar_type(), "ls_axpy_fwd", [&] {
+    launch_ls_axpy_fwd<scalar_t>((const scalar_t*)x.data_ptr(),
+                                 (const scalar_t*)res.data_ptr(),
+                                 (const scalar_t*)gamma.data_ptr(),
+                                 (scalar_t*)out.data_ptr(), rows, D, current_stream());
+  });
+  return out;
+}
+
+std::vector<torch::Tensor> ls_axpy_bwd(torch::Tensor dout, torch::Tensor res,
+                                       torch::Tensor gamma) {
+  CHECK_INPUT(dout);
+  const int D = dout.size(-1);
+  const long rows = dout.numel() / D;
+  auto dres = torch::empty_like(dout);
+  auto dgamma = torch::zeros({D}, dout.options().dtype(torch::kFloat));
+  DISPATCH_FLOAT_BF16(dout.scalar_type(), "ls_axpy_bwd", [&] {
+    launch_ls_axpy_bwd<scalar_t>((const scalar_t*)dout.data_ptr(),
+                                 (const scalar_t*)res.data_ptr(),
+                                 (const scalar_t*)gamma.data_ptr(),
+                                 (scalar_t*)dres.data_ptr(), dgamma.data_ptr<float>(),
+                                 rows, D, current_stream());
+  });
+  return {dres, dgamma.to(dout.scalar_type())};
 }
 
 torch::Tensor swiglu_fwd(torch::Tensor x12) {
@@ -715,6 +752,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, mod) {
   mod.def("l2norm_bwd", &l2norm_bwd);
   mod.def("bias_gelu_fwd", &bias_gelu_fwd);
   mod.def("bias_gelu_bwd", &bias_gelu_bwd);
+  mod.def("ls_axpy_fwd", &ls_axpy_fwd);
+  mod.def("ls_axpy_bwd", &ls_axpy_bwd);
   mod.def("swiglu_fwd", &swiglu_fwd);
   mod.def("swiglu_bwd", &swiglu_bwd);
   mod.def("rope_fwd", &rope_fwd);

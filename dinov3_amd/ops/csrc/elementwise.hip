@@ -77,6 +77,59 @@ __global__ void bias_gelu_bwd_kernel(const T* __restrict__ dy, const T* __restri
   for (int e = 0; e < 8; ++e) atomicAdd(dbias + col8 + e, db[e]);
 }
 
+// -------------------- LayerScale + residual add (K10) -------------------
+// out = x + gamma * res ; bwd: dx = dout (aliased), dres = dout * gamma,
+// dgamma = colsum(dout * res). 8-wide; same fixed-column-slice register
+// accumulation as bias_gelu_bwd.
+
+template <typename T>
+__global__ void ls_axpy_fwd_kernel(const T* __restrict__ x, const T* __restrict__ res,
+                                   const T* __restrict__ gamma, T* __restrict__ out,
+                                   long rows, int D) {
+  const long total8 = rows * (long)(D / 8);
+  for (long idx = blockIdx.x * (long)blockDim.x + threadIdx.x; idx < total8;
+       idx += (long)gridDim.x * blockDim.x) {
+    const int col8 = (int)(idx % (D / 8)) * 8;
+    T xb[8], rb[8], gb[8], ob[8];
+    Vec8<T>::load(xb, x + idx * 8);
+    Vec8<T>::load(rb, res + idx * 8);
+    Vec8<T>::load(gb, gamma + col8);
+#pragma unroll
+    for (int e = 0; e < 8; ++e) {
+      float v = ScalarOps<T>::load(xb + e) +
+                ScalarOps<T>::load(gb + e) * ScalarOps<T>::load(rb + e);
+      ScalarOps<T>::store(ob + e, v);
+    }
+    Vec8<T>::store(out + idx * 8, ob);
+  }
+}
+
+template <typename T>
+__global__ void ls_axpy_bwd_kernel(const T* __restrict__ dout, const T* __restrict__ res,
+                                   const T* __restrict__ gamma, T* __restrict__ dres,
+                                   float* __restrict__ dgamma, long rows, int D) {
+  const int col8 = (blockIdx.y * blockDim.x + threadIdx.x) * 8;
+  if (col8 >= D) return;
+  T gb[8];
+  Vec8<T>::load(gb, gamma + col8);
+  float dg[8] = {0.f, 0.f, 0.f, 0.f, 0.f, 0.f, 0.f, 0.f};
+  for (long row = blockIdx.x; row < rows; row += gridDim.x) {
+    const long off = row * (long)D + col8;
+    T db[8], rb[8], ob[8];
+    Vec8<T>::load(db, dout + off);
+    Vec8<T>::load(rb, res + off);
+#pragma unroll
+    for (int e = 0; e < 8; ++e) {
+      const float g = ScalarOps<T>::load(db + e);
+      dg[e] += g * ScalarOps<T>::load(rb + e);
+      ScalarOps<T>::store(ob + e, g * ScalarOps<T>::load(gb + e));
+    }
+    Vec8<T>::store(dres + off, ob);
+  }
+#pragma unroll
+  for (int e = 0; e < 8; ++e) atomicAdd(dgamma + col8 + e, dg[e]);
+}
+
 // ------------------------------ swiglu ---------------------------------
 // x12 = [rows, 2H] as [x1 | x2]; y = silu(x1) * x2.
 
@@ -174,6 +227,24 @@ void launch_bias_gelu_bwd(const T* dy, const T* x, const T* bias, T* dx, float* 
 }
 
 template <typename T>
+void launch_ls_axpy_fwd(const T* x, const T* res, const T* gamma, T* out, long rows,
+                        int D, hipStream_t stream) {
+  long total8 = rows * (long)(D / 8);
+  int grid = (int)min((total8 + EW_BLOCK - 1) / EW_BLOCK, (long)2048);
+  hipLaunchKernelGGL((ls_axpy_fwd_kernel<T>), dim3(grid), dim3(EW_BLOCK), 0, stream, x,
+                     res, gamma, out, rows, D);
+}
+
+template <typename T>
+void launch_ls_axpy_bwd(const T* dout, const T* res, const T* gamma, T* dres,
+                        float* dgamma, long rows, int D, hipStream_t stream) {
+  const int col_tiles = (D / 8 + EW_BLOCK - 1) / EW_BLOCK;
+  int row_grid = (int)min(rows, (long)(2048 / col_tiles + 1));
+  hipLaunchKernelGGL((ls_axpy_bwd_kernel<T>), dim3(row_grid, col_tiles), dim3(EW_BLOCK),
+                     0, stream, dout, res, gamma, dres, dgamma, rows, D);
+}
+
+template <typename T>
 void launch_swiglu_fwd(const T* x12, T* y, long rows, int H, hipStream_t stream) {
   long total = rows * (long)H;
   int grid = (int)min((total + EW_BLOCK - 1) / EW_BLOCK, (long)2048);
@@ -202,6 +273,10 @@ void launch_rope_fwd(const T* x, const float* sin_t, const float* cos_t, T* y, l
 #define INSTANTIATE_EW(T)                                                            \
   template void launch_bias_gelu_fwd<T>(const T*, const T*, T*, long, int,           \
                                         hipStream_t);                                \
+  template void launch_ls_axpy_fwd<T>(const T*, const T*, const T*, T*, long, int,   \
+                                      hipStream_t);                                  \
+  template void launch_ls_axpy_bwd<T>(const T*, const T*, const T*, T*, float*,      \
+                                      long, int, hipStream_t);                       \
   template void launch_bias_gelu_bwd<T>(const T*, const T*, const T*, T*, float*,    \
                                         long, int, hipStream_t);                     \
   template void launch_swiglu_fwd<T>(const T*, T*, long, int, hipStream_t);          \

@@ -1,32 +1,64 @@
 // Row-wise normalization kernels: LayerNorm / RMSNorm / L2-norm, fwd + bwd.
 // (SURVEY K2/K3 + the DINO-head bottleneck normalize of K15.)
 //
-// Shape model: x is [rows, D] contiguous, bf16 or fp32 I/O, fp32 math.
-// One 256-thread block per row (grid-stride over rows); vectorized 8-wide
-// bf16 loads on the fast path (guideline 13). Weight grads accumulate into a
-// per-block LDS column buffer and leave via one atomicAdd per column per
-// block (guideline 12).
+// x is [rows, D] contiguous, bf16 or fp32 I/O, fp32 math. One 256-thread
+// block per row (grid-stride over rows). V8=true takes the 8-wide vector
+// path (16 B/lane bf16 loads — guideline 13); weight grads accumulate in
+// registers per fixed column slice in LDS and leave via one atomicAdd per
+// column per block.
 
 #include "common.h"
 
 #define NORM_BLOCK 256
 
+template <typename T, bool V8>
+DEV_INLINE void row_load8(const T* p, int i, float* out) {
+  if constexpr (V8) {
+    T buf[8];
+    Vec8<T>::load(buf, p + i);
+#pragma unroll
+    for (int e = 0; e < 8; ++e) out[e] = ScalarOps<T>::load(buf + e);
+  } else {
+    out[0] = ScalarOps<T>::load(p + i);
+  }
+}
+
+template <typename T, bool V8>
+DEV_INLINE void row_store8(T* p, int i, const float* in) {
+  if constexpr (V8) {
+    T buf[8];
+#pragma unroll
+    for (int e = 0; e < 8; ++e) ScalarOps<T>::store(buf + e, in[e]);
+    Vec8<T>::store(p + i, buf);
+  } else {
+    ScalarOps<T>::store(p + i, in[0]);
+  }
+}
+
+#define ROW_LOOP(i) \
+  for (int i = threadIdx.x * EW; i < D; i += blockDim.x * EW)
+
 // ------------------------------ LayerNorm ------------------------------
 
-template <typename T>
+template <typename T, bool V8>
 __global__ void layernorm_fwd_kernel(
     const T* __restrict__ x, const T* __restrict__ w, const T* __restrict__ b,
     T* __restrict__ y, float* __restrict__ mean_out, float* __restrict__ rstd_out,
     long rows, int D, float eps) {
+  constexpr int EW = V8 ? 8 : 1;
   __shared__ float red[16];
   for (long row = blockIdx.x; row < rows; row += gridDim.x) {
     const T* xr = x + row * (long)D;
     T* yr = y + row * (long)D;
     float s = 0.f, s2 = 0.f;
-    for (int i = threadIdx.x; i < D; i += blockDim.x) {
-      float v = ScalarOps<T>::load(xr + i);
-      s += v;
-      s2 += v * v;
+    ROW_LOOP(i) {
+      float v[EW];
+      row_load8<T, V8>(xr, i, v);
+#pragma unroll
+      for (int e = 0; e < EW; ++e) {
+        s += v[e];
+        s2 += v[e] * v[e];
+      }
     }
     s = block_reduce_sum(s, red);
     __syncthreads();
@@ -38,22 +70,26 @@ __global__ void layernorm_fwd_kernel(
       mean_out[row] = mean;
       rstd_out[row] = rstd;
     }
-    for (int i = threadIdx.x; i < D; i += blockDim.x) {
-      float v = ScalarOps<T>::load(xr + i);
-      float wi = ScalarOps<T>::load(w + i);
-      float bi = ScalarOps<T>::load(b + i);
-      ScalarOps<T>::store(yr + i, (v - mean) * rstd * wi + bi);
+    ROW_LOOP(i) {
+      float v[EW], wv[EW], bv[EW], o[EW];
+      row_load8<T, V8>(xr, i, v);
+      row_load8<T, V8>(w, i, wv);
+      row_load8<T, V8>(b, i, bv);
+#pragma unroll
+      for (int e = 0; e < EW; ++e) o[e] = (v[e] - mean) * rstd * wv[e] + bv[e];
+      row_store8<T, V8>(yr, i, o);
     }
     __syncthreads();
   }
 }
 
-template <typename T>
+template <typename T, bool V8>
 __global__ void layernorm_bwd_kernel(
     const T* __restrict__ dy, const T* __restrict__ x, const T* __restrict__ w,
     const float* __restrict__ mean, const float* __restrict__ rstd,
     T* __restrict__ dx, float* __restrict__ dw, float* __restrict__ db,
     long rows, int D) {
+  constexpr int EW = V8 ? 8 : 1;
   extern __shared__ float smem[];  // [16 red] + [D dw] + [D db]
   float* red = smem;
   float* dw_acc = smem + 16;
@@ -69,26 +105,36 @@ __global__ void layernorm_bwd_kernel(
     T* dxr = dx + row * (long)D;
     const float m = mean[row], r = rstd[row];
     float sum_dyw = 0.f, sum_dyw_xhat = 0.f;
-    for (int i = threadIdx.x; i < D; i += blockDim.x) {
-      float g = ScalarOps<T>::load(dyr + i);
-      float xhat = (ScalarOps<T>::load(xr + i) - m) * r;
-      float wi = ScalarOps<T>::load(w + i);
-      float gw = g * wi;
-      sum_dyw += gw;
-      sum_dyw_xhat += gw * xhat;
-      dw_acc[i] += g * xhat;
-      db_acc[i] += g;
+    ROW_LOOP(i) {
+      float g[EW], v[EW], wv[EW];
+      row_load8<T, V8>(dyr, i, g);
+      row_load8<T, V8>(xr, i, v);
+      row_load8<T, V8>(w, i, wv);
+#pragma unroll
+      for (int e = 0; e < EW; ++e) {
+        const float xhat = (v[e] - m) * r;
+        const float gw = g[e] * wv[e];
+        sum_dyw += gw;
+        sum_dyw_xhat += gw * xhat;
+        dw_acc[i + e] += g[e] * xhat;
+        db_acc[i + e] += g[e];
+      }
     }
     sum_dyw = block_reduce_sum(sum_dyw, red);
     __syncthreads();
     sum_dyw_xhat = block_reduce_sum(sum_dyw_xhat, red);
     const float inv_d = 1.0f / D;
-    for (int i = threadIdx.x; i < D; i += blockDim.x) {
-      float g = ScalarOps<T>::load(dyr + i);
-      float xhat = (ScalarOps<T>::load(xr + i) - m) * r;
-      float wi = ScalarOps<T>::load(w + i);
-      float v = (g * wi - (sum_dyw + xhat * sum_dyw_xhat) * inv_d) * r;
-      ScalarOps<T>::store(dxr + i, v);
+    ROW_LOOP(i) {
+      float g[EW], v[EW], wv[EW], o[EW];
+      row_load8<T, V8>(dyr, i, g);
+      row_load8<T, V8>(xr, i, v);
+      row_load8<T, V8>(w, i, wv);
+#pragma unroll
+      for (int e = 0; e < EW; ++e) {
+        const float xhat = (v[e] - m) * r;
+        o[e] = (g[e] * wv[e] - (sum_dyw + xhat * sum_dyw_xhat) * inv_d) * r;
+      }
+      row_store8<T, V8>(dxr, i, o);
     }
     __syncthreads();
   }
@@ -100,36 +146,43 @@ __global__ void layernorm_bwd_kernel(
 
 // ------------------------------ RMSNorm --------------------------------
 
-template <typename T>
+template <typename T, bool V8>
 __global__ void rmsnorm_fwd_kernel(
     const T* __restrict__ x, const T* __restrict__ w, T* __restrict__ y,
     float* __restrict__ rstd_out, long rows, int D, float eps) {
+  constexpr int EW = V8 ? 8 : 1;
   __shared__ float red[16];
   for (long row = blockIdx.x; row < rows; row += gridDim.x) {
     const T* xr = x + row * (long)D;
     T* yr = y + row * (long)D;
     float s2 = 0.f;
-    for (int i = threadIdx.x; i < D; i += blockDim.x) {
-      float v = ScalarOps<T>::load(xr + i);
-      s2 += v * v;
+    ROW_LOOP(i) {
+      float v[EW];
+      row_load8<T, V8>(xr, i, v);
+#pragma unroll
+      for (int e = 0; e < EW; ++e) s2 += v[e] * v[e];
     }
     s2 = block_reduce_sum(s2, red);
     float rstd = rsqrtf(s2 / D + eps);
     if (threadIdx.x == 0) rstd_out[row] = rstd;
-    for (int i = threadIdx.x; i < D; i += blockDim.x) {
-      float v = ScalarOps<T>::load(xr + i);
-      float wi = ScalarOps<T>::load(w + i);
-      ScalarOps<T>::store(yr + i, v * rstd * wi);
+    ROW_LOOP(i) {
+      float v[EW], wv[EW], o[EW];
+      row_load8<T, V8>(xr, i, v);
+      row_load8<T, V8>(w, i, wv);
+#pragma unroll
+      for (int e = 0; e < EW; ++e) o[e] = v[e] * rstd * wv[e];
+      row_store8<T, V8>(yr, i, o);
     }
     __syncthreads();
   }
 }
 
-template <typename T>
+template <typename T, bool V8>
 __global__ void rmsnorm_bwd_kernel(
     const T* __restrict__ dy, const T* __restrict__ x, const T* __restrict__ w,
     const float* __restrict__ rstd, T* __restrict__ dx, float* __restrict__ dw,
     long rows, int D) {
+  constexpr int EW = V8 ? 8 : 1;
   extern __shared__ float smem[];
   float* red = smem;
   float* dw_acc = smem + 16;
@@ -141,20 +194,27 @@ __global__ void rmsnorm_bwd_kernel(
     T* dxr = dx + row * (long)D;
     const float r = rstd[row];
     float sum_gxw = 0.f;
-    for (int i = threadIdx.x; i < D; i += blockDim.x) {
-      float g = ScalarOps<T>::load(dyr + i);
-      float v = ScalarOps<T>::load(xr + i);
-      float wi = ScalarOps<T>::load(w + i);
-      sum_gxw += g * wi * v;
-      dw_acc[i] += g * v * r;
+    ROW_LOOP(i) {
+      float g[EW], v[EW], wv[EW];
+      row_load8<T, V8>(dyr, i, g);
+      row_load8<T, V8>(xr, i, v);
+      row_load8<T, V8>(w, i, wv);
+#pragma unroll
+      for (int e = 0; e < EW; ++e) {
+        sum_gxw += g[e] * wv[e] * v[e];
+        dw_acc[i + e] += g[e] * v[e] * r;
+      }
     }
     sum_gxw = block_reduce_sum(sum_gxw, red);
     const float c = sum_gxw * r * r * r / D;
-    for (int i = threadIdx.x; i < D; i += blockDim.x) {
-      float g = ScalarOps<T>::load(dyr + i);
-      float v = ScalarOps<T>::load(xr + i);
-      float wi = ScalarOps<T>::load(w + i);
-      ScalarOps<T>::store(dxr + i, g * wi * r - v * c);
+    ROW_LOOP(i) {
+      float g[EW], v[EW], wv[EW], o[EW];
+      row_load8<T, V8>(dyr, i, g);
+      row_load8<T, V8>(xr, i, v);
+      row_load8<T, V8>(w, i, wv);
+#pragma unroll
+      for (int e = 0; e < EW; ++e) o[e] = g[e] * wv[e] * r - v[e] * c;
+      row_store8<T, V8>(dxr, i, o);
     }
     __syncthreads();
   }
@@ -164,52 +224,66 @@ __global__ void rmsnorm_bwd_kernel(
 // ------------------------------ L2 norm --------------------------------
 // y = x / (||x|| + eps); saves s = 1/(||x|| + eps). ||x|| = 1/s - eps.
 
-template <typename T>
+template <typename T, bool V8>
 __global__ void l2norm_fwd_kernel(
     const T* __restrict__ x, T* __restrict__ y, float* __restrict__ s_out,
     long rows, int D, float eps) {
+  constexpr int EW = V8 ? 8 : 1;
   __shared__ float red[16];
   for (long row = blockIdx.x; row < rows; row += gridDim.x) {
     const T* xr = x + row * (long)D;
     T* yr = y + row * (long)D;
     float s2 = 0.f;
-    for (int i = threadIdx.x; i < D; i += blockDim.x) {
-      float v = ScalarOps<T>::load(xr + i);
-      s2 += v * v;
+    ROW_LOOP(i) {
+      float v[EW];
+      row_load8<T, V8>(xr, i, v);
+#pragma unroll
+      for (int e = 0; e < EW; ++e) s2 += v[e] * v[e];
     }
     s2 = block_reduce_sum(s2, red);
     float s = 1.0f / (sqrtf(s2) + eps);
     if (threadIdx.x == 0) s_out[row] = s;
-    for (int i = threadIdx.x; i < D; i += blockDim.x) {
-      float v = ScalarOps<T>::load(xr + i);
-      ScalarOps<T>::store(yr + i, v * s);
+    ROW_LOOP(i) {
+      float v[EW], o[EW];
+      row_load8<T, V8>(xr, i, v);
+#pragma unroll
+      for (int e = 0; e < EW; ++e) o[e] = v[e] * s;
+      row_store8<T, V8>(yr, i, o);
     }
     __syncthreads();
   }
 }
 
-template <typename T>
+template <typename T, bool V8>
 __global__ void l2norm_bwd_kernel(
     const T* __restrict__ dy, const T* __restrict__ y, const float* __restrict__ s_in,
     T* __restrict__ dx, long rows, int D, float eps) {
+  constexpr int EW = V8 ? 8 : 1;
   __shared__ float red[16];
   for (long row = blockIdx.x; row < rows; row += gridDim.x) {
     const T* dyr = dy + row * (long)D;
     const T* yr = y + row * (long)D;
     T* dxr = dx + row * (long)D;
     const float s = s_in[row];
-    const float n = 1.0f / s - eps;  // the original norm
+    const float n = 1.0f / s - eps;
     float dot = 0.f;
-    for (int i = threadIdx.x; i < D; i += blockDim.x) {
-      dot += ScalarOps<T>::load(dyr + i) * ScalarOps<T>::load(yr + i);
+    ROW_LOOP(i) {
+      float g[EW], v[EW];
+      row_load8<T, V8>(dyr, i, g);
+      row_load8<T, V8>(yr, i, v);
+#pragma unroll
+      for (int e = 0; e < EW; ++e) dot += g[e] * v[e];
     }
     dot = block_reduce_sum(dot, red);
     // y = x * s, s = 1/(n+eps)  =>  dx = s*dy - (dot(dy,y)/n) * y
     const float c = dot / fmaxf(n, 1e-20f);
-    for (int i = threadIdx.x; i < D; i += blockDim.x) {
-      float g = ScalarOps<T>::load(dyr + i);
-      float yv = ScalarOps<T>::load(yr + i);
-      ScalarOps<T>::store(dxr + i, s * g - c * yv);
+    ROW_LOOP(i) {
+      float g[EW], v[EW], o[EW];
+      row_load8<T, V8>(dyr, i, g);
+      row_load8<T, V8>(yr, i, v);
+#pragma unroll
+      for (int e = 0; e < EW; ++e) o[e] = s * g[e] - c * v[e];
+      row_store8<T, V8>(dxr, i, o);
     }
     __syncthreads();
   }
@@ -217,55 +291,69 @@ __global__ void l2norm_bwd_kernel(
 
 // ------------------------------ C wrappers -----------------------------
 
+#define DISPATCH_V8(D, ...)            \
+  if ((D) % 8 == 0) {                  \
+    constexpr bool V8 = true;          \
+    __VA_ARGS__;                       \
+  } else {                             \
+    constexpr bool V8 = false;         \
+    __VA_ARGS__;                       \
+  }
+
 template <typename T>
 void launch_layernorm_fwd(const T* x, const T* w, const T* b, T* y, float* mean,
                           float* rstd, long rows, int D, float eps, hipStream_t stream) {
   int grid = (int)min(rows, (long)8192);
-  hipLaunchKernelGGL((layernorm_fwd_kernel<T>), dim3(grid), dim3(NORM_BLOCK), 0, stream,
-                     x, w, b, y, mean, rstd, rows, D, eps);
+  DISPATCH_V8(D, hipLaunchKernelGGL(HIP_KERNEL_NAME(layernorm_fwd_kernel<T, V8>), dim3(grid),
+                                     dim3(NORM_BLOCK), 0, stream, x, w, b, y, mean, rstd,
+                                     rows, D, eps));
 }
 
 template <typename T>
 void launch_layernorm_bwd(const T* dy, const T* x, const T* w, const float* mean,
                           const float* rstd, T* dx, float* dw, float* db, long rows,
                           int D, hipStream_t stream) {
-  int grid = (int)min(rows, (long)1024);
+  int grid = (int)min(rows, (long)2048);
   size_t shmem = (16 + 2 * (size_t)D) * sizeof(float);
-  hipLaunchKernelGGL((layernorm_bwd_kernel<T>), dim3(grid), dim3(NORM_BLOCK), shmem, stream,
-                     dy, x, w, mean, rstd, dx, dw, db, rows, D);
+  DISPATCH_V8(D, hipLaunchKernelGGL(HIP_KERNEL_NAME(layernorm_bwd_kernel<T, V8>), dim3(grid),
+                                     dim3(NORM_BLOCK), shmem, stream, dy, x, w, mean,
+                                     rstd, dx, dw, db, rows, D));
 }
 
 template <typename T>
 void launch_rmsnorm_fwd(const T* x, const T* w, T* y, float* rstd, long rows, int D,
                         float eps, hipStream_t stream) {
   int grid = (int)min(rows, (long)8192);
-  hipLaunchKernelGGL((rmsnorm_fwd_kernel<T>), dim3(grid), dim3(NORM_BLOCK), 0, stream,
-                     x, w, y, rstd, rows, D, eps);
+  DISPATCH_V8(D, hipLaunchKernelGGL(HIP_KERNEL_NAME(rmsnorm_fwd_kernel<T, V8>), dim3(grid),
+                                     dim3(NORM_BLOCK), 0, stream, x, w, y, rstd, rows, D,
+                                     eps));
 }
 
 template <typename T>
 void launch_rmsnorm_bwd(const T* dy, const T* x, const T* w, const float* rstd, T* dx,
                         float* dw, long rows, int D, hipStream_t stream) {
-  int grid = (int)min(rows, (long)1024);
+  int grid = (int)min(rows, (long)2048);
   size_t shmem = (16 + (size_t)D) * sizeof(float);
-  hipLaunchKernelGGL((rmsnorm_bwd_kernel<T>), dim3(grid), dim3(NORM_BLOCK), shmem, stream,
-                     dy, x, w, rstd, dx, dw, rows, D);
+  DISPATCH_V8(D, hipLaunchKernelGGL(HIP_KERNEL_NAME(rmsnorm_bwd_kernel<T, V8>), dim3(grid),
+                                     dim3(NORM_BLOCK), shmem, stream, dy, x, w, rstd, dx,
+                                     dw, rows, D));
 }
 
 template <typename T>
 void launch_l2norm_fwd(const T* x, T* y, float* s, long rows, int D, float eps,
                        hipStream_t stream) {
   int grid = (int)min(rows, (long)8192);
-  hipLaunchKernelGGL((l2norm_fwd_kernel<T>), dim3(grid), dim3(NORM_BLOCK), 0, stream,
-                     x, y, s, rows, D, eps);
+  DISPATCH_V8(D, hipLaunchKernelGGL(HIP_KERNEL_NAME(l2norm_fwd_kernel<T, V8>), dim3(grid),
+                                     dim3(NORM_BLOCK), 0, stream, x, y, s, rows, D, eps));
 }
 
 template <typename T>
 void launch_l2norm_bwd(const T* dy, const T* y, const float* s, T* dx, long rows, int D,
                        float eps, hipStream_t stream) {
   int grid = (int)min(rows, (long)8192);
-  hipLaunchKernelGGL((l2norm_bwd_kernel<T>), dim3(grid), dim3(NORM_BLOCK), 0, stream,
-                     dy, y, s, dx, rows, D, eps);
+  DISPATCH_V8(D, hipLaunchKernelGGL(HIP_KERNEL_NAME(l2norm_bwd_kernel<T, V8>), dim3(grid),
+                                     dim3(NORM_BLOCK), 0, stream, dy, y, s, dx, rows, D,
+                                     eps));
 }
 
 // explicit instantiations used by bindings.cpp
