@@ -116,9 +116,9 @@ def main():
     model.train()
 
     groups = model.get_params_groups()
-    optimizer = FusedAdamW(groups, beta1=cfg.optim.adamw_beta1, beta2=cfg.optim.adamw_beta2)
-    params = [p for g in groups for p in g["params"]]
-    reducer = GradReducer(params, reduce_dtype=torch.float32)
+    from dinov3_amd.train.train import build_training_engine
+
+    optimizer, finalize_backward = build_training_engine(cfg, groups)
 
     batches = make_synthetic_batch(cfg, device, dtype)
     clip = cfg.optim.clip_grad
@@ -127,7 +127,7 @@ def main():
         data = batches[i % len(batches)]
         loss, _ = model(data, teacher_temp=0.07, iteration=i)
         loss.backward()
-        reducer.finalize()
+        finalize_backward()
         clip_scales = None
         if clip:
             sums = optimizer.grad_norm_sums()
@@ -168,6 +168,8 @@ def main():
     ms_per_step = elapsed / args.steps * 1000.0
 
     if rank == 0:
+        parallelism = ("fsdp" if (world > 1 and cfg.compute_precision.sharding_strategy
+                                  in ("SHARD_GRAD_OP", "FULL_SHARD")) else "dp")
         result = {
             "metric": "images/sec (whole node) ViT-L/16 DINOv3 pretrain 224px",
             "value": img_per_sec,
@@ -186,7 +188,7 @@ def main():
                 "global_batch": args.batch_size * world,
                 "img_size": args.global_size,
                 "local_crops": args.local_crops,
-                "parallelism": f"dp{world}",
+                "parallelism": f"{parallelism}{world}",
             },
         }
         print(json.dumps(result))

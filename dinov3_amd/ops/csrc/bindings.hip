@@ -90,8 +90,8 @@ void launch_multi_tensor_adamw(T* const*, const T* const*, float* const*, float*
 template <typename T>
 void launch_multi_tensor_l2norm_sq(const T* const*, const long*, const int*, const long*,
                                    int, float*, hipStream_t);
-template <typename T>
-void launch_multi_tensor_adamw_planned(T* const*, const T* const*, float* const*,
+template <typename T, typename GT>
+void launch_multi_tensor_adamw_planned(T* const*, const GT* const*, float* const*,
                                        float* const*, float* const*, const long*,
                                        const int*, const long*, int, const float*,
                                        const float*, const float*, const int*,
@@ -560,12 +560,21 @@ void multi_tensor_adamw_planned(torch::Tensor ptrs, torch::Tensor sizes, torch::
                                 torch::Tensor sub_id, torch::Tensor clip, double lr,
                                 double last_lr, double wd, double beta1, double beta2,
                                 double eps, double bc1, double bc2, bool has_master,
-                                bool is_bf16) {
+                                bool is_bf16, bool grad_is_f32) {
   const int n_chunks = (int)ct.numel();
   long* pp = ptrs.data_ptr<long>();
   const long n = n_tensors;
-  if (is_bf16) {
-    launch_multi_tensor_adamw_planned<__hip_bfloat16>(
+  if (is_bf16 && grad_is_f32) {
+    launch_multi_tensor_adamw_planned<__hip_bfloat16, float>(
+        (__hip_bfloat16* const*)pp, (const float* const*)(pp + n),
+        (float* const*)(pp + 2 * n), (float* const*)(pp + 3 * n),
+        has_master ? (float* const*)(pp + 4 * n) : nullptr, sizes.data_ptr<long>(),
+        ct.data_ptr<int>(), co.data_ptr<long>(), n_chunks, lr_mult.data_ptr<float>(),
+        wd_mult.data_ptr<float>(), is_last.data_ptr<float>(), sub_id.data_ptr<int>(),
+        clip.data_ptr<float>(), (float)lr, (float)last_lr, (float)wd, (float)beta1,
+        (float)beta2, (float)eps, (float)bc1, (float)bc2, has_master, current_stream());
+  } else if (is_bf16) {
+    launch_multi_tensor_adamw_planned<__hip_bfloat16, __hip_bfloat16>(
         (__hip_bfloat16* const*)pp, (const __hip_bfloat16* const*)(pp + n),
         (float* const*)(pp + 2 * n), (float* const*)(pp + 3 * n),
         has_master ? (float* const*)(pp + 4 * n) : nullptr, sizes.data_ptr<long>(),
@@ -574,7 +583,7 @@ void multi_tensor_adamw_planned(torch::Tensor ptrs, torch::Tensor sizes, torch::
         clip.data_ptr<float>(), (float)lr, (float)last_lr, (float)wd, (float)beta1,
         (float)beta2, (float)eps, (float)bc1, (float)bc2, has_master, current_stream());
   } else {
-    launch_multi_tensor_adamw_planned<float>(
+    launch_multi_tensor_adamw_planned<float, float>(
         (float* const*)pp, (const float* const*)(pp + n), (float* const*)(pp + 2 * n),
         (float* const*)(pp + 3 * n), has_master ? (float* const*)(pp + 4 * n) : nullptr,
         sizes.data_ptr<long>(), ct.data_ptr<int>(), co.data_ptr<long>(), n_chunks,

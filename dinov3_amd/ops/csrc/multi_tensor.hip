@@ -106,9 +106,9 @@ __global__ void multi_tensor_l2norm_sq_kernel(
 // wd_mult[t]; grad scale = clip[submodel_id[t]] (per-submodel global-norm
 // clip factors, uploaded per step as a tiny array).
 
-template <typename T, bool HAS_MASTER>
+template <typename T, typename GT, bool HAS_MASTER>
 __global__ void multi_tensor_adamw_planned_kernel(
-    T* const* __restrict__ p_ptrs, const T* const* __restrict__ g_ptrs,
+    T* const* __restrict__ p_ptrs, const GT* const* __restrict__ g_ptrs,
     float* const* __restrict__ m_ptrs, float* const* __restrict__ v_ptrs,
     float* const* __restrict__ w_ptrs, const long* __restrict__ sizes,
     const int* __restrict__ chunk_tensor, const long* __restrict__ chunk_offset,
@@ -123,7 +123,7 @@ __global__ void multi_tensor_adamw_planned_kernel(
     const long off = chunk_offset[c];
     const long n = min((long)MT_CHUNK, sizes[tid] - off);
     T* p = p_ptrs[tid] + off;
-    const T* g = g_ptrs[tid] + off;
+    const GT* g = g_ptrs[tid] + off;
     float* m = m_ptrs[tid] + off;
     float* v = v_ptrs[tid] + off;
     float* w = HAS_MASTER ? (w_ptrs[tid] + off) : nullptr;
@@ -131,7 +131,7 @@ __global__ void multi_tensor_adamw_planned_kernel(
     const float gwd = wd * wd_mult[tid];
     const float gscale = clip[sub_id[tid]];
     for (long i = threadIdx.x; i < n; i += blockDim.x) {
-      float gv = ScalarOps<T>::load(g + i) * gscale;
+      float gv = ScalarOps<GT>::load(g + i) * gscale;
       float mv = beta1 * m[i] + (1.0f - beta1) * gv;
       float vv = beta2 * v[i] + (1.0f - beta2) * gv * gv;
       m[i] = mv;
@@ -169,24 +169,24 @@ __global__ void multi_tensor_l2norm_planned_kernel(
   }
 }
 
-template <typename T>
+template <typename T, typename GT>
 void launch_multi_tensor_adamw_planned(
-    T* const* p, const T* const* g, float* const* m, float* const* v, float* const* w,
+    T* const* p, const GT* const* g, float* const* m, float* const* v, float* const* w,
     const long* sizes, const int* chunk_tensor, const long* chunk_offset, int n_chunks,
     const float* lr_mult, const float* wd_mult, const float* is_last, const int* sub_id,
     const float* clip, float lr, float last_lr, float wd, float beta1, float beta2,
     float eps, float bc1, float bc2, bool has_master, hipStream_t stream) {
   int grid = min(n_chunks, 4096);
   if (has_master) {
-    hipLaunchKernelGGL((multi_tensor_adamw_planned_kernel<T, true>), dim3(grid),
-                       dim3(MT_BLOCK), 0, stream, p, g, m, v, w, sizes, chunk_tensor,
-                       chunk_offset, n_chunks, lr_mult, wd_mult, is_last, sub_id, clip, lr,
-                       last_lr, wd, beta1, beta2, eps, bc1, bc2);
+    hipLaunchKernelGGL(HIP_KERNEL_NAME(multi_tensor_adamw_planned_kernel<T, GT, true>),
+                       dim3(grid), dim3(MT_BLOCK), 0, stream, p, g, m, v, w, sizes,
+                       chunk_tensor, chunk_offset, n_chunks, lr_mult, wd_mult, is_last,
+                       sub_id, clip, lr, last_lr, wd, beta1, beta2, eps, bc1, bc2);
   } else {
-    hipLaunchKernelGGL((multi_tensor_adamw_planned_kernel<T, false>), dim3(grid),
-                       dim3(MT_BLOCK), 0, stream, p, g, m, v, w, sizes, chunk_tensor,
-                       chunk_offset, n_chunks, lr_mult, wd_mult, is_last, sub_id, clip, lr,
-                       last_lr, wd, beta1, beta2, eps, bc1, bc2);
+    hipLaunchKernelGGL(HIP_KERNEL_NAME(multi_tensor_adamw_planned_kernel<T, GT, false>),
+                       dim3(grid), dim3(MT_BLOCK), 0, stream, p, g, m, v, w, sizes,
+                       chunk_tensor, chunk_offset, n_chunks, lr_mult, wd_mult, is_last,
+                       sub_id, clip, lr, last_lr, wd, beta1, beta2, eps, bc1, bc2);
   }
 }
 
@@ -201,18 +201,24 @@ void launch_multi_tensor_l2norm_planned(const T* const* g, const long* sizes,
                      out);
 }
 
-#define INSTANTIATE_MT_PLANNED(T)                                                         \
-  template void launch_multi_tensor_adamw_planned<T>(                                     \
-      T* const*, const T* const*, float* const*, float* const*, float* const*,           \
+#define INSTANTIATE_MT_PLANNED(T, GT)                                                     \
+  template void launch_multi_tensor_adamw_planned<T, GT>(                                 \
+      T* const*, const GT* const*, float* const*, float* const*, float* const*,          \
       const long*, const int*, const long*, int, const float*, const float*,             \
       const float*, const int*, const float*, float, float, float, float, float, float,  \
-      float, float, bool, hipStream_t);                                                  \
+      float, float, bool, hipStream_t);
+
+INSTANTIATE_MT_PLANNED(float, float)
+INSTANTIATE_MT_PLANNED(__hip_bfloat16, __hip_bfloat16)
+INSTANTIATE_MT_PLANNED(__hip_bfloat16, float)
+
+#define INSTANTIATE_MT_L2P(T)                                                             \
   template void launch_multi_tensor_l2norm_planned<T>(const T* const*, const long*,       \
                                                       const int*, const long*, int,       \
                                                       const int*, float*, hipStream_t);
 
-INSTANTIATE_MT_PLANNED(float)
-INSTANTIATE_MT_PLANNED(__hip_bfloat16)
+INSTANTIATE_MT_L2P(float)
+INSTANTIATE_MT_L2P(__hip_bfloat16)
 
 // ---------------------------- C wrappers -------------------------------
 

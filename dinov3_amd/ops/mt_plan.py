@@ -59,10 +59,11 @@ def adamw_planned(plan: MultiTensorPlan, lr_mult: torch.Tensor, wd_mult: torch.T
     plan.check_pointers()
     bc1 = 1.0 - beta1**step
     bc2 = 1.0 - beta2**step
+    grad_is_f32 = plan.lists[1][0].dtype == torch.float32
     hip_ops().multi_tensor_adamw_planned(
         plan.ptrs, plan.sizes, plan.ct, plan.co, plan.n_tensors, lr_mult, wd_mult,
         is_last, sub_id, clip, lr, last_lr, wd, beta1, beta2, eps, bc1, bc2, has_master,
-        plan.is_bf16,
+        plan.is_bf16, grad_is_f32,
     )
 
 

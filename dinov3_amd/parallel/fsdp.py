@@ -1,0 +1,312 @@
+"""Sharded training engine: ZeRO-2-style parameter/grad/optimizer sharding
+over RCCL on xGMI (the config's SHARD_GRAD_OP strategy).
+
+MI355X-first design (deliberately NOT the reference's gather-on-compute
+wrapper, dinov3_jax/fsdp/utils.py:19-94): with 288 GB HBM3E per GPU, full
+bf16 parameters stay RESIDENT on every rank — no per-forward all-gather at
+all. Per step:
+
+  backward  ->  per-bucket fp32 reduce-scatter (async, comm stream,
+                overlapped with the remaining backward)   [C2 in SURVEY §2.3]
+  step      ->  ONE planned AdamW kernel over this rank's fp32 grad shards /
+                sharded fp32 m/v/master -> updated bf16 param shards
+            ->  per-bucket bf16 all-gather republishes full params (async)
+
+Communication per step: 2 bytes/param fp32-RS-equivalent + 2 bytes/param AG
+— half of a fp32 DDP all-reduce — with optimizer state and master weights
+sharded 1/world. Buckets are the fused param groups (uniform lr/wd within a
+bucket), padded to world*64 elements; gradient buckets are persistent flat
+buffers that autograd accumulates into via .grad views.
+
+Works on gloo/CPU for tests (reduce_scatter emulated with all_reduce).
+"""
+
+from __future__ import annotations
+
+import logging
+from typing import Dict, List, Optional
+
+import torch
+import torch.distributed as dist
+
+logger = logging.getLogger("dinov3")
+
+
+def _pad_to(n: int, multiple: int) -> int:
+    return (n + multiple - 1) // multiple * multiple
+
+
+def _reduce_scatter(shard: torch.Tensor, full: torch.Tensor, group) -> Optional[dist.Work]:
+    """shard <- sum over ranks of full's rank-th slice. Async when supported."""
+    backend = dist.get_backend(group)
+    if backend == "nccl":
+        return dist.reduce_scatter_tensor(shard, full, op=dist.ReduceOp.SUM,
+                                          group=group, async_op=True)
+    # gloo fallback: all-reduce the full bucket, then slice-copy
+    work = dist.all_reduce(full, op=dist.ReduceOp.SUM, group=group, async_op=True)
+    work.wait()
+    rank = dist.get_rank(group)
+    n = shard.numel()
+    shard.copy_(full[rank * n: (rank + 1) * n])
+    return None
+
+
+def _all_gather(full: torch.Tensor, shard: torch.Tensor, group) -> Optional[dist.Work]:
+    backend = dist.get_backend(group)
+    if backend == "nccl":
+        return dist.all_gather_into_tensor(full, shard, group=group, async_op=True)
+    world = dist.get_world_size(group)
+    n = shard.numel()
+    chunks = list(full.split(n))
+    work = dist.all_gather(chunks, shard.contiguous(), group=group, async_op=True)
+    work.wait()
+    return None
+
+
+class ShardedEngine:
+    """Owns the student's fused param groups: flat param/grad buckets,
+    sharded optimizer state, backward-overlapped gradient reduction."""
+
+    def __init__(self, param_groups: List[dict], beta1: float = 0.9, beta2: float = 0.999,
+                 eps: float = 1e-8, process_group=None, align: int = 64,
+                 comm_dtype: torch.dtype = torch.float32):
+        assert dist.is_available() and dist.is_initialized(), "ShardedEngine needs torch.distributed"
+        self.group = process_group
+        self.world = dist.get_world_size(process_group)
+        self.rank = dist.get_rank(process_group)
+        self.beta1, self.beta2, self.eps = beta1, beta2, eps
+        self.step_count = 0
+        self.comm_dtype = comm_dtype
+        self.groups = param_groups
+        self.submodels = sorted({g["submodel"] for g in param_groups})
+        self._sub_idx = {s: i for i, s in enumerate(self.submodels)}
+
+        self.buckets: List[dict] = []
+        for g in param_groups:
+            params = g["params"]
+            device = params[0].device
+            dtype = params[0].dtype
+            total = sum(p.numel() for p in params)
+            padded = _pad_to(total, self.world * align)
+            shard_n = padded // self.world
+            flat = torch.zeros(padded, dtype=dtype, device=device)
+            # move param storages into the flat bucket (full params resident)
+            off = 0
+            for p in params:
+                n = p.numel()
+                flat[off: off + n].copy_(p.detach().reshape(-1))
+                p.data = flat[off: off + n].view(p.shape)
+                off += n
+            grad_flat = torch.zeros(padded, dtype=dtype, device=device)
+            off = 0
+            for p in params:
+                n = p.numel()
+                p.grad = grad_flat[off: off + n].view(p.shape)
+                off += n
+            s0, s1 = self.rank * shard_n, (self.rank + 1) * shard_n
+            bucket = {
+                "group": g,
+                "flat": flat,
+                "grad_flat": grad_flat,
+                "param_shard": flat[s0:s1],
+                "grad_shard": torch.zeros(shard_n, dtype=comm_dtype, device=device),
+                "exp_avg": torch.zeros(shard_n, dtype=torch.float32, device=device),
+                "exp_avg_sq": torch.zeros(shard_n, dtype=torch.float32, device=device),
+                "master": flat[s0:s1].float() if dtype != torch.float32 else None,
+                "n_params": len(params),
+                "pending": 0,
+                "rs_work": None,
+                "rs_buf": None,
+            }
+            self.buckets.append(bucket)
+
+        # hooks: count grads per bucket, fire async reduce-scatter when full
+        self._param_bucket: Dict[int, dict] = {}
+        self._hooks = []
+        for b in self.buckets:
+            for p in b["group"]["params"]:
+                self._param_bucket[id(p)] = b
+                self._hooks.append(p.register_post_accumulate_grad_hook(self._on_grad))
+        self._reset_pending()
+        self._plan = None
+        logger.info("ShardedEngine: %d buckets, world %d, %.1fM params/rank shard",
+                    len(self.buckets), self.world,
+                    sum(b["param_shard"].numel() for b in self.buckets) / 1e6)
+
+    # ------------------------------------------------------------------
+    def _reset_pending(self) -> None:
+        for b in self.buckets:
+            b["pending"] = b["n_params"]
+
+    def _on_grad(self, p: torch.nn.Parameter) -> None:
+        b = self._param_bucket[id(p)]
+        b["pending"] -= 1
+        if b["pending"] == 0:
+            self._launch_rs(b)
+
+    def _launch_rs(self, b: dict) -> None:
+        full = b["grad_flat"]
+        if full.dtype != self.comm_dtype:
+            full = full.to(self.comm_dtype)
+        b["rs_buf"] = full  # keep alive until work completes
+        b["rs_work"] = _reduce_scatter(b["grad_shard"], full, self.group)
+
+    def finalize_backward(self) -> None:
+        """Wait for all grad reductions; average (sum -> mean)."""
+        for b in self.buckets:
+            if b["pending"] > 0:  # params without grads this step
+                self._launch_rs(b)
+                b["pending"] = 0
+        for b in self.buckets:
+            if b["rs_work"] is not None:
+                b["rs_work"].wait()
+                b["rs_work"] = None
+            b["rs_buf"] = None
+            b["grad_shard"].div_(self.world)
+        self._reset_pending()
+
+    # ------------------------------------------------------------------
+    @torch.no_grad()
+    def grad_norm_sums(self) -> torch.Tensor:
+        """Per-submodel sum of squared grads over THIS rank's shards; caller
+        all-reduces across ranks (shards are disjoint -> sum is global)."""
+        device = self.buckets[0]["grad_shard"].device
+        sums = torch.zeros(len(self.submodels), dtype=torch.float32, device=device)
+        if device.type == "cuda":
+            from ..ops.mt_plan import MultiTensorPlan, l2norm_planned
+
+            if getattr(self, "_gplan", None) is None:
+                self._gplan = MultiTensorPlan([[b["grad_shard"] for b in self.buckets]])
+                self._gsub = torch.tensor(
+                    [self._sub_idx[b["group"]["submodel"]] for b in self.buckets],
+                    dtype=torch.int32, device=device,
+                )
+            return l2norm_planned(self._gplan, self._gsub, len(self.submodels))
+        for b in self.buckets:
+            sums[self._sub_idx[b["group"]["submodel"]]] += b["grad_shard"].float().pow(2).sum()
+        return sums
+
+    def clip_factors(self, sums: torch.Tensor, clip: float) -> torch.Tensor:
+        return torch.clamp(clip / (sums.sqrt() + 1e-6), max=1.0)
+
+    # ------------------------------------------------------------------
+    @torch.no_grad()
+    def step(self, lr: float, weight_decay: float, last_layer_lr: Optional[float] = None,
+             clip_scales=None) -> None:
+        self.step_count += 1
+        if last_layer_lr is None:
+            last_layer_lr = lr
+        device = self.buckets[0]["flat"].device
+        if device.type == "cuda":
+            self._step_planned(lr, weight_decay, last_layer_lr, clip_scales)
+        else:
+            self._step_eager(lr, weight_decay, last_layer_lr, clip_scales)
+        # republish updated param shards (async all-gathers, then wait)
+        works = []
+        for b in self.buckets:
+            w = _all_gather(b["flat"], b["param_shard"], self.group)
+            if w is not None:
+                works.append(w)
+        for w in works:
+            w.wait()
+
+    def _clip_tensor(self, clip_scales, device) -> torch.Tensor:
+        if clip_scales is None:
+            return torch.ones(len(self.submodels), dtype=torch.float32, device=device)
+        if isinstance(clip_scales, dict):
+            return torch.tensor([clip_scales.get(s, 1.0) for s in self.submodels],
+                                dtype=torch.float32, device=device)
+        return clip_scales.float().to(device)
+
+    def _step_planned(self, lr, wd, last_lr, clip_scales) -> None:
+        from ..ops.mt_plan import MultiTensorPlan, adamw_planned
+
+        if self._plan is None:
+            lists = [
+                [b["param_shard"] for b in self.buckets],
+                [b["grad_shard"] for b in self.buckets],
+                [b["exp_avg"] for b in self.buckets],
+                [b["exp_avg_sq"] for b in self.buckets],
+            ]
+            has_master = self.buckets[0]["master"] is not None
+            if has_master:
+                lists.append([b["master"] for b in self.buckets])
+            device = self.buckets[0]["flat"].device
+            self._plan = MultiTensorPlan(lists)
+            self._lr_mult = torch.tensor([b["group"]["lr_multiplier"] for b in self.buckets],
+                                         dtype=torch.float32, device=device)
+            self._wd_mult = torch.tensor([b["group"]["wd_multiplier"] for b in self.buckets],
+                                         dtype=torch.float32, device=device)
+            self._is_last = torch.tensor(
+                [1.0 if b["group"]["is_last_layer"] else 0.0 for b in self.buckets],
+                dtype=torch.float32, device=device)
+            self._sub_id = torch.tensor(
+                [self._sub_idx[b["group"]["submodel"]] for b in self.buckets],
+                dtype=torch.int32, device=device)
+            self._has_master = has_master
+        clip_t = self._clip_tensor(clip_scales, self.buckets[0]["flat"].device)
+        adamw_planned(self._plan, self._lr_mult, self._wd_mult, self._is_last,
+                      self._sub_id, clip_t, lr, last_lr, wd, self.beta1, self.beta2,
+                      self.eps, self.step_count, self._has_master)
+
+    def _step_eager(self, lr, wd, last_lr, clip_scales) -> None:
+        bc1 = 1.0 - self.beta1**self.step_count
+        bc2 = 1.0 - self.beta2**self.step_count
+        clip_t = self._clip_tensor(clip_scales, self.buckets[0]["flat"].device)
+        for b in self.buckets:
+            g = b["group"]
+            glr = (last_lr if g["is_last_layer"] else lr) * g["lr_multiplier"]
+            gwd = wd * g["wd_multiplier"]
+            scale = float(clip_t[self._sub_idx[g["submodel"]]])
+            grad = b["grad_shard"].float() * scale
+            m, v = b["exp_avg"], b["exp_avg_sq"]
+            m.mul_(self.beta1).add_(grad, alpha=1 - self.beta1)
+            v.mul_(self.beta2).addcmul_(grad, grad, value=1 - self.beta2)
+            work = b["master"] if b["master"] is not None else b["param_shard"]
+            if b["master"] is not None:
+                work = b["master"]
+            work.mul_(1.0 - glr * gwd)
+            work.add_((m / bc1) / ((v / bc2).sqrt() + self.eps), alpha=-glr)
+            if b["master"] is not None:
+                b["param_shard"].copy_(work.to(b["param_shard"].dtype))
+
+    # ------------------------------------------------------------------
+    def zero_grad(self, set_to_none: bool = False) -> None:
+        for b in self.buckets:
+            b["grad_flat"].zero_()
+            b["grad_shard"].zero_()
+
+    def state_dict(self) -> dict:
+        return {
+            "step_count": self.step_count,
+            "world": self.world,
+            "rank": self.rank,
+            "shards": [
+                {
+                    "names": b["group"]["names"],
+                    "exp_avg": b["exp_avg"],
+                    "exp_avg_sq": b["exp_avg_sq"],
+                    "master": b["master"],
+                }
+                for b in self.buckets
+            ],
+        }
+
+    def load_state_dict(self, state: dict) -> None:
+        assert state["world"] == self.world, "resharded restore not supported yet"
+        self.step_count = state["step_count"]
+        by_names = {tuple(s["names"]): s for s in state["shards"]}
+        for b in self.buckets:
+            s = by_names.get(tuple(b["group"]["names"]))
+            if s is None:
+                logger.warning("shard %s missing from checkpoint", b["group"]["names"][:1])
+                continue
+            b["exp_avg"].copy_(s["exp_avg"])
+            b["exp_avg_sq"].copy_(s["exp_avg_sq"])
+            if b["master"] is not None and s["master"] is not None:
+                b["master"].copy_(s["master"])
+
+    def remove_hooks(self) -> None:
+        for h in self._hooks:
+            h.remove()
+        self._hooks = []

@@ -116,6 +116,31 @@ def build_optimizer(cfg, params_groups) -> FusedAdamW:
     )
 
 
+def build_training_engine(cfg, params_groups):
+    """Select the distributed engine per compute_precision.sharding_strategy.
+
+    SHARD_GRAD_OP / FULL_SHARD + world>1 -> ShardedEngine (ZeRO-2 style:
+    resident full params, fp32 grad reduce-scatter overlapped with backward,
+    sharded optimizer state, bf16 param all-gather after the step).
+    Otherwise -> FusedAdamW + bucketed all-reduce GradReducer (NO_SHARD/DDP).
+
+    Returns (optimizer-like, finalize_backward_fn).
+    """
+    world = parallel.get_world_size()
+    strategy = cfg.compute_precision.sharding_strategy
+    if world > 1 and strategy in ("SHARD_GRAD_OP", "FULL_SHARD"):
+        from ..parallel.fsdp import ShardedEngine
+
+        engine = ShardedEngine(params_groups, beta1=cfg.optim.adamw_beta1,
+                               beta2=cfg.optim.adamw_beta2)
+        return engine, engine.finalize_backward
+    optimizer = build_optimizer(cfg, params_groups)
+    student_params = [p for g in params_groups for p in g["params"]]
+    reducer = GradReducer(student_params, reduce_dtype=torch.float32
+                          if cfg.compute_precision.reduce_dtype == "fp32" else None)
+    return optimizer, reducer.finalize
+
+
 def build_data_loader_from_cfg(cfg, model: SSLMetaArch, sampler_advance: int = 0):
     img_size = cfg.crops.global_crops_size
     patch_size = cfg.student.patch_size
@@ -184,10 +209,7 @@ def do_train(cfg, model: SSLMetaArch, resume: bool = True, max_iterations: int =
 
     schedulers = build_schedulers(cfg)
     params_groups = model.get_params_groups()
-    optimizer = build_optimizer(cfg, params_groups)
-    student_params = [p for g in params_groups for p in g["params"]]
-    reducer = GradReducer(student_params, reduce_dtype=torch.float32
-                          if cfg.compute_precision.reduce_dtype == "fp32" else None)
+    optimizer, finalize_backward = build_training_engine(cfg, params_groups)
 
     start_iter = 0
     output_dir = cfg.train.output_dir
@@ -251,7 +273,7 @@ def do_train(cfg, model: SSLMetaArch, resume: bool = True, max_iterations: int =
         nan_count = 0
 
         loss.backward()
-        reducer.finalize()
+        finalize_backward()
 
         clip_scales = None
         if clip is not None and clip > 0:
