@@ -87,16 +87,18 @@ class SelfAttentionBlock(nn.Module):
             flat = self._add_scaled(flat, self.mlp(self.norm2(flat)), self.ls2)
             return flat
 
+        from ..ops.row_ops import gather_rows, scatter_add_rows
+
         keep_ratio = 1.0 - self.sample_drop_ratio
         rows1, metas1, scale1 = _subset_rows(metas, keep_ratio, flat.device)
-        sub = flat.index_select(0, rows1)
+        sub = gather_rows(flat, rows1)
         res = self.ls1(self.attn.forward_flat(self.norm1(sub), metas1))
-        flat = flat.index_add(0, rows1, (res * scale1.unsqueeze(1)).to(flat.dtype))
+        flat = scatter_add_rows(flat, rows1, res, scale1)
 
         rows2, _, scale2 = _subset_rows(metas, keep_ratio, flat.device)
-        sub = flat.index_select(0, rows2)
+        sub = gather_rows(flat, rows2)
         res = self.ls2(self.mlp(self.norm2(sub)))
-        return flat.index_add(0, rows2, (res * scale2.unsqueeze(1)).to(flat.dtype))
+        return scatter_add_rows(flat, rows2, res, scale2)
 
     # ------------------------------------------------------------------
     def forward(self, x: torch.Tensor, rope: Optional[RopeSinCos] = None) -> torch.Tensor:

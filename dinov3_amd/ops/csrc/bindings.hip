@@ -36,6 +36,14 @@ void launch_ls_axpy_fwd(const T*, const T*, const T*, T*, long, int, hipStream_t
 template <typename T>
 void launch_ls_axpy_bwd(const T*, const T*, const T*, T*, float*, long, int, hipStream_t);
 template <typename T>
+void launch_row_gather(const T*, const long*, T*, long, int, hipStream_t);
+template <typename T>
+void launch_row_scatter_add(T*, const long*, const T*, const float*, long, int,
+                            hipStream_t);
+template <typename T>
+void launch_row_gather_scaled(const T*, const long*, const float*, T*, long, int,
+                              hipStream_t);
+template <typename T>
 void launch_swiglu_fwd(const T*, T*, long, int, hipStream_t);
 template <typename T>
 void launch_swiglu_bwd(const T*, const T*, T*, long, int, hipStream_t);
@@ -277,6 +285,47 @@ std::vector<torch::Tensor> ls_axpy_bwd(torch::Tensor dout, torch::Tensor res,
                                  rows, D, current_stream());
   });
   return {dres, dgamma.to(dout.scalar_type())};
+}
+
+torch::Tensor row_gather(torch::Tensor src, torch::Tensor idx) {
+  CHECK_INPUT(src);
+  const int D = src.size(-1);
+  TORCH_CHECK(D % 8 == 0, "row ops need D % 8 == 0");
+  const long M = idx.numel();
+  auto out = torch::empty({M, D}, src.options());
+  DISPATCH_FLOAT_BF16(src.scalar_type(), "row_gather", [&] {
+    launch_row_gather<scalar_t>((const scalar_t*)src.data_ptr(), idx.data_ptr<long>(),
+                                (scalar_t*)out.data_ptr(), M, D, current_stream());
+  });
+  return out;
+}
+
+void row_scatter_add_(torch::Tensor dst, torch::Tensor idx, torch::Tensor src,
+                      torch::Tensor scale) {
+  CHECK_INPUT(dst);
+  CHECK_INPUT(src);
+  const int D = dst.size(-1);
+  const long M = idx.numel();
+  const float* sp = (scale.defined() && scale.numel() > 0) ? scale.data_ptr<float>() : nullptr;
+  DISPATCH_FLOAT_BF16(dst.scalar_type(), "row_scatter_add", [&] {
+    launch_row_scatter_add<scalar_t>((scalar_t*)dst.data_ptr(), idx.data_ptr<long>(),
+                                     (const scalar_t*)src.data_ptr(), sp, M, D,
+                                     current_stream());
+  });
+}
+
+torch::Tensor row_gather_scaled(torch::Tensor src, torch::Tensor idx, torch::Tensor scale) {
+  CHECK_INPUT(src);
+  const int D = src.size(-1);
+  const long M = idx.numel();
+  auto out = torch::empty({M, D}, src.options());
+  const float* sp = (scale.defined() && scale.numel() > 0) ? scale.data_ptr<float>() : nullptr;
+  DISPATCH_FLOAT_BF16(src.scalar_type(), "row_gather_scaled", [&] {
+    launch_row_gather_scaled<scalar_t>((const scalar_t*)src.data_ptr(),
+                                       idx.data_ptr<long>(), sp, (scalar_t*)out.data_ptr(),
+                                       M, D, current_stream());
+  });
+  return out;
 }
 
 torch::Tensor swiglu_fwd(torch::Tensor x12) {
@@ -761,6 +810,9 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, mod) {
   mod.def("l2norm_bwd", &l2norm_bwd);
   mod.def("bias_gelu_fwd", &bias_gelu_fwd);
   mod.def("bias_gelu_bwd", &bias_gelu_bwd);
+  mod.def("row_gather", &row_gather);
+  mod.def("row_scatter_add_", &row_scatter_add_);
+  mod.def("row_gather_scaled", &row_gather_scaled);
   mod.def("ls_axpy_fwd", &ls_axpy_fwd);
   mod.def("ls_axpy_bwd", &ls_axpy_bwd);
   mod.def("swiglu_fwd", &swiglu_fwd);

@@ -5,12 +5,19 @@
 
 #define EW_BLOCK 256
 
+// fast tanh via the hardware exp pipe: tanh(y) = 1 - 2/(1 + exp(2y)).
+// tanhf() is a slow polyline in libm; __expf is one v_exp_f32 — the GELU
+// kernels were VALU-bound on tanhf in the ViT-L profile.
+DEV_INLINE float fast_tanh(float y) {
+  return 1.0f - 2.0f / (1.0f + __expf(2.0f * y));
+}
+
 DEV_INLINE float gelu_tanh(float x) {
   // tanh approximation (matches torch F.gelu(approximate="tanh"))
   const float k0 = 0.7978845608028654f;  // sqrt(2/pi)
   const float k1 = 0.044715f;
   float inner = k0 * (x + k1 * x * x * x);
-  return 0.5f * x * (1.0f + tanhf(inner));
+  return 0.5f * x * (1.0f + fast_tanh(inner));
 }
 
 DEV_INLINE float gelu_tanh_grad(float x) {
@@ -18,7 +25,7 @@ DEV_INLINE float gelu_tanh_grad(float x) {
   const float k1 = 0.044715f;
   float x2 = x * x;
   float inner = k0 * (x + k1 * x * x2);
-  float t = tanhf(inner);
+  float t = fast_tanh(inner);
   float sech2 = 1.0f - t * t;
   return 0.5f * (1.0f + t) + 0.5f * x * sech2 * k0 * (1.0f + 3.0f * k1 * x2);
 }
@@ -170,6 +177,73 @@ __global__ void swiglu_bwd_kernel(const T* __restrict__ dy, const T* __restrict_
   }
 }
 
+// -------------------- row gather / scatter-add (K11) --------------------
+// Stochastic-depth subset compute: gather kept samples' token rows, compute,
+// scatter-add the scaled residual back. Rows are disjoint (no atomics).
+
+template <typename T>
+__global__ void row_gather_kernel(const T* __restrict__ src, const long* __restrict__ idx,
+                                  T* __restrict__ out, long M, int D) {
+  const int d8 = D / 8;
+  long total = M * (long)d8;
+  for (long i = blockIdx.x * (long)blockDim.x + threadIdx.x; i < total;
+       i += (long)gridDim.x * blockDim.x) {
+    const long r = i / d8;
+    const int c8 = (int)(i % d8) * 8;
+    const long srow = idx[r];
+    T buf[8];
+    Vec8<T>::load(buf, src + srow * (long)D + c8);
+    Vec8<T>::store(out + r * (long)D + c8, buf);
+  }
+}
+
+template <typename T>
+__global__ void row_scatter_add_kernel(T* __restrict__ dst, const long* __restrict__ idx,
+                                       const T* __restrict__ src,
+                                       const float* __restrict__ scale, long M, int D) {
+  const int d8 = D / 8;
+  long total = M * (long)d8;
+  for (long i = blockIdx.x * (long)blockDim.x + threadIdx.x; i < total;
+       i += (long)gridDim.x * blockDim.x) {
+    const long r = i / d8;
+    const int c8 = (int)(i % d8) * 8;
+    const long drow = idx[r];
+    const float s = scale != nullptr ? scale[r] : 1.0f;
+    T sb[8], db[8];
+    Vec8<T>::load(sb, src + r * (long)D + c8);
+    Vec8<T>::load(db, dst + drow * (long)D + c8);
+#pragma unroll
+    for (int e = 0; e < 8; ++e) {
+      ScalarOps<T>::store(db + e,
+                          ScalarOps<T>::load(db + e) + s * ScalarOps<T>::load(sb + e));
+    }
+    Vec8<T>::store(dst + drow * (long)D + c8, db);
+  }
+}
+
+// gather with per-row scale (backward of scatter-add)
+template <typename T>
+__global__ void row_gather_scaled_kernel(const T* __restrict__ src,
+                                         const long* __restrict__ idx,
+                                         const float* __restrict__ scale,
+                                         T* __restrict__ out, long M, int D) {
+  const int d8 = D / 8;
+  long total = M * (long)d8;
+  for (long i = blockIdx.x * (long)blockDim.x + threadIdx.x; i < total;
+       i += (long)gridDim.x * blockDim.x) {
+    const long r = i / d8;
+    const int c8 = (int)(i % d8) * 8;
+    const long srow = idx[r];
+    const float s = scale != nullptr ? scale[r] : 1.0f;
+    T buf[8];
+    Vec8<T>::load(buf, src + srow * (long)D + c8);
+#pragma unroll
+    for (int e = 0; e < 8; ++e)
+      ScalarOps<T>::store(buf + e, s * ScalarOps<T>::load(buf + e));
+    Vec8<T>::store(out + r * (long)D + c8, buf);
+  }
+}
+
 // ------------------------------- RoPE ----------------------------------
 // x: [B, Hh, N, hd] contiguous; sin/cos: [P, hd] fp32 with P = N - prefix.
 // rotate-half: out[j] = x[j]*cos[j] - x[j+hd/2]*sin[j]          (j < hd/2)
@@ -245,6 +319,36 @@ void launch_ls_axpy_bwd(const T* dout, const T* res, const T* gamma, T* dres,
 }
 
 template <typename T>
+void launch_row_gather(const T* src, const long* idx, T* out, long M, int D,
+                       hipStream_t stream) {
+  long total = M * (long)(D / 8);
+  int grid = (int)min((total + EW_BLOCK - 1) / EW_BLOCK, (long)2048);
+  if (grid == 0) return;
+  hipLaunchKernelGGL((row_gather_kernel<T>), dim3(grid), dim3(EW_BLOCK), 0, stream, src,
+                     idx, out, M, D);
+}
+
+template <typename T>
+void launch_row_scatter_add(T* dst, const long* idx, const T* src, const float* scale,
+                            long M, int D, hipStream_t stream) {
+  long total = M * (long)(D / 8);
+  int grid = (int)min((total + EW_BLOCK - 1) / EW_BLOCK, (long)2048);
+  if (grid == 0) return;
+  hipLaunchKernelGGL((row_scatter_add_kernel<T>), dim3(grid), dim3(EW_BLOCK), 0, stream,
+                     dst, idx, src, scale, M, D);
+}
+
+template <typename T>
+void launch_row_gather_scaled(const T* src, const long* idx, const float* scale, T* out,
+                              long M, int D, hipStream_t stream) {
+  long total = M * (long)(D / 8);
+  int grid = (int)min((total + EW_BLOCK - 1) / EW_BLOCK, (long)2048);
+  if (grid == 0) return;
+  hipLaunchKernelGGL((row_gather_scaled_kernel<T>), dim3(grid), dim3(EW_BLOCK), 0, stream,
+                     src, idx, scale, out, M, D);
+}
+
+template <typename T>
 void launch_swiglu_fwd(const T* x12, T* y, long rows, int H, hipStream_t stream) {
   long total = rows * (long)H;
   int grid = (int)min((total + EW_BLOCK - 1) / EW_BLOCK, (long)2048);
@@ -277,6 +381,12 @@ void launch_rope_fwd(const T* x, const float* sin_t, const float* cos_t, T* y, l
                                       hipStream_t);                                  \
   template void launch_ls_axpy_bwd<T>(const T*, const T*, const T*, T*, float*,      \
                                       long, int, hipStream_t);                       \
+  template void launch_row_gather<T>(const T*, const long*, T*, long, int,           \
+                                     hipStream_t);                                   \
+  template void launch_row_scatter_add<T>(T*, const long*, const T*, const float*,   \
+                                          long, int, hipStream_t);                   \
+  template void launch_row_gather_scaled<T>(const T*, const long*, const float*, T*, \
+                                            long, int, hipStream_t);                 \
   template void launch_bias_gelu_bwd<T>(const T*, const T*, const T*, T*, float*,    \
                                         long, int, hipStream_t);                     \
   template void launch_swiglu_fwd<T>(const T*, T*, long, int, hipStream_t);          \

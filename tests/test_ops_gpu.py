@@ -232,3 +232,62 @@ def test_multi_tensor_l2norm_gpu(ops):
     got = ops.multi_tensor_l2norm_sq(g)
     want = sum(x.float().pow(2).sum() for x in g)
     _assert_close(got, want, atol=0.5, rtol=1e-2, what="l2norm_sq")
+
+
+def test_ls_axpy_fwd_bwd(ops):
+    torch.manual_seed(10)
+    R, D = 300, 1024
+    x = torch.randn(R, D, device=DEV).bfloat16().requires_grad_(True)
+    res = torch.randn(R, D, device=DEV).bfloat16().requires_grad_(True)
+    gamma = (torch.randn(D, device=DEV) * 0.01).bfloat16().requires_grad_(True)
+    from dinov3_amd.ops.ls_axpy import ls_axpy
+
+    out = ls_axpy(x, res, gamma)
+    xr = x.detach().float().requires_grad_(True)
+    rr = res.detach().float().requires_grad_(True)
+    gr = gamma.detach().float().requires_grad_(True)
+    ref = xr + gr * rr
+    _assert_close(out, ref, atol=0.03, what="ls_axpy fwd")
+    dy = torch.randn_like(out)
+    out.backward(dy)
+    ref.backward(dy.float())
+    _assert_close(x.grad, xr.grad, atol=0.02, what="ls_axpy dx")
+    _assert_close(res.grad, rr.grad, atol=0.02, what="ls_axpy dres")
+    _assert_close(gamma.grad, gr.grad, atol=0.3, rtol=2e-2, what="ls_axpy dgamma")
+
+
+def test_row_gather_scatter(ops):
+    torch.manual_seed(11)
+    R, D, M = 64, 256, 24
+    flat = torch.randn(R, D, device=DEV).bfloat16().requires_grad_(True)
+    idx = torch.randperm(R, device=DEV)[:M]
+    from dinov3_amd.ops.row_ops import gather_rows, scatter_add_rows
+
+    sub = gather_rows(flat, idx)
+    assert torch.equal(sub, flat.detach()[idx])
+    res = torch.randn(M, D, device=DEV).bfloat16().requires_grad_(True)
+    scale = torch.full((M,), 1.5, device=DEV)
+    out = scatter_add_rows(flat, idx, res, scale)
+    ref_in = flat.detach().float().requires_grad_(True)
+    res_ref = res.detach().float().requires_grad_(True)
+    ref = ref_in.index_add(0, idx, res_ref * 1.5)
+    _assert_close(out, ref, atol=0.05, what="scatter fwd")
+    dy = torch.randn_like(out)
+    out.backward(dy)
+    ref.backward(dy.float())
+    _assert_close(flat.grad, ref_in.grad, atol=0.02, what="scatter dflat")
+    _assert_close(res.grad, res_ref.grad, atol=0.03, what="scatter dres")
+
+
+def test_gather_rows_backward(ops):
+    torch.manual_seed(12)
+    R, D, M = 32, 64, 10
+    flat = torch.randn(R, D, device=DEV).bfloat16().requires_grad_(True)
+    idx = torch.randperm(R, device=DEV)[:M]
+    from dinov3_amd.ops.row_ops import gather_rows
+
+    sub = gather_rows(flat, idx)
+    (sub.float() ** 2).sum().backward()
+    ref = torch.zeros_like(flat, dtype=torch.float32)
+    ref[idx] = 2 * flat.detach()[idx].float()
+    _assert_close(flat.grad, ref, atol=0.05, what="gather dflat")
