@@ -72,6 +72,41 @@ def _make_sampler(*, dataset, type: Optional[SamplerType], shuffle: bool, seed: 
     return None
 
 
+class CombinedDataLoader:
+    """Sample batches from several loaders with given probabilities.
+
+    The reference references a missing `CombineDataLoader` for its
+    multi-resolution crop schedules (train.py:763) — this is
+    the working equivalent: every __iter__ draws a loader by ratio, yielding
+    its next batch (each batch is internally one resolution).
+    """
+
+    def __init__(self, loaders, ratios=None, seed: int = 0):
+        import numpy as np
+
+        self.loaders = list(loaders)
+        if ratios is None:
+            ratios = [1.0] * len(self.loaders)
+        total = float(sum(ratios))
+        self.ratios = [r / total for r in ratios]
+        self._rng = np.random.default_rng(seed)
+
+    def __iter__(self):
+        import numpy as np
+
+        iters = [iter(dl) for dl in self.loaders]
+        while True:
+            choice = int(self._rng.choice(len(iters), p=self.ratios))
+            try:
+                yield next(iters[choice])
+            except StopIteration:
+                iters[choice] = iter(self.loaders[choice])
+                yield next(iters[choice])
+
+    def __len__(self):
+        return sum(len(dl) for dl in self.loaders)
+
+
 def make_data_loader(
     *,
     dataset,

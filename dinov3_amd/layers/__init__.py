@@ -1,4 +1,5 @@
 from .attention import LinearKMaskedBias, SelfAttention
+from .causal_attention import CausalSelfAttention, CausalSelfAttentionBlock
 from .block import SelfAttentionBlock
 from .dino_head import DINOHead
 from .ffn_layers import FFN_LAYERS, Mlp, SwiGLUFFN
@@ -8,6 +9,8 @@ from .rope import RopePositionEmbedding
 
 __all__ = [
     "SelfAttention",
+    "CausalSelfAttention",
+    "CausalSelfAttentionBlock",
     "LinearKMaskedBias",
     "SelfAttentionBlock",
     "DINOHead",

@@ -6,6 +6,7 @@ import logging
 
 from . import vision_transformer as vits
 from .vision_transformer import DinoVisionTransformer  # noqa: F401
+from .convnext import ConvNeXt  # noqa: F401
 
 logger = logging.getLogger("dinov3")
 

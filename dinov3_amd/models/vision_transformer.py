@@ -115,7 +115,12 @@ class DinoVisionTransformer(nn.Module):
         self.norm = norm_factory(embed_dim)
         self.cls_norm = norm_factory(embed_dim) if untie_cls_and_patch_norms else None
         self.local_cls_norm = norm_factory(embed_dim) if untie_global_and_local_cls_norm else None
+        self.grad_checkpointing = False  # selective recompute per block (288 GB
+        # HBM makes this OFF by default up to ViT-g; flip for 7B-class runs)
         self._init_weights()
+
+    def set_grad_checkpointing(self, enabled: bool = True) -> None:
+        self.grad_checkpointing = enabled
 
     def _init_weights(self) -> None:
         for m in self.modules():
@@ -183,8 +188,14 @@ class DinoVisionTransformer(nn.Module):
             metas.append(SelfAttention._meta_for(t, rope_tables[key], off))
             off += t.shape[0] * t.shape[1]
 
-        for block in self.blocks:
-            flat = block.forward_flat(flat, metas)
+        if self.grad_checkpointing and self.training and torch.is_grad_enabled():
+            from torch.utils.checkpoint import checkpoint
+
+            for block in self.blocks:
+                flat = checkpoint(block.forward_flat, flat, metas, use_reentrant=False)
+        else:
+            for block in self.blocks:
+                flat = block.forward_flat(flat, metas)
         tokens = uncat_with_shapes(flat, shapes, counts)
 
         output = []

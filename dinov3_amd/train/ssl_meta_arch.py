@@ -99,6 +99,12 @@ class SSLMetaArch(nn.Module):
         else:
             self.gram_backbone = None
 
+        # activation checkpointing (selective per-block recompute) on request
+        if config.train.checkpointing or config.train.checkpointing_full:
+            if hasattr(self.student_backbone, "set_grad_checkpointing"):
+                self.student_backbone.set_grad_checkpointing(True)
+                logger.info("activation checkpointing enabled on the student backbone")
+
         # the teacher tower never takes gradients and starts as a copy of the student
         self._sync_teacher_from_student()
         for module in (self.teacher_backbone, self.teacher_dino_head, self.teacher_ibot_head):

@@ -141,6 +141,32 @@ def build_training_engine(cfg, params_groups):
     return optimizer, reducer.finalize
 
 
+def build_multi_resolution_data_loader_from_cfg(cfg, model: SSLMetaArch, sampler_advance: int = 0):
+    """Multi-resolution crop schedules: crops.global_crops_size may be a list
+    of sizes (paired with local sizes + ratios) — one loader per resolution,
+    combined by sampling ratio (reference train.py:718-769)."""
+    g_sizes = cfg.crops.global_crops_size
+    if isinstance(g_sizes, int):
+        return build_data_loader_from_cfg(cfg, model, sampler_advance)
+    from ..data.loaders import CombinedDataLoader
+    import copy
+
+    l_sizes = cfg.crops.local_crops_size
+    if isinstance(l_sizes, int):
+        l_sizes = [l_sizes] * len(g_sizes)
+    ratios = cfg.crops.global_local_crop_pairs_ratios
+    if isinstance(ratios, (int, float)):
+        ratios = [float(ratios)] * len(g_sizes)
+    loaders = []
+    for gs, ls in zip(g_sizes, l_sizes):
+        sub = copy.deepcopy(cfg)
+        sub.crops.global_crops_size = gs
+        sub.crops.local_crops_size = ls
+        loaders.append(build_data_loader_from_cfg(sub, model, sampler_advance))
+    logger.info("multi-resolution loader: sizes %s ratios %s", list(g_sizes), ratios)
+    return CombinedDataLoader(loaders, ratios, seed=cfg.train.seed)
+
+
 def build_data_loader_from_cfg(cfg, model: SSLMetaArch, sampler_advance: int = 0):
     img_size = cfg.crops.global_crops_size
     patch_size = cfg.student.patch_size
@@ -220,7 +246,7 @@ def do_train(cfg, model: SSLMetaArch, resume: bool = True, max_iterations: int =
             start_iter = payload["iteration"] + 1
             logger.info("resumed at iteration %d", start_iter)
 
-    data_loader = build_data_loader_from_cfg(cfg, model)
+    data_loader = build_multi_resolution_data_loader_from_cfg(cfg, model)
     total_iterations = schedulers["total_iterations"]
     if max_iterations > 0:
         total_iterations = min(total_iterations, start_iter + max_iterations)
@@ -320,7 +346,35 @@ def do_train(cfg, model: SSLMetaArch, resume: bool = True, max_iterations: int =
 
 
 def do_test(cfg, model, iteration: int):
-    raise NotImplementedError("eval harness lands with the eval subsystem")
+    """k-NN + linear-probe eval of the teacher backbone on the configured
+    dataset (synthetic decode in this offline environment)."""
+    from ..data import SamplerType, make_data_loader, make_dataset
+    from ..data.transforms import make_eval_transform
+    from ..eval import evaluate_knn, evaluate_linear_probe, extract_features
+
+    device = parallel.device()
+    backbone = model.teacher_backbone if hasattr(model, "teacher_backbone") else model
+    backbone = backbone.to(device)
+    transform = make_eval_transform(crop_size=cfg.crops.global_crops_size,
+                                    mean=cfg.crops.rgb_mean, std=cfg.crops.rgb_std)
+    base = cfg.train.dataset_path.split(":")[0]
+    train_ds = make_dataset(dataset_str=f"{base}:split=TRAIN:length=2048"
+                            if base == "Synthetic" else f"{base}:split=TRAIN",
+                            transform=transform)
+    val_ds = make_dataset(dataset_str=f"{base}:split=VAL:length=512"
+                          if base == "Synthetic" else f"{base}:split=VAL",
+                          transform=transform)
+    kwargs = dict(batch_size=64, num_workers=cfg.train.num_workers, shuffle=False,
+                  sampler_type=SamplerType.EPOCH, drop_last=False)
+    train_feats, train_labels = extract_features(backbone, make_data_loader(dataset=train_ds, **kwargs))
+    val_feats, val_labels = extract_features(backbone, make_data_loader(dataset=val_ds, **kwargs))
+    results = {
+        "iteration": iteration,
+        "knn_top1": evaluate_knn(train_feats, train_labels, val_feats, val_labels),
+        "linear_top1": evaluate_linear_probe(train_feats, train_labels, val_feats, val_labels),
+    }
+    logger.info("eval results: %s", results)
+    return results
 
 
 def main(argv=None):
