@@ -108,7 +108,7 @@ __global__ __launch_bounds__(256) void fwd_kernel(
   constexpr int DTILES = HD / 32;
   constexpr int KVB = 32;
   constexpr int LDS_STRIDE = HD + 8;
-  constexpr int VT_STRIDE = KVB + 8;
+  constexpr int VT_STRIDE = 2 * KVB + 8;
   constexpr int HALF = HD / 2;
 
   const int bh = blockIdx.x;
@@ -129,7 +129,7 @@ __global__ __launch_bounds__(256) void fwd_kernel(
 
   extern __shared__ __attribute__((aligned(16))) char smem_raw[];
   __hip_bfloat16* k_lds = reinterpret_cast<__hip_bfloat16*>(smem_raw);
-  __hip_bfloat16* vt_lds = k_lds + KVB * LDS_STRIDE;
+  __hip_bfloat16* vt_lds = k_lds + 2 * KVB * LDS_STRIDE;
 
   // Q fragments (+ rope)
   bf16x8 qf[KSLICES];
@@ -156,18 +156,18 @@ __global__ __launch_bounds__(256) void fwd_kernel(
     for (int r = 0; r < 16; ++r) o_acc[t][r] = 0.f;
   float m_run = -INFINITY, l_run = 0.f;
 
-  const int n_kv = (N + KVB - 1) / KVB;
-  for (int kt = 0; kt < n_kv; ++kt) {
-    const int kbase = kt * KVB;
+  const int n_super = (N + 2 * KVB - 1) / (2 * KVB);  // 64 keys per barrier pair
+  for (int kt = 0; kt < n_super; ++kt) {
+    const int kbase0 = kt * 2 * KVB;
     __syncthreads();
     {
       // K staging with rope: each thread owns a (row, pair-chunk) — loads the
       // lo/hi halves, rotates, writes both into k_lds.
       constexpr int PAIRS_PER_ROW = HALF / 8;
-      for (int idx = threadIdx.x; idx < KVB * PAIRS_PER_ROW; idx += 256) {
+      for (int idx = threadIdx.x; idx < 2 * KVB * PAIRS_PER_ROW; idx += 256) {
         const int row = idx / PAIRS_PER_ROW;
         const int c0 = (idx % PAIRS_PER_ROW) * 8;
-        const int krow = kbase + row;
+        const int krow = kbase0 + row;
         bf16x8 lo{}, hi{};
         if (krow < N) {
           lo = load8(qv.at(krow, 1, c0));
@@ -182,10 +182,10 @@ __global__ __launch_bounds__(256) void fwd_kernel(
       }
       // V transposed staging
       constexpr int PER_ROW = HD / 8;
-      for (int idx = threadIdx.x; idx < KVB * PER_ROW; idx += 256) {
+      for (int idx = threadIdx.x; idx < 2 * KVB * PER_ROW; idx += 256) {
         const int row = idx / PER_ROW;
         const int c8 = (idx % PER_ROW) * 8;
-        const int krow = kbase + row;
+        const int krow = kbase0 + row;
         bf16x8 vv = (krow < N) ? load8(qv.at(krow, 2, c8)) : bf16x8{};
 #pragma unroll
         for (int e = 0; e < 8; ++e)
@@ -194,10 +194,14 @@ __global__ __launch_bounds__(256) void fwd_kernel(
     }
     __syncthreads();
 
+   for (int sub = 0; sub < 2 && kbase0 + sub * KVB < N; ++sub) {
+    const int kbase = kbase0 + sub * KVB;
+    const int krow_off = sub * KVB;
+
     f32x16 s_acc = {};
 #pragma unroll
     for (int s = 0; s < KSLICES; ++s) {
-      bf16x8 af = load8(&k_lds[l31 * LDS_STRIDE + s * 16 + hhalf * 8]);
+      bf16x8 af = load8(&k_lds[(krow_off + l31) * LDS_STRIDE + s * 16 + hhalf * 8]);
       s_acc = MFMA32(af, qf[s], s_acc);
     }
     float sv[16];
@@ -230,8 +234,8 @@ __global__ __launch_bounds__(256) void fwd_kernel(
     bf16x8 p1 = pack_fragment(sv, 8);
 #pragma unroll
     for (int t = 0; t < DTILES; ++t) {
-      bf16x8 a0 = load8(&vt_lds[(t * 32 + l31) * VT_STRIDE + hhalf * 8]);
-      bf16x8 a1 = load8(&vt_lds[(t * 32 + l31) * VT_STRIDE + 16 + hhalf * 8]);
+      bf16x8 a0 = load8(&vt_lds[(t * 32 + l31) * VT_STRIDE + krow_off + hhalf * 8]);
+      bf16x8 a1 = load8(&vt_lds[(t * 32 + l31) * VT_STRIDE + krow_off + 16 + hhalf * 8]);
       f32x16 acc;
 #pragma unroll
       for (int r = 0; r < 16; ++r) acc[r] = o_acc[t][r];
@@ -240,6 +244,7 @@ __global__ __launch_bounds__(256) void fwd_kernel(
 #pragma unroll
       for (int r = 0; r < 16; ++r) o_acc[t][r] = acc[r];
     }
+   }
   }
 
   const float inv_l = 1.0f / l_run;
@@ -271,7 +276,7 @@ __global__ __launch_bounds__(256) void bwd_dq_kernel(
   constexpr int DTILES = HD / 32;
   constexpr int KVB = 32;
   constexpr int LDS_STRIDE = HD + 8;
-  constexpr int KT_STRIDE = KVB + 8;
+  constexpr int KT_STRIDE = 2 * KVB + 8;
   constexpr int HALF = HD / 2;
 
   const int bh = blockIdx.x;
@@ -294,8 +299,8 @@ __global__ __launch_bounds__(256) void bwd_dq_kernel(
 
   extern __shared__ __attribute__((aligned(16))) char smem_raw[];
   __hip_bfloat16* k_lds = reinterpret_cast<__hip_bfloat16*>(smem_raw);   // rope'd K
-  __hip_bfloat16* v_lds = k_lds + KVB * LDS_STRIDE;
-  __hip_bfloat16* kt_lds = v_lds + KVB * LDS_STRIDE;                      // rope'd K^T
+  __hip_bfloat16* v_lds = k_lds + 2 * KVB * LDS_STRIDE;
+  __hip_bfloat16* kt_lds = v_lds + 2 * KVB * LDS_STRIDE;                  // rope'd K^T
 
   const int qrow = q0 + l31;
   const int safe = qrow < N ? qrow : (N - 1);
@@ -324,16 +329,16 @@ __global__ __launch_bounds__(256) void bwd_dq_kernel(
 #pragma unroll
     for (int r = 0; r < 16; ++r) dq_acc[t][r] = 0.f;
 
-  const int n_kv = (N + KVB - 1) / KVB;
-  for (int kt = 0; kt < n_kv; ++kt) {
-    const int kbase = kt * KVB;
+  const int n_super = (N + 2 * KVB - 1) / (2 * KVB);
+  for (int kt = 0; kt < n_super; ++kt) {
+    const int kbase0 = kt * 2 * KVB;
     __syncthreads();
     {
       constexpr int PAIRS_PER_ROW = HALF / 8;
-      for (int idx = threadIdx.x; idx < KVB * PAIRS_PER_ROW; idx += 256) {
+      for (int idx = threadIdx.x; idx < 2 * KVB * PAIRS_PER_ROW; idx += 256) {
         const int row = idx / PAIRS_PER_ROW;
         const int c0 = (idx % PAIRS_PER_ROW) * 8;
-        const int krow = kbase + row;
+        const int krow = kbase0 + row;
         bf16x8 lo{}, hi{};
         if (krow < N) {
           lo = load8(qv.at(krow, 1, c0));
@@ -351,21 +356,24 @@ __global__ __launch_bounds__(256) void bwd_dq_kernel(
         }
       }
       constexpr int PER_ROW = HD / 8;
-      for (int idx = threadIdx.x; idx < KVB * PER_ROW; idx += 256) {
+      for (int idx = threadIdx.x; idx < 2 * KVB * PER_ROW; idx += 256) {
         const int row = idx / PER_ROW;
         const int c8 = (idx % PER_ROW) * 8;
-        const int krow = kbase + row;
+        const int krow = kbase0 + row;
         bf16x8 vv = (krow < N) ? load8(qv.at(krow, 2, c8)) : bf16x8{};
         *reinterpret_cast<bf16x8*>(&v_lds[row * LDS_STRIDE + c8]) = vv;
       }
     }
     __syncthreads();
 
+   for (int sub = 0; sub < 2 && kbase0 + sub * KVB < N; ++sub) {
+    const int kbase = kbase0 + sub * KVB;
+    const int krow_off = sub * KVB;
     f32x16 s_acc = {}, dp_acc = {};
 #pragma unroll
     for (int s = 0; s < KSLICES; ++s) {
-      bf16x8 kf = load8(&k_lds[l31 * LDS_STRIDE + s * 16 + hhalf * 8]);
-      bf16x8 vf = load8(&v_lds[l31 * LDS_STRIDE + s * 16 + hhalf * 8]);
+      bf16x8 kf = load8(&k_lds[(krow_off + l31) * LDS_STRIDE + s * 16 + hhalf * 8]);
+      bf16x8 vf = load8(&v_lds[(krow_off + l31) * LDS_STRIDE + s * 16 + hhalf * 8]);
       s_acc = MFMA32(kf, qf[s], s_acc);
       dp_acc = MFMA32(vf, dof[s], dp_acc);
     }
@@ -380,8 +388,8 @@ __global__ __launch_bounds__(256) void bwd_dq_kernel(
     bf16x8 f1 = pack_fragment(ds, 8);
 #pragma unroll
     for (int t = 0; t < DTILES; ++t) {
-      bf16x8 a0 = load8(&kt_lds[(t * 32 + l31) * KT_STRIDE + hhalf * 8]);
-      bf16x8 a1 = load8(&kt_lds[(t * 32 + l31) * KT_STRIDE + 16 + hhalf * 8]);
+      bf16x8 a0 = load8(&kt_lds[(t * 32 + l31) * KT_STRIDE + krow_off + hhalf * 8]);
+      bf16x8 a1 = load8(&kt_lds[(t * 32 + l31) * KT_STRIDE + krow_off + 16 + hhalf * 8]);
       f32x16 acc;
 #pragma unroll
       for (int r = 0; r < 16; ++r) acc[r] = dq_acc[t][r];
@@ -390,6 +398,7 @@ __global__ __launch_bounds__(256) void bwd_dq_kernel(
 #pragma unroll
       for (int r = 0; r < 16; ++r) dq_acc[t][r] = acc[r];
     }
+   }
   }
 
   if (qrow < N) {
@@ -487,17 +496,17 @@ __global__ __launch_bounds__(256) void bwd_dkv_kernel(
       dv_acc[t][r] = 0.f;
     }
 
-  const int n_q = (N + QB - 1) / QB;
-  for (int qt = 0; qt < n_q; ++qt) {
-    const int qbase = qt * QB;
+  const int n_super = (N + 2 * QB - 1) / (2 * QB);
+  for (int qt = 0; qt < n_super; ++qt) {
+    const int qbase0 = qt * 2 * QB;
     __syncthreads();
     {
       // Q^T staging with rope (pair chunks), dO^T staging
       constexpr int PAIRS_PER_ROW = HALF / 8;
-      for (int idx = threadIdx.x; idx < QB * PAIRS_PER_ROW; idx += 256) {
+      for (int idx = threadIdx.x; idx < 2 * QB * PAIRS_PER_ROW; idx += 256) {
         const int row = idx / PAIRS_PER_ROW;
         const int c0 = (idx % PAIRS_PER_ROW) * 8;
-        const int qrow = qbase + row;
+        const int qrow = qbase0 + row;
         bf16x8 lo{}, hi{};
         if (qrow < N) {
           lo = load8(qv.at(qrow, 0, c0));
@@ -513,23 +522,26 @@ __global__ __launch_bounds__(256) void bwd_dkv_kernel(
         }
       }
       constexpr int PER_ROW = HD / 8;
-      for (int idx = threadIdx.x; idx < QB * PER_ROW; idx += 256) {
+      for (int idx = threadIdx.x; idx < 2 * QB * PER_ROW; idx += 256) {
         const int row = idx / PER_ROW;
         const int c8 = (idx % PER_ROW) * 8;
-        const int qrow = qbase + row;
+        const int qrow = qbase0 + row;
         bf16x8 dov = (qrow < N) ? load8(do_base + (long)qrow * do_stride + c8) : bf16x8{};
 #pragma unroll
         for (int e = 0; e < 8; ++e)
           dot_lds[(c8 + e) * QT_STRIDE + row] = reinterpret_cast<__hip_bfloat16*>(&dov)[e];
       }
-      for (int row = threadIdx.x; row < QB; row += 256) {
-        const int qrow = qbase + row;
+      for (int row = threadIdx.x; row < 2 * QB; row += 256) {
+        const int qrow = qbase0 + row;
         lse_lds[row] = (qrow < N) ? lse[((long)b * H + h) * N + qrow] : INFINITY;
         d_lds[row] = (qrow < N) ? D[((long)b * H + h) * N + qrow] : 0.f;
       }
     }
     __syncthreads();
 
+   for (int sub = 0; sub < 2 && qbase0 + sub * QB < N; ++sub) {
+    const int qbase = qbase0 + sub * QB;
+    const int qrow_off = sub * QB;
     f32x16 s_acc = {}, dp_acc = {};
 #pragma unroll
     for (int s = 0; s < KSLICES; ++s) {
@@ -537,8 +549,8 @@ __global__ __launch_bounds__(256) void bwd_dkv_kernel(
 #pragma unroll
       for (int e = 0; e < 8; ++e) {
         const int d = s * 16 + hhalf * 8 + e;
-        reinterpret_cast<__hip_bfloat16*>(&aq)[e] = qt_lds[d * QT_STRIDE + l31];
-        reinterpret_cast<__hip_bfloat16*>(&ad)[e] = dot_lds[d * QT_STRIDE + l31];
+        reinterpret_cast<__hip_bfloat16*>(&aq)[e] = qt_lds[d * QT_STRIDE + qrow_off + l31];
+        reinterpret_cast<__hip_bfloat16*>(&ad)[e] = dot_lds[d * QT_STRIDE + qrow_off + l31];
       }
       s_acc = MFMA32(aq.v8, kf[s], s_acc);
       dp_acc = MFMA32(ad.v8, vf[s], dp_acc);
@@ -548,10 +560,10 @@ __global__ __launch_bounds__(256) void bwd_dkv_kernel(
     for (int r = 0; r < 16; ++r) {
       const int qrow = qbase + c_row(r, hhalf);
       const bool valid = (qrow < N) && (krow < N);
-      const float l = lse_lds[c_row(r, hhalf)];
+      const float l = lse_lds[qrow_off + c_row(r, hhalf)];
       float p = valid ? __expf(s_acc[r] * scale - l) : 0.f;
       pv[r] = p;
-      ds[r] = p * (dp_acc[r] - d_lds[c_row(r, hhalf)]) * scale;
+      ds[r] = p * (dp_acc[r] - d_lds[qrow_off + c_row(r, hhalf)]) * scale;
     }
     bf16x8 p0 = pack_fragment(pv, 0);
     bf16x8 p1 = pack_fragment(pv, 8);
@@ -562,10 +574,10 @@ __global__ __launch_bounds__(256) void bwd_dkv_kernel(
       union { unsigned u[4]; bf16x8 v8; } ado0, ado1, aq0, aq1;
 #pragma unroll
       for (int e = 0; e < 8; ++e) {
-        reinterpret_cast<__hip_bfloat16*>(&ado0)[e] = dot_lds[(t * 32 + l31) * QT_STRIDE + hhalf * 8 + e];
-        reinterpret_cast<__hip_bfloat16*>(&ado1)[e] = dot_lds[(t * 32 + l31) * QT_STRIDE + 16 + hhalf * 8 + e];
-        reinterpret_cast<__hip_bfloat16*>(&aq0)[e] = qt_lds[(t * 32 + l31) * QT_STRIDE + hhalf * 8 + e];
-        reinterpret_cast<__hip_bfloat16*>(&aq1)[e] = qt_lds[(t * 32 + l31) * QT_STRIDE + 16 + hhalf * 8 + e];
+        reinterpret_cast<__hip_bfloat16*>(&ado0)[e] = dot_lds[(t * 32 + l31) * QT_STRIDE + qrow_off + hhalf * 8 + e];
+        reinterpret_cast<__hip_bfloat16*>(&ado1)[e] = dot_lds[(t * 32 + l31) * QT_STRIDE + qrow_off + 16 + hhalf * 8 + e];
+        reinterpret_cast<__hip_bfloat16*>(&aq0)[e] = qt_lds[(t * 32 + l31) * QT_STRIDE + qrow_off + hhalf * 8 + e];
+        reinterpret_cast<__hip_bfloat16*>(&aq1)[e] = qt_lds[(t * 32 + l31) * QT_STRIDE + qrow_off + 16 + hhalf * 8 + e];
       }
       f32x16 accv, acck;
 #pragma unroll
@@ -583,6 +595,7 @@ __global__ __launch_bounds__(256) void bwd_dkv_kernel(
         dk_acc[t][r] = acck[r];
       }
     }
+   }
   }
 
   if (krow < N) {
@@ -659,11 +672,11 @@ void launch_fmha_rope_fwd(const __hip_bfloat16* qkv, const float* sin_t,
                           int N, int P, int HD, float scale, hipStream_t stream) {
   dim3 grid(B * H, (N + 127) / 128);
   if (HD == 64) {
-    size_t shmem = (32 * 72 + 64 * 40) * sizeof(__hip_bfloat16);
+    size_t shmem = (64 * 72 + 64 * 72) * sizeof(__hip_bfloat16);
     hipLaunchKernelGGL((fmha_rope::fwd_kernel<64>), grid, dim3(256), shmem, stream, qkv,
                        sin_t, cos_t, o, lse, B, H, N, P, scale);
   } else if (HD == 128) {
-    size_t shmem = (32 * 136 + 128 * 40) * sizeof(__hip_bfloat16);
+    size_t shmem = (64 * 136 + 128 * 72) * sizeof(__hip_bfloat16);
     hipLaunchKernelGGL((fmha_rope::fwd_kernel<128>), grid, dim3(256), shmem, stream, qkv,
                        sin_t, cos_t, o, lse, B, H, N, P, scale);
   }
@@ -683,11 +696,11 @@ void launch_fmha_rope_bwd_dq(const __hip_bfloat16* qkv, const __hip_bfloat16* do
                              int P, int HD, float scale, hipStream_t stream) {
   dim3 grid(B * H, (N + 127) / 128);
   if (HD == 64) {
-    size_t shmem = (2 * 32 * 72 + 64 * 40) * sizeof(__hip_bfloat16);
+    size_t shmem = (2 * 64 * 72 + 64 * 72) * sizeof(__hip_bfloat16);
     hipLaunchKernelGGL((fmha_rope::bwd_dq_kernel<64>), grid, dim3(256), shmem, stream, qkv,
                        dout, sin_t, cos_t, lse, D, dqkv, B, H, N, P, scale);
   } else if (HD == 128) {
-    size_t shmem = (2 * 32 * 136 + 128 * 40) * sizeof(__hip_bfloat16);
+    size_t shmem = (2 * 64 * 136 + 128 * 72) * sizeof(__hip_bfloat16);
     hipLaunchKernelGGL((fmha_rope::bwd_dq_kernel<128>), grid, dim3(256), shmem, stream,
                        qkv, dout, sin_t, cos_t, lse, D, dqkv, B, H, N, P, scale);
   }
@@ -699,11 +712,11 @@ void launch_fmha_rope_bwd_dkv(const __hip_bfloat16* qkv, const __hip_bfloat16* d
                               int P, int HD, float scale, hipStream_t stream) {
   dim3 grid(B * H, (N + 127) / 128);
   if (HD == 64) {
-    size_t shmem = 2 * 64 * 40 * sizeof(__hip_bfloat16) + 64 * sizeof(float);
+    size_t shmem = 2 * 64 * 72 * sizeof(__hip_bfloat16) + 128 * sizeof(float);
     hipLaunchKernelGGL((fmha_rope::bwd_dkv_kernel<64>), grid, dim3(256), shmem, stream,
                        qkv, dout, sin_t, cos_t, lse, D, dqkv, B, H, N, P, scale);
   } else if (HD == 128) {
-    size_t shmem = 2 * 128 * 40 * sizeof(__hip_bfloat16) + 64 * sizeof(float);
+    size_t shmem = 2 * 128 * 72 * sizeof(__hip_bfloat16) + 128 * sizeof(float);
     hipLaunchKernelGGL((fmha_rope::bwd_dkv_kernel<128>), grid, dim3(256), shmem, stream,
                        qkv, dout, sin_t, cos_t, lse, D, dqkv, B, H, N, P, scale);
   }
