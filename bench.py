@@ -24,6 +24,17 @@ import torch
 REPO_ROOT = os.path.dirname(os.path.abspath(__file__))
 sys.path.insert(0, REPO_ROOT)
 
+# Pre-tuned hipBLASLt algorithm selections for the ViT-L step's GEMM shapes
+# (PyTorch TunableOp, tuned on MI355X — +4% step time). Tuning itself stays
+# off; delete the env vars to disable.
+_TUNED_DIR = os.path.join(REPO_ROOT, "dinov3_amd", "tunableop")
+if os.path.isdir(_TUNED_DIR):
+    # torch appends the device ordinal before .csv: ship tunableop{0..7}.csv
+    os.environ.setdefault("PYTORCH_TUNABLEOP_ENABLED", "1")
+    os.environ.setdefault("PYTORCH_TUNABLEOP_TUNING", "0")
+    os.environ.setdefault("PYTORCH_TUNABLEOP_FILENAME",
+                          os.path.join(_TUNED_DIR, "tunableop.csv"))
+
 BASELINE_IMG_PER_SEC_PER_GPU = 2048 / 0.57 / 32  # Meta RSC anchor: 0.57 s/iter @ 2048 global batch, 32 GPUs
 
 

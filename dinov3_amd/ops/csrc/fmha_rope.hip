@@ -496,14 +496,14 @@ __global__ __launch_bounds__(256) void bwd_dkv_kernel(
       dv_acc[t][r] = 0.f;
     }
 
-  const int n_super = (N + 2 * QB - 1) / (2 * QB);
-  for (int qt = 0; qt < n_super; ++qt) {
-    const int qbase0 = qt * 2 * QB;
+  const int n_q = (N + QB - 1) / QB;
+  for (int qt = 0; qt < n_q; ++qt) {
+    const int qbase0 = qt * QB;
     __syncthreads();
     {
       // Q^T staging with rope (pair chunks), dO^T staging
       constexpr int PAIRS_PER_ROW = HALF / 8;
-      for (int idx = threadIdx.x; idx < 2 * QB * PAIRS_PER_ROW; idx += 256) {
+      for (int idx = threadIdx.x; idx < QB * PAIRS_PER_ROW; idx += 256) {
         const int row = idx / PAIRS_PER_ROW;
         const int c0 = (idx % PAIRS_PER_ROW) * 8;
         const int qrow = qbase0 + row;
@@ -522,7 +522,7 @@ __global__ __launch_bounds__(256) void bwd_dkv_kernel(
         }
       }
       constexpr int PER_ROW = HD / 8;
-      for (int idx = threadIdx.x; idx < 2 * QB * PER_ROW; idx += 256) {
+      for (int idx = threadIdx.x; idx < QB * PER_ROW; idx += 256) {
         const int row = idx / PER_ROW;
         const int c8 = (idx % PER_ROW) * 8;
         const int qrow = qbase0 + row;
@@ -531,7 +531,7 @@ __global__ __launch_bounds__(256) void bwd_dkv_kernel(
         for (int e = 0; e < 8; ++e)
           dot_lds[(c8 + e) * QT_STRIDE + row] = reinterpret_cast<__hip_bfloat16*>(&dov)[e];
       }
-      for (int row = threadIdx.x; row < 2 * QB; row += 256) {
+      for (int row = threadIdx.x; row < QB; row += 256) {
         const int qrow = qbase0 + row;
         lse_lds[row] = (qrow < N) ? lse[((long)b * H + h) * N + qrow] : INFINITY;
         d_lds[row] = (qrow < N) ? D[((long)b * H + h) * N + qrow] : 0.f;
@@ -539,9 +539,9 @@ __global__ __launch_bounds__(256) void bwd_dkv_kernel(
     }
     __syncthreads();
 
-   for (int sub = 0; sub < 2 && qbase0 + sub * QB < N; ++sub) {
-    const int qbase = qbase0 + sub * QB;
-    const int qrow_off = sub * QB;
+   {
+    const int qbase = qbase0;
+    const int qrow_off = 0;
     f32x16 s_acc = {}, dp_acc = {};
 #pragma unroll
     for (int s = 0; s < KSLICES; ++s) {
