@@ -32,7 +32,13 @@ class DINOLoss(nn.Module):
 
     @torch.no_grad()
     def sinkhorn_knopp_teacher(self, teacher_output: torch.Tensor, teacher_temp: float,
-                               n_iterations: int = 3) -> torch.Tensor:
+                               n_iterations: int = 3):
+        from ..ops import use_hip
+        from ..ops.proto_scores import sinkhorn_knopp_factored
+
+        if use_hip(teacher_output) and teacher_output.dtype == torch.bfloat16:
+            # factored form: the CE kernel consumes exp(x/T)*u*v lazily
+            return sinkhorn_knopp_factored(teacher_output, teacher_temp, n_iterations)
         return sinkhorn_knopp(teacher_output, teacher_temp, n_iterations=n_iterations)
 
     def forward(self, student_logits: torch.Tensor, teacher_probs: torch.Tensor,

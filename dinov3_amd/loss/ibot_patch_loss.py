@@ -32,7 +32,14 @@ class iBOTPatchLoss(nn.Module):
 
     @torch.no_grad()
     def sinkhorn_knopp_teacher(self, teacher_output: torch.Tensor, teacher_temp: float,
-                               n_masked_patches_tensor: torch.Tensor, n_iterations: int = 3) -> torch.Tensor:
+                               n_masked_patches_tensor: torch.Tensor, n_iterations: int = 3):
+        from ..ops import use_hip
+        from ..ops.proto_scores import sinkhorn_knopp_factored
+
+        if use_hip(teacher_output) and teacher_output.dtype == torch.bfloat16:
+            # factored form (the per-iteration B scalar cancels in the final
+            # row-normalize, so the global masked count is not needed)
+            return sinkhorn_knopp_factored(teacher_output, teacher_temp, n_iterations)
         B = n_masked_patches_tensor.clone().float()
         if B.ndim > 0:
             B = B.sum()

@@ -61,6 +61,24 @@ void launch_ibot_ce_bwd(const __hip_bfloat16*, const float*, const float*, const
                         const float*, const float*, __hip_bfloat16*, int, long, float,
                         hipStream_t);
 void launch_sinkhorn_exp(const __hip_bfloat16*, float*, float*, long, float, hipStream_t);
+void launch_sinkhorn_fact_colsum(const __hip_bfloat16*, const float*, float*, int, long,
+                                 float, hipStream_t);
+void launch_sinkhorn_fact_rowsum(const __hip_bfloat16*, const float*, float*, int, long,
+                                 float, hipStream_t);
+void launch_ibot_ce_fact_fwd(const __hip_bfloat16*, const __hip_bfloat16*, const float*,
+                             const float*, const float*, float*, float*, float*, int,
+                             long, float, float, hipStream_t);
+void launch_ibot_ce_fact_bwd(const __hip_bfloat16*, const __hip_bfloat16*, const float*,
+                             const float*, const float*, const float*, const float*,
+                             const float*, __hip_bfloat16*, int, long, float, float,
+                             hipStream_t);
+void launch_dino_ce_fact_fwd(const __hip_bfloat16*, const __hip_bfloat16*, const float*,
+                             const float*, float*, float*, float*, int, int, int, long,
+                             float, float, bool, hipStream_t);
+void launch_dino_ce_fact_bwd(const __hip_bfloat16*, const __hip_bfloat16*, const float*,
+                             const float*, const float*, const float*, const float*,
+                             __hip_bfloat16*, int, int, int, long, float, float, bool,
+                             hipStream_t);
 void launch_sinkhorn_colsum(const float*, float*, int, long, const float*, hipStream_t);
 void launch_sinkhorn_div_row(float*, const float*, int, long, float, const float*, bool,
                              hipStream_t);
@@ -436,6 +454,99 @@ torch::Tensor ibot_ce_bwd(torch::Tensor g, torch::Tensor x, torch::Tensor t,
                      w.data_ptr<float>(), lse.data_ptr<float>(), st.data_ptr<float>(),
                      g.data_ptr<float>(), (__hip_bfloat16*)dx.data_ptr(), M, K,
                      (float)(1.0 / temp), current_stream());
+  return dx;
+}
+
+torch::Tensor sinkhorn_fact_colsum(torch::Tensor x, torch::Tensor u, double temp) {
+  CHECK_INPUT(x);
+  const int M = x.size(0);
+  const long K = x.size(1);
+  auto A = torch::empty({K}, x.options().dtype(torch::kFloat));
+  const float* up = (u.defined() && u.numel() > 0) ? u.data_ptr<float>() : nullptr;
+  launch_sinkhorn_fact_colsum((const __hip_bfloat16*)x.data_ptr(), up,
+                              A.data_ptr<float>(), M, K, (float)(1.0 / temp),
+                              current_stream());
+  return A;
+}
+
+torch::Tensor sinkhorn_fact_rowsum(torch::Tensor x, torch::Tensor v, double temp) {
+  CHECK_INPUT(x);
+  const int M = x.size(0);
+  const long K = x.size(1);
+  auto u = torch::empty({M}, x.options().dtype(torch::kFloat));
+  launch_sinkhorn_fact_rowsum((const __hip_bfloat16*)x.data_ptr(), v.data_ptr<float>(),
+                              u.data_ptr<float>(), M, K, (float)(1.0 / temp),
+                              current_stream());
+  return u;
+}
+
+std::vector<torch::Tensor> ibot_ce_fact_fwd(torch::Tensor x, torch::Tensor xt,
+                                            torch::Tensor u, torch::Tensor v,
+                                            torch::Tensor w, double temp, double tt) {
+  CHECK_INPUT(x);
+  CHECK_INPUT(xt);
+  const int M = x.size(0);
+  const long K = x.size(1);
+  auto lse = torch::empty({M}, x.options().dtype(torch::kFloat));
+  auto st = torch::empty({M}, x.options().dtype(torch::kFloat));
+  auto loss = torch::zeros({}, x.options().dtype(torch::kFloat));
+  launch_ibot_ce_fact_fwd((const __hip_bfloat16*)x.data_ptr(),
+                          (const __hip_bfloat16*)xt.data_ptr(), u.data_ptr<float>(),
+                          v.data_ptr<float>(), w.data_ptr<float>(), lse.data_ptr<float>(),
+                          st.data_ptr<float>(), loss.data_ptr<float>(), M, K,
+                          (float)(1.0 / temp), (float)(1.0 / tt), current_stream());
+  return {loss, lse, st};
+}
+
+torch::Tensor ibot_ce_fact_bwd(torch::Tensor g, torch::Tensor x, torch::Tensor xt,
+                               torch::Tensor u, torch::Tensor v, torch::Tensor w,
+                               torch::Tensor lse, torch::Tensor st, double temp,
+                               double tt) {
+  const int M = x.size(0);
+  const long K = x.size(1);
+  auto dx = torch::empty_like(x);
+  launch_ibot_ce_fact_bwd((const __hip_bfloat16*)x.data_ptr(),
+                          (const __hip_bfloat16*)xt.data_ptr(), u.data_ptr<float>(),
+                          v.data_ptr<float>(), w.data_ptr<float>(), lse.data_ptr<float>(),
+                          st.data_ptr<float>(), g.data_ptr<float>(),
+                          (__hip_bfloat16*)dx.data_ptr(), M, K, (float)(1.0 / temp),
+                          (float)(1.0 / tt), current_stream());
+  return dx;
+}
+
+std::vector<torch::Tensor> dino_ce_fact_fwd(torch::Tensor x, torch::Tensor xt,
+                                            torch::Tensor u, torch::Tensor v, double temp,
+                                            double tt, bool ignore_diag) {
+  CHECK_INPUT(x);
+  CHECK_INPUT(xt);
+  const int S = x.size(0), B = x.size(1);
+  const int T = xt.size(0);
+  const long K = x.size(2);
+  auto lse = torch::empty({S * B}, x.options().dtype(torch::kFloat));
+  auto st = torch::empty({S * B}, x.options().dtype(torch::kFloat));
+  auto loss = torch::zeros({}, x.options().dtype(torch::kFloat));
+  launch_dino_ce_fact_fwd((const __hip_bfloat16*)x.data_ptr(),
+                          (const __hip_bfloat16*)xt.data_ptr(), u.data_ptr<float>(),
+                          v.data_ptr<float>(), lse.data_ptr<float>(), st.data_ptr<float>(),
+                          loss.data_ptr<float>(), S, T, B, K, (float)(1.0 / temp),
+                          (float)(1.0 / tt), ignore_diag, current_stream());
+  return {loss, lse, st};
+}
+
+torch::Tensor dino_ce_fact_bwd(torch::Tensor g, torch::Tensor x, torch::Tensor xt,
+                               torch::Tensor u, torch::Tensor v, torch::Tensor lse,
+                               torch::Tensor st, double temp, double tt,
+                               bool ignore_diag) {
+  const int S = x.size(0), B = x.size(1);
+  const int T = xt.size(0);
+  const long K = x.size(2);
+  auto dx = torch::empty_like(x);
+  launch_dino_ce_fact_bwd((const __hip_bfloat16*)x.data_ptr(),
+                          (const __hip_bfloat16*)xt.data_ptr(), u.data_ptr<float>(),
+                          v.data_ptr<float>(), lse.data_ptr<float>(), st.data_ptr<float>(),
+                          g.data_ptr<float>(), (__hip_bfloat16*)dx.data_ptr(), S, T, B, K,
+                          (float)(1.0 / temp), (float)(1.0 / tt), ignore_diag,
+                          current_stream());
   return dx;
 }
 
@@ -824,6 +935,12 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, mod) {
   mod.def("ibot_ce_fwd", &ibot_ce_fwd);
   mod.def("ibot_ce_bwd", &ibot_ce_bwd);
   mod.def("sinkhorn_exp", &sinkhorn_exp);
+  mod.def("sinkhorn_fact_colsum", &sinkhorn_fact_colsum);
+  mod.def("sinkhorn_fact_rowsum", &sinkhorn_fact_rowsum);
+  mod.def("ibot_ce_fact_fwd", &ibot_ce_fact_fwd);
+  mod.def("ibot_ce_fact_bwd", &ibot_ce_fact_bwd);
+  mod.def("dino_ce_fact_fwd", &dino_ce_fact_fwd);
+  mod.def("dino_ce_fact_bwd", &dino_ce_fact_bwd);
   mod.def("sinkhorn_colsum", &sinkhorn_colsum);
   mod.def("sinkhorn_div_row", &sinkhorn_div_row);
   mod.def("fmha_fwd", &fmha_fwd);

@@ -205,6 +205,178 @@ __global__ void sinkhorn_div_row_kernel(float* __restrict__ Q,
   for (long k = threadIdx.x; k < K; k += blockDim.x) qr[k] *= mul;
 }
 
+// ---------------- factored Sinkhorn-Knopp ----------------
+// Sinkhorn is diagonal scaling: Q_final = exp(x/T) ∘ u ⊗ v. Iterating on the
+// [M] and [K] scale vectors only costs 2 read passes of the bf16 logits per
+// iteration and never materializes the [M, 65536] fp32 Q (profiles/: the
+// materialized version spent ~7 ms/step on Q traffic). Per-iteration scalar
+// factors cancel in the final row-normalize, so no 1/sum_Q pass is needed.
+
+// A[k] = sum_m exp(x[m,k]*inv_temp) * u[m]   (u == nullptr -> 1)
+__global__ void sinkhorn_fact_colsum_kernel(const __hip_bfloat16* __restrict__ x,
+                                            const float* __restrict__ u,
+                                            float* __restrict__ A, int M, long K,
+                                            float inv_temp) {
+  for (long k = blockIdx.x * (long)blockDim.x + threadIdx.x; k < K;
+       k += (long)gridDim.x * blockDim.x) {
+    float acc = 0.f;
+    for (int m = 0; m < M; ++m) {
+      float e = __expf(bf16_to_f32(*(const short*)(x + (long)m * K + k)) * inv_temp);
+      acc += (u != nullptr ? u[m] : 1.0f) * e;
+    }
+    A[k] = acc;
+  }
+}
+
+// u[m] = 1 / sum_k exp(x[m,k]*inv_temp) * v[k]
+__global__ void sinkhorn_fact_rowsum_kernel(const __hip_bfloat16* __restrict__ x,
+                                            const float* __restrict__ v,
+                                            float* __restrict__ u, long K,
+                                            float inv_temp) {
+  __shared__ float red[16];
+  const int m = blockIdx.x;
+  const __hip_bfloat16* xr = x + (long)m * K;
+  float acc = 0.f;
+  for (long k = threadIdx.x; k < K; k += blockDim.x) {
+    acc += __expf(bf16_to_f32(*(const short*)(xr + k)) * inv_temp) * v[k];
+  }
+  acc = block_reduce_sum(acc, red);
+  if (threadIdx.x == 0) u[m] = 1.0f / acc;
+}
+
+// ---------------- CE with factored teacher ----------------
+// t[m,k] = exp(xt[m,k]*inv_tt) * u[m] * v[k]; rows sum to 1 by construction.
+
+__global__ void ibot_ce_fact_fwd_kernel(
+    const __hip_bfloat16* __restrict__ x, const __hip_bfloat16* __restrict__ xt,
+    const float* __restrict__ u, const float* __restrict__ v,
+    const float* __restrict__ w, float* __restrict__ lse_out, float* __restrict__ st_out,
+    float* __restrict__ loss_sum, long K, float inv_temp, float inv_tt) {
+  __shared__ float red[16];
+  const int row = blockIdx.x;
+  const __hip_bfloat16* xr = x + (long)row * K;
+  const __hip_bfloat16* xtr = xt + (long)row * K;
+  const float um = u[row];
+
+  float m = -INFINITY;
+  for (long i = threadIdx.x; i < K; i += blockDim.x)
+    m = fmaxf(m, bf16_to_f32(*(const short*)(xr + i)) * inv_temp);
+  m = block_reduce_max(m, red);
+  __syncthreads();
+
+  float sumexp = 0.f, dot = 0.f, st = 0.f;
+  for (long i = threadIdx.x; i < K; i += blockDim.x) {
+    const float xi = bf16_to_f32(*(const short*)(xr + i)) * inv_temp;
+    sumexp += __expf(xi - m);
+    const float ts = __expf(bf16_to_f32(*(const short*)(xtr + i)) * inv_tt) * um * v[i];
+    dot += ts * xi;
+    st += ts;
+  }
+  sumexp = block_reduce_sum(sumexp, red);
+  __syncthreads();
+  dot = block_reduce_sum(dot, red);
+  __syncthreads();
+  st = block_reduce_sum(st, red);
+  const float lse = m + __logf(sumexp);
+  if (threadIdx.x == 0) {
+    lse_out[row] = lse;
+    st_out[row] = st;
+    atomicAdd(loss_sum, w[row] * (st * lse - dot));
+  }
+}
+
+__global__ void ibot_ce_fact_bwd_kernel(
+    const __hip_bfloat16* __restrict__ x, const __hip_bfloat16* __restrict__ xt,
+    const float* __restrict__ u, const float* __restrict__ v,
+    const float* __restrict__ w, const float* __restrict__ lse_in,
+    const float* __restrict__ st_in, const float* __restrict__ g,
+    __hip_bfloat16* __restrict__ dx, long K, float inv_temp, float inv_tt) {
+  const int row = blockIdx.x;
+  const __hip_bfloat16* xr = x + (long)row * K;
+  const __hip_bfloat16* xtr = xt + (long)row * K;
+  const float um = u[row];
+  const float lse = lse_in[row];
+  const float st = st_in[row];
+  const float scale = g[0] * inv_temp * w[row];
+  __hip_bfloat16* dxr = dx + (long)row * K;
+  for (long i = threadIdx.x; i < K; i += blockDim.x) {
+    const float xi = bf16_to_f32(*(const short*)(xr + i)) * inv_temp;
+    const float sm = __expf(xi - lse);
+    const float ts = __expf(bf16_to_f32(*(const short*)(xtr + i)) * inv_tt) * um * v[i];
+    *reinterpret_cast<short*>(dxr + i) = f32_to_bf16(scale * (st * sm - ts));
+  }
+}
+
+__global__ void dino_ce_fact_fwd_kernel(
+    const __hip_bfloat16* __restrict__ x, const __hip_bfloat16* __restrict__ xt,
+    const float* __restrict__ u, const float* __restrict__ v,
+    float* __restrict__ lse_out, float* __restrict__ st_out, float* __restrict__ loss_sum,
+    int S, int T, int B, long K, float inv_temp, float inv_tt, bool ignore_diag) {
+  __shared__ float red[16];
+  const int row = blockIdx.x;
+  const int s = row / B;
+  const int b = row % B;
+  const __hip_bfloat16* xr = x + (long)row * K;
+
+  float m = -INFINITY;
+  for (long i = threadIdx.x; i < K; i += blockDim.x)
+    m = fmaxf(m, bf16_to_f32(*(const short*)(xr + i)) * inv_temp);
+  m = block_reduce_max(m, red);
+  __syncthreads();
+
+  float sumexp = 0.f, dot = 0.f, st = 0.f;
+  for (long i = threadIdx.x; i < K; i += blockDim.x) {
+    const float xi = bf16_to_f32(*(const short*)(xr + i)) * inv_temp;
+    sumexp += __expf(xi - m);
+    float ts = 0.f;
+    for (int tt = 0; tt < T; ++tt) {
+      if (ignore_diag && tt == s) continue;
+      const long trow = (long)tt * B + b;
+      ts += __expf(bf16_to_f32(*(const short*)(xt + trow * K + i)) * inv_tt) * u[trow] * v[i];
+    }
+    dot += ts * xi;
+    st += ts;
+  }
+  sumexp = block_reduce_sum(sumexp, red);
+  __syncthreads();
+  dot = block_reduce_sum(dot, red);
+  __syncthreads();
+  st = block_reduce_sum(st, red);
+  const float lse = m + __logf(sumexp);
+  if (threadIdx.x == 0) {
+    lse_out[row] = lse;
+    st_out[row] = st;
+    atomicAdd(loss_sum, st * lse - dot);
+  }
+}
+
+__global__ void dino_ce_fact_bwd_kernel(
+    const __hip_bfloat16* __restrict__ x, const __hip_bfloat16* __restrict__ xt,
+    const float* __restrict__ u, const float* __restrict__ v,
+    const float* __restrict__ lse_in, const float* __restrict__ st_in,
+    const float* __restrict__ g, __hip_bfloat16* __restrict__ dx, int S, int T, int B,
+    long K, float inv_temp, float inv_tt, bool ignore_diag) {
+  const int row = blockIdx.x;
+  const int s = row / B;
+  const int b = row % B;
+  const __hip_bfloat16* xr = x + (long)row * K;
+  __hip_bfloat16* dxr = dx + (long)row * K;
+  const float lse = lse_in[row];
+  const float st = st_in[row];
+  const float scale = g[0] * inv_temp;
+  for (long i = threadIdx.x; i < K; i += blockDim.x) {
+    const float xi = bf16_to_f32(*(const short*)(xr + i)) * inv_temp;
+    const float sm = __expf(xi - lse);
+    float ts = 0.f;
+    for (int tt = 0; tt < T; ++tt) {
+      if (ignore_diag && tt == s) continue;
+      const long trow = (long)tt * B + b;
+      ts += __expf(bf16_to_f32(*(const short*)(xt + trow * K + i)) * inv_tt) * u[trow] * v[i];
+    }
+    *reinterpret_cast<short*>(dxr + i) = f32_to_bf16(scale * (st * sm - ts));
+  }
+}
+
 // ---------------- launchers ----------------
 
 void launch_dino_ce_fwd(const __hip_bfloat16* x, const float* t, float* lse, float* st,
@@ -237,6 +409,56 @@ void launch_ibot_ce_bwd(const __hip_bfloat16* x, const float* t, const float* w,
   if (M == 0) return;
   hipLaunchKernelGGL(ibot_ce_bwd_kernel, dim3(M), dim3(CE_BLOCK), 0, stream, x, t, w,
                      lse, st, g, dx, K, inv_temp);
+}
+
+void launch_sinkhorn_fact_colsum(const __hip_bfloat16* x, const float* u, float* A,
+                                 int M, long K, float inv_temp, hipStream_t stream) {
+  int grid = (int)min((K + CE_BLOCK - 1) / CE_BLOCK, (long)2048);
+  hipLaunchKernelGGL(sinkhorn_fact_colsum_kernel, dim3(grid), dim3(CE_BLOCK), 0, stream,
+                     x, u, A, M, K, inv_temp);
+}
+
+void launch_sinkhorn_fact_rowsum(const __hip_bfloat16* x, const float* v, float* u,
+                                 int M, long K, float inv_temp, hipStream_t stream) {
+  hipLaunchKernelGGL(sinkhorn_fact_rowsum_kernel, dim3(M), dim3(CE_BLOCK), 0, stream, x,
+                     v, u, K, inv_temp);
+}
+
+void launch_ibot_ce_fact_fwd(const __hip_bfloat16* x, const __hip_bfloat16* xt,
+                             const float* u, const float* v, const float* w, float* lse,
+                             float* st, float* loss_sum, int M, long K, float inv_temp,
+                             float inv_tt, hipStream_t stream) {
+  if (M == 0) return;
+  hipLaunchKernelGGL(ibot_ce_fact_fwd_kernel, dim3(M), dim3(CE_BLOCK), 0, stream, x, xt,
+                     u, v, w, lse, st, loss_sum, K, inv_temp, inv_tt);
+}
+
+void launch_ibot_ce_fact_bwd(const __hip_bfloat16* x, const __hip_bfloat16* xt,
+                             const float* u, const float* v, const float* w,
+                             const float* lse, const float* st, const float* g,
+                             __hip_bfloat16* dx, int M, long K, float inv_temp,
+                             float inv_tt, hipStream_t stream) {
+  if (M == 0) return;
+  hipLaunchKernelGGL(ibot_ce_fact_bwd_kernel, dim3(M), dim3(CE_BLOCK), 0, stream, x, xt,
+                     u, v, w, lse, st, g, dx, K, inv_temp, inv_tt);
+}
+
+void launch_dino_ce_fact_fwd(const __hip_bfloat16* x, const __hip_bfloat16* xt,
+                             const float* u, const float* v, float* lse, float* st,
+                             float* loss_sum, int S, int T, int B, long K, float inv_temp,
+                             float inv_tt, bool ignore_diag, hipStream_t stream) {
+  hipLaunchKernelGGL(dino_ce_fact_fwd_kernel, dim3(S * B), dim3(CE_BLOCK), 0, stream, x,
+                     xt, u, v, lse, st, loss_sum, S, T, B, K, inv_temp, inv_tt,
+                     ignore_diag);
+}
+
+void launch_dino_ce_fact_bwd(const __hip_bfloat16* x, const __hip_bfloat16* xt,
+                             const float* u, const float* v, const float* lse,
+                             const float* st, const float* g, __hip_bfloat16* dx, int S,
+                             int T, int B, long K, float inv_temp, float inv_tt,
+                             bool ignore_diag, hipStream_t stream) {
+  hipLaunchKernelGGL(dino_ce_fact_bwd_kernel, dim3(S * B), dim3(CE_BLOCK), 0, stream, x,
+                     xt, u, v, lse, st, g, dx, S, T, B, K, inv_temp, inv_tt, ignore_diag);
 }
 
 void launch_sinkhorn_exp(const __hip_bfloat16* x, float* Q, float* total, long n,

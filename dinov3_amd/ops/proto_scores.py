@@ -23,6 +23,106 @@ def _world() -> int:
     return dist.get_world_size() if (dist.is_available() and dist.is_initialized()) else 1
 
 
+class FactoredProbs:
+    """Lazy sinkhorn output: t = exp(logits/temp) * u[..., None] * v.
+
+    Sinkhorn-Knopp is diagonal scaling of exp(x/T); carrying only the row/col
+    scale vectors lets the CE kernels consume the teacher distribution without
+    ever materializing the [rows, 65536] fp32 tensor.
+    """
+
+    def __init__(self, logits: torch.Tensor, u: torch.Tensor, v: torch.Tensor, temp: float):
+        self.logits = logits
+        self.u = u
+        self.v = v
+        self.temp = temp
+
+    @property
+    def shape(self):
+        return self.logits.shape
+
+    def reshape(self, *shape):
+        return FactoredProbs(self.logits.reshape(*shape), self.u.reshape(*shape[:-1]),
+                             self.v, self.temp)
+
+    def __getitem__(self, idx):
+        return FactoredProbs(self.logits[idx], self.u[idx], self.v, self.temp)
+
+    def materialize(self) -> torch.Tensor:
+        t = torch.exp(self.logits.float() / self.temp)
+        return t * self.u.unsqueeze(-1) * self.v
+
+    def float(self):
+        return self.materialize()
+
+
+@torch.no_grad()
+def sinkhorn_knopp_factored(teacher_logits: torch.Tensor, teacher_temp: float,
+                            n_iterations: int = 3) -> FactoredProbs:
+    """Factored sinkhorn: 2 bf16 read passes per iteration, no Q tensor.
+
+    Per-iteration scalar factors (sum_Q, K, B) cancel in the final
+    row-normalize, so only the doubly-stochastic scaling vectors survive —
+    identical output to the materialized reference."""
+    from . import hip_ops
+
+    ops = hip_ops()
+    world = _world()
+    M, K = teacher_logits.shape
+    x = teacher_logits.contiguous()
+    u = torch.empty(0)
+    for it in range(n_iterations):
+        A = ops.sinkhorn_fact_colsum(x, u if it > 0 else torch.empty(0, device=x.device),
+                                     teacher_temp)
+        if world > 1:
+            dist.all_reduce(A)
+        v = torch.reciprocal(A * K)
+        u = ops.sinkhorn_fact_rowsum(x, v, teacher_temp)
+    return FactoredProbs(x, u, v, teacher_temp)
+
+
+class _DinoCEFactFn(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, x, xt, u, v, temps, tempt, ignore_diag, denom):
+        from . import hip_ops
+
+        loss_sum, lse, st = hip_ops().dino_ce_fact_fwd(x, xt, u, v, temps, tempt, ignore_diag)
+        ctx.save_for_backward(x, xt, u, v, lse, st)
+        ctx.meta = (temps, tempt, ignore_diag, denom)
+        return loss_sum / denom
+
+    @staticmethod
+    def backward(ctx, g):
+        from . import hip_ops
+
+        x, xt, u, v, lse, st = ctx.saved_tensors
+        temps, tempt, ignore_diag, denom = ctx.meta
+        gs = (g / denom).float().contiguous()
+        dx = hip_ops().dino_ce_fact_bwd(gs, x, xt, u, v, lse, st, temps, tempt, ignore_diag)
+        return dx, None, None, None, None, None, None, None
+
+
+class _IbotCEFactFn(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, x, xt, u, v, w, temps, tempt):
+        from . import hip_ops
+
+        loss_sum, lse, st = hip_ops().ibot_ce_fact_fwd(x, xt, u, v, w, temps, tempt)
+        ctx.save_for_backward(x, xt, u, v, w, lse, st)
+        ctx.meta = (temps, tempt)
+        return loss_sum
+
+    @staticmethod
+    def backward(ctx, g):
+        from . import hip_ops
+
+        x, xt, u, v, w, lse, st = ctx.saved_tensors
+        temps, tempt = ctx.meta
+        dx = hip_ops().ibot_ce_fact_bwd(g.float().contiguous(), x, xt, u, v, w, lse, st,
+                                        temps, tempt)
+        return dx, None, None, None, None, None, None
+
+
 class _DinoCEFn(torch.autograd.Function):
     @staticmethod
     def forward(ctx, x, t, temp, ignore_diag, denom):
@@ -153,10 +253,18 @@ def dino_softmax_ce(student_logits: torch.Tensor, teacher_probs: torch.Tensor,
     T = teacher_probs.shape[0]
     if use_hip(student_logits) and student_logits.dtype == torch.bfloat16:
         denom = float(B * S * T - B * min(S, T)) if ignore_diagonal else float(B * S * T)
+        if isinstance(teacher_probs, FactoredProbs):
+            return _DinoCEFactFn.apply(
+                student_logits.contiguous(), teacher_probs.logits.contiguous(),
+                teacher_probs.u.reshape(-1).contiguous(), teacher_probs.v,
+                student_temp, teacher_probs.temp, ignore_diagonal, denom,
+            )
         return _DinoCEFn.apply(
             student_logits.contiguous(), teacher_probs.float().contiguous(),
             student_temp, ignore_diagonal, denom,
         )
+    if isinstance(teacher_probs, FactoredProbs):
+        teacher_probs = teacher_probs.materialize()
     logp = F.log_softmax(student_logits.float() / student_temp, dim=-1)
     tp = teacher_probs.float()
     if ignore_diagonal:
@@ -185,11 +293,20 @@ def ibot_softmax_ce(student_patch_logits: torch.Tensor, teacher_patch_probs: tor
         w = (masks_weight.float().contiguous() if masks_weight is not None
              else torch.full((M,), 1.0 / max(n_total_rows, 1),
                              device=student_patch_logits.device))
+        if isinstance(teacher_patch_probs, FactoredProbs):
+            return _IbotCEFactFn.apply(
+                student_patch_logits.contiguous(),
+                teacher_patch_probs.logits.contiguous(),
+                teacher_patch_probs.u.contiguous(), teacher_patch_probs.v, w,
+                student_temp, teacher_patch_probs.temp,
+            )
         denom = 1.0  # weights carry the normalization; caller divides by rows
         return _IbotCEFn.apply(
             student_patch_logits.contiguous(), teacher_patch_probs.float().contiguous(),
             w, student_temp, denom,
         )
+    if isinstance(teacher_patch_probs, FactoredProbs):
+        teacher_patch_probs = teacher_patch_probs.materialize()
     logp = F.log_softmax(student_patch_logits.float() / student_temp, dim=-1)
     per_row = (teacher_patch_probs.float() * logp).sum(dim=-1)  # [M]
     if masks_weight is not None:

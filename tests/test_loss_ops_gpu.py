@@ -94,3 +94,63 @@ def test_sinkhorn_fused_ibot_total_columns():
     finally:
         del os.environ["DINOV3_DISABLE_HIP"]
     _close(Q_hip, Q_ref, 2e-3, "sinkhorn ibot")
+
+
+def test_sinkhorn_factored_matches_materialized():
+    torch.manual_seed(5)
+    M, K = 40, 4096
+    logits = torch.randn(M, K, device=DEV).bfloat16()
+    from dinov3_amd.ops.proto_scores import sinkhorn_knopp, sinkhorn_knopp_factored
+
+    fact = sinkhorn_knopp_factored(logits, teacher_temp=0.1)
+    Q_fact = fact.materialize()
+    import os
+
+    os.environ["DINOV3_DISABLE_HIP"] = "1"
+    try:
+        Q_ref = sinkhorn_knopp(logits.float(), teacher_temp=0.1)
+    finally:
+        del os.environ["DINOV3_DISABLE_HIP"]
+    _close(Q_fact, Q_ref, 2e-3, "sinkhorn factored vs materialized")
+    _close(Q_fact.sum(dim=1), torch.ones(M, device=DEV), 1e-3, "factored rowsum")
+
+
+def test_dino_ce_factored_matches_dense():
+    torch.manual_seed(6)
+    S, T, B, K = 3, 2, 4, 2048
+    x = torch.randn(S, B, K, device=DEV).bfloat16().requires_grad_(True)
+    xt = torch.randn(T, B, K, device=DEV).bfloat16()
+    from dinov3_amd.ops.proto_scores import (FactoredProbs, dino_softmax_ce,
+                                             sinkhorn_knopp_factored)
+
+    fact = sinkhorn_knopp_factored(xt.reshape(T * B, K), 0.07)
+    fact3 = fact.reshape(T, B, K)
+    loss = dino_softmax_ce(x, fact3, student_temp=0.1)
+    loss.backward()
+    g_fact = x.grad.clone()
+    x.grad = None
+    dense = fact3.materialize()
+    loss_dense = dino_softmax_ce(x, dense, student_temp=0.1)
+    loss_dense.backward()
+    _close(loss, loss_dense, 5e-3, "dino fact loss")
+    _close(g_fact, x.grad, 1e-4 + 0.02 * x.grad.abs().max().item(), "dino fact grad")
+
+
+def test_ibot_ce_factored_matches_dense():
+    torch.manual_seed(7)
+    M, K = 30, 2048
+    x = torch.randn(M, K, device=DEV).bfloat16().requires_grad_(True)
+    xt = torch.randn(M, K, device=DEV).bfloat16()
+    w = torch.rand(M, device=DEV) + 0.1
+    from dinov3_amd.ops.proto_scores import ibot_softmax_ce, sinkhorn_knopp_factored
+
+    fact = sinkhorn_knopp_factored(xt, 0.07)
+    loss = ibot_softmax_ce(x, fact, n_total_rows=M, student_temp=0.1, masks_weight=w)
+    loss.backward()
+    g_fact = x.grad.clone()
+    x.grad = None
+    dense = fact.materialize()
+    loss_dense = ibot_softmax_ce(x, dense, n_total_rows=M, student_temp=0.1, masks_weight=w)
+    loss_dense.backward()
+    _close(loss, loss_dense, max(5e-3, 1e-4 * abs(loss_dense.item())), "ibot fact loss")
+    _close(g_fact, x.grad, 1e-4 + 0.02 * x.grad.abs().max().item(), "ibot fact grad")
