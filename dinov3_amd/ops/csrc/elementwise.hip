@@ -66,7 +66,31 @@ __global__ void bias_gelu_bwd_kernel(const T* __restrict__ dy, const T* __restri
   T bb[8];
   Vec8<T>::load(bb, bias + col8);
   float db[8] = {0.f, 0.f, 0.f, 0.f, 0.f, 0.f, 0.f, 0.f};
-  for (long row = blockIdx.x; row < rows; row += gridDim.x) {
+  // 4 rows in flight per iteration: the fixed-column walk is a strided
+  // stream, so batch the loads for ILP/prefetch depth
+  long row = blockIdx.x * 4L;
+  const long rstep = (long)gridDim.x * 4;
+  for (; row + 3 < rows; row += rstep) {
+    T xb[4][8], gb[4][8], ob[4][8];
+#pragma unroll
+    for (int rr = 0; rr < 4; ++rr) {
+      const long off = (row + rr) * (long)H + col8;
+      Vec8<T>::load(xb[rr], x + off);
+      Vec8<T>::load(gb[rr], dy + off);
+    }
+#pragma unroll
+    for (int rr = 0; rr < 4; ++rr) {
+#pragma unroll
+      for (int e = 0; e < 8; ++e) {
+        float v = ScalarOps<T>::load(xb[rr] + e) + ScalarOps<T>::load(bb + e);
+        float g = ScalarOps<T>::load(gb[rr] + e) * gelu_tanh_grad(v);
+        ScalarOps<T>::store(ob[rr] + e, g);
+        db[e] += g;
+      }
+      Vec8<T>::store(dx + (row + rr) * (long)H + col8, ob[rr]);
+    }
+  }
+  for (; row < rows; ++row) {
     const long off = row * (long)H + col8;
     T xb[8], gb[8], ob[8];
     Vec8<T>::load(xb, x + off);
@@ -295,7 +319,7 @@ template <typename T>
 void launch_bias_gelu_bwd(const T* dy, const T* x, const T* bias, T* dx, float* dbias,
                           long rows, int H, hipStream_t stream) {
   const int col_tiles = (H / 8 + EW_BLOCK - 1) / EW_BLOCK;
-  int row_grid = (int)min(rows, (long)(2048 / col_tiles + 1));
+  int row_grid = (int)min((rows + 3) / 4, (long)(2048 / col_tiles + 1));
   hipLaunchKernelGGL((bias_gelu_bwd_kernel<T>), dim3(row_grid, col_tiles), dim3(EW_BLOCK),
                      0, stream, dy, x, bias, dx, dbias, rows, H);
 }

@@ -11,6 +11,13 @@
 
 #define NORM_BLOCK 256
 
+// block size matched to the row width so no lanes idle in the 8-wide loops
+static inline int norm_block_for(int D) {
+  if (D >= 2048) return 256;
+  if (D >= 1024) return 128;
+  return 64;
+}
+
 template <typename T, bool V8>
 DEV_INLINE void row_load8(const T* p, int i, float* out) {
   if constexpr (V8) {
@@ -305,7 +312,7 @@ void launch_layernorm_fwd(const T* x, const T* w, const T* b, T* y, float* mean,
                           float* rstd, long rows, int D, float eps, hipStream_t stream) {
   int grid = (int)min(rows, (long)8192);
   DISPATCH_V8(D, hipLaunchKernelGGL(HIP_KERNEL_NAME(layernorm_fwd_kernel<T, V8>), dim3(grid),
-                                     dim3(NORM_BLOCK), 0, stream, x, w, b, y, mean, rstd,
+                                     dim3(norm_block_for(D)), 0, stream, x, w, b, y, mean, rstd,
                                      rows, D, eps));
 }
 
@@ -316,7 +323,7 @@ void launch_layernorm_bwd(const T* dy, const T* x, const T* w, const float* mean
   int grid = (int)min(rows, (long)2048);
   size_t shmem = (16 + 2 * (size_t)D) * sizeof(float);
   DISPATCH_V8(D, hipLaunchKernelGGL(HIP_KERNEL_NAME(layernorm_bwd_kernel<T, V8>), dim3(grid),
-                                     dim3(NORM_BLOCK), shmem, stream, dy, x, w, mean,
+                                     dim3(norm_block_for(D)), shmem, stream, dy, x, w, mean,
                                      rstd, dx, dw, db, rows, D));
 }
 
@@ -325,7 +332,7 @@ void launch_rmsnorm_fwd(const T* x, const T* w, T* y, float* rstd, long rows, in
                         float eps, hipStream_t stream) {
   int grid = (int)min(rows, (long)8192);
   DISPATCH_V8(D, hipLaunchKernelGGL(HIP_KERNEL_NAME(rmsnorm_fwd_kernel<T, V8>), dim3(grid),
-                                     dim3(NORM_BLOCK), 0, stream, x, w, y, rstd, rows, D,
+                                     dim3(norm_block_for(D)), 0, stream, x, w, y, rstd, rows, D,
                                      eps));
 }
 
@@ -335,7 +342,7 @@ void launch_rmsnorm_bwd(const T* dy, const T* x, const T* w, const float* rstd, 
   int grid = (int)min(rows, (long)2048);
   size_t shmem = (16 + (size_t)D) * sizeof(float);
   DISPATCH_V8(D, hipLaunchKernelGGL(HIP_KERNEL_NAME(rmsnorm_bwd_kernel<T, V8>), dim3(grid),
-                                     dim3(NORM_BLOCK), shmem, stream, dy, x, w, rstd, dx,
+                                     dim3(norm_block_for(D)), shmem, stream, dy, x, w, rstd, dx,
                                      dw, rows, D));
 }
 
@@ -344,7 +351,7 @@ void launch_l2norm_fwd(const T* x, T* y, float* s, long rows, int D, float eps,
                        hipStream_t stream) {
   int grid = (int)min(rows, (long)8192);
   DISPATCH_V8(D, hipLaunchKernelGGL(HIP_KERNEL_NAME(l2norm_fwd_kernel<T, V8>), dim3(grid),
-                                     dim3(NORM_BLOCK), 0, stream, x, y, s, rows, D, eps));
+                                     dim3(norm_block_for(D)), 0, stream, x, y, s, rows, D, eps));
 }
 
 template <typename T>
@@ -352,7 +359,7 @@ void launch_l2norm_bwd(const T* dy, const T* y, const float* s, T* dx, long rows
                        float eps, hipStream_t stream) {
   int grid = (int)min(rows, (long)8192);
   DISPATCH_V8(D, hipLaunchKernelGGL(HIP_KERNEL_NAME(l2norm_bwd_kernel<T, V8>), dim3(grid),
-                                     dim3(NORM_BLOCK), 0, stream, dy, y, s, dx, rows, D,
+                                     dim3(norm_block_for(D)), 0, stream, dy, y, s, dx, rows, D,
                                      eps));
 }
 

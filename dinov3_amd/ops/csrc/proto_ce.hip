@@ -213,18 +213,35 @@ __global__ void sinkhorn_div_row_kernel(float* __restrict__ Q,
 // factors cancel in the final row-normalize, so no 1/sum_Q pass is needed.
 
 // A[k] = sum_m exp(x[m,k]*inv_temp) * u[m]   (u == nullptr -> 1)
+// 2D tiling: block (kw, mt) reduces a 256-row m-tile over a 2048-col window
+// (contiguous 4 KB per row per block: streams, unlike a per-thread column
+// walk with a 128 KB stride), then one atomicAdd per column per m-tile.
+#define SK_MT 256
 __global__ void sinkhorn_fact_colsum_kernel(const __hip_bfloat16* __restrict__ x,
                                             const float* __restrict__ u,
                                             float* __restrict__ A, int M, long K,
                                             float inv_temp) {
-  for (long k = blockIdx.x * (long)blockDim.x + threadIdx.x; k < K;
-       k += (long)gridDim.x * blockDim.x) {
-    float acc = 0.f;
-    for (int m = 0; m < M; ++m) {
-      float e = __expf(bf16_to_f32(*(const short*)(x + (long)m * K + k)) * inv_temp);
-      acc += (u != nullptr ? u[m] : 1.0f) * e;
+  const long k0 = (long)blockIdx.x * (blockDim.x * 8);
+  const int m0 = blockIdx.y * SK_MT;
+  const int m1 = min(m0 + SK_MT, M);
+  float acc[8] = {0.f, 0.f, 0.f, 0.f, 0.f, 0.f, 0.f, 0.f};
+  const long kb = k0 + threadIdx.x * 8;
+  if (kb >= K) return;
+  for (int m = m0; m < m1; ++m) {
+    const float um = (u != nullptr ? u[m] : 1.0f);
+    __hip_bfloat16 xb[8];
+    Vec8<__hip_bfloat16>::load(xb, x + (long)m * K + kb);
+#pragma unroll
+    for (int e = 0; e < 8; ++e) {
+      acc[e] += um * __expf(bf16_to_f32(*(short*)(xb + e)) * inv_temp);
     }
-    A[k] = acc;
+  }
+  if (gridDim.y == 1) {
+#pragma unroll
+    for (int e = 0; e < 8; ++e) A[kb + e] = acc[e];
+  } else {
+#pragma unroll
+    for (int e = 0; e < 8; ++e) atomicAdd(A + kb + e, acc[e]);
   }
 }
 
@@ -413,8 +430,10 @@ void launch_ibot_ce_bwd(const __hip_bfloat16* x, const float* t, const float* w,
 
 void launch_sinkhorn_fact_colsum(const __hip_bfloat16* x, const float* u, float* A,
                                  int M, long K, float inv_temp, hipStream_t stream) {
-  int grid = (int)min((K + CE_BLOCK - 1) / CE_BLOCK, (long)2048);
-  hipLaunchKernelGGL(sinkhorn_fact_colsum_kernel, dim3(grid), dim3(CE_BLOCK), 0, stream,
+  const int kw = (int)((K + CE_BLOCK * 8 - 1) / (CE_BLOCK * 8));
+  const int mt = (M + SK_MT - 1) / SK_MT;
+  if (mt > 1) hipMemsetAsync(A, 0, K * sizeof(float), stream);
+  hipLaunchKernelGGL(sinkhorn_fact_colsum_kernel, dim3(kw, mt), dim3(CE_BLOCK), 0, stream,
                      x, u, A, M, K, inv_temp);
 }
 

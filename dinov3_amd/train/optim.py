@@ -169,16 +169,17 @@ class FusedAdamW:
 
     def zero_grad(self, set_to_none: Optional[bool] = None) -> None:
         # once a plan exists, grads must keep their storage: zero in place
+        # (one foreach launch, not one fill per tensor)
         if set_to_none is None:
             set_to_none = self._plan is None
-        for g in self.groups:
-            for p in g["params"]:
-                if p.grad is None:
-                    continue
-                if set_to_none:
+        if set_to_none:
+            for g in self.groups:
+                for p in g["params"]:
                     p.grad = None
-                else:
-                    p.grad.zero_()
+            return
+        grads = [p.grad for g in self.groups for p in g["params"] if p.grad is not None]
+        if grads:
+            torch._foreach_zero_(grads)
 
     # ------------------------------------------------------------------
     def state_dict(self) -> dict:
