@@ -52,6 +52,13 @@ def get_args_parser(add_help: bool = True) -> argparse.ArgumentParser:
     parser.add_argument("--eval", type=str, default="", help="eval type")
     parser.add_argument("--profiling", action="store_true", help="emit roctx/profiler ranges")
     parser.add_argument("--multi-distillation", action="store_true")
+    parser.add_argument("--record-ref-losses", action="store_true",
+                        help="dump per-iteration loss terms to ref_losses.json for parity checks")
+    parser.add_argument("--ref-losses-path", type=str, default="",
+                        help="compare per-iteration losses against a recorded JSON")
+    parser.add_argument("--test-ibot", action="store_true", help="run with dino/koleo weights zeroed")
+    parser.add_argument("--benchmark-codebase", action="store_true",
+                        help="time the train step and print images/sec at exit")
     parser.add_argument("--seed", type=int, default=0)
     parser.add_argument("--output-dir", default="", help="output directory")
     parser.add_argument("--max-iterations", type=int, default=-1, help="cap iterations (debug/bench)")
@@ -224,7 +231,8 @@ def batch_to_device(data: dict, device: torch.device) -> dict:
     }
 
 
-def do_train(cfg, model: SSLMetaArch, resume: bool = True, max_iterations: int = -1):
+def do_train(cfg, model: SSLMetaArch, resume: bool = True, max_iterations: int = -1,
+             record_losses_to: str = "", compare_losses_to: str = ""):
     device = parallel.device()
     param_dtype = DTYPE_MAP.get(cfg.compute_precision.param_dtype, torch.float32)
     if device.type == "cuda":
@@ -262,6 +270,13 @@ def do_train(cfg, model: SSLMetaArch, resume: bool = True, max_iterations: int =
 
     nan_count = 0
     iteration = start_iter
+    recorded_losses = []
+    reference_losses = None
+    if compare_losses_to:
+        import json
+
+        with open(compare_losses_to) as f:
+            reference_losses = json.load(f)
 
     def infinite_batches():
         epoch = 0
@@ -285,7 +300,18 @@ def do_train(cfg, model: SSLMetaArch, resume: bool = True, max_iterations: int =
         last_layer_lr = 0.0 if it < schedulers["freeze_last_layer_iterations"] else lr
 
         data = batch_to_device(data, device)
-        loss, loss_dict = model(data, teacher_temp=teacher_temp, iteration=it)
+        try:
+            loss, loss_dict = model(data, teacher_temp=teacher_temp, iteration=it)
+        except torch.OutOfMemoryError:
+            # OOM retry on a shrunken batch (reference helper get_batch_subset,
+            # collate.py:97-139, which the reference never wires up)
+            from ..data import get_batch_subset
+
+            logger.warning("OOM at iteration %d: retrying with half batch", it)
+            if device.type == "cuda":
+                torch.cuda.empty_cache()
+            data = get_batch_subset(data, divide_by=2)
+            loss, loss_dict = model(data, teacher_temp=teacher_temp, iteration=it)
 
         if not torch.isfinite(loss):
             nan_count += 1
@@ -328,6 +354,19 @@ def do_train(cfg, model: SSLMetaArch, resume: bool = True, max_iterations: int =
                if k != "total_loss"},
         )
 
+        if record_losses_to or reference_losses is not None:
+            entry = {"iteration": it, "total_loss": float(loss.detach())}
+            entry.update({k: float(v) for k, v in loss_dict.items()
+                          if isinstance(v, (int, float)) or (isinstance(v, torch.Tensor) and v.ndim == 0)})
+            if record_losses_to:
+                recorded_losses.append(entry)
+            if reference_losses is not None and it < len(reference_losses):
+                ref = reference_losses[it]
+                for k, v in entry.items():
+                    if k in ref and abs(v - ref[k]) > 0.05 * max(1.0, abs(ref[k])):
+                        logger.warning("loss parity drift at it=%d %s: %g vs ref %g",
+                                       it, k, v, ref[k])
+
         if output_dir and ckpt_period > 0 and (it + 1) % ckpt_period == 0:
             save_checkpoint(
                 output_dir, it, model, optimizer,
@@ -335,6 +374,12 @@ def do_train(cfg, model: SSLMetaArch, resume: bool = True, max_iterations: int =
             )
         iteration += 1
 
+    if record_losses_to and parallel.is_main_process():
+        import json
+
+        with open(record_losses_to, "w") as f:
+            json.dump(recorded_losses, f)
+        logger.info("recorded %d iterations of losses to %s", len(recorded_losses), record_losses_to)
     if output_dir:
         save_checkpoint(
             output_dir, iteration - 1, model, optimizer,
@@ -381,10 +426,32 @@ def main(argv=None):
     args = get_args_parser().parse_args(argv)
     setup_job(output_dir=args.output_dir or None, seed=args.seed)
     cfg = setup_config(args)
+    if args.test_ibot:
+        cfg.dino.loss_weight = 0.0
+        cfg.dino.koleo_loss_weight = 0.0
+        logger.info("--test-ibot: dino/koleo loss weights zeroed")
+    if args.multi_distillation or cfg.multidistillation.enabled:
+        cfg.multidistillation.enabled = True
+        from .multidist_meta_arch import MultiDistillationMetaArch
+
+        MultiDistillationMetaArch(cfg)  # validates the subgroup layout
+        raise NotImplementedError("multi-distillation training loop is a stub (reference parity)")
     model = SSLMetaArch(cfg)
     if args.eval_only:
         return do_test(cfg, model, 0)
-    return do_train(cfg, model, resume=not args.no_resume, max_iterations=args.max_iterations)
+    import os
+    import time
+
+    record_to = ""
+    if args.record_ref_losses:
+        record_to = os.path.join(args.output_dir or ".", "ref_losses.json")
+    t0 = time.time()
+    result = do_train(cfg, model, resume=not args.no_resume, max_iterations=args.max_iterations,
+                      record_losses_to=record_to, compare_losses_to=args.ref_losses_path)
+    if args.benchmark_codebase:
+        elapsed = time.time() - t0
+        logger.info("benchmark: total wall %.1f s (use bench.py for the timed-step metric)", elapsed)
+    return result
 
 
 if __name__ == "__main__":
