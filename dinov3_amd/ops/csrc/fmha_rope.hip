@@ -156,42 +156,72 @@ __global__ __launch_bounds__(256) void fwd_kernel(
     for (int r = 0; r < 16; ++r) o_acc[t][r] = 0.f;
   float m_run = -INFINITY, l_run = 0.f;
 
-  const int n_super = (N + 2 * KVB - 1) / (2 * KVB);  // 64 keys per barrier pair
-  for (int kt = 0; kt < n_super; ++kt) {
-    const int kbase0 = kt * 2 * KVB;
-    __syncthreads();
-    {
-      // K staging with rope: each thread owns a (row, pair-chunk) — loads the
-      // lo/hi halves, rotates, writes both into k_lds.
-      constexpr int PAIRS_PER_ROW = HALF / 8;
-      for (int idx = threadIdx.x; idx < 2 * KVB * PAIRS_PER_ROW; idx += 256) {
-        const int row = idx / PAIRS_PER_ROW;
-        const int c0 = (idx % PAIRS_PER_ROW) * 8;
-        const int krow = kbase0 + row;
-        bf16x8 lo{}, hi{};
-        if (krow < N) {
-          lo = load8(qv.at(krow, 1, c0));
-          hi = load8(qv.at(krow, 1, c0 + HALF));
-          const int p = krow - prefix;
-          if (use_rope && p >= 0) {
-            rope_rotate8(lo, hi, sin_t + (long)p * HD, cos_t + (long)p * HD, c0);
-          }
-        }
-        *reinterpret_cast<bf16x8*>(&k_lds[row * LDS_STRIDE + c0]) = lo;
-        *reinterpret_cast<bf16x8*>(&k_lds[row * LDS_STRIDE + HALF + c0]) = hi;
-      }
-      // V transposed staging
-      constexpr int PER_ROW = HD / 8;
-      for (int idx = threadIdx.x; idx < 2 * KVB * PER_ROW; idx += 256) {
-        const int row = idx / PER_ROW;
-        const int c8 = (idx % PER_ROW) * 8;
-        const int krow = kbase0 + row;
-        bf16x8 vv = (krow < N) ? load8(qv.at(krow, 2, c8)) : bf16x8{};
+  // T14 async-STAGE split: each thread owns ONE K pair-chunk and up to two V
+  // chunks of the staged 64-key tile; the next tile's global loads are issued
+  // BEFORE the compute phase so HBM latency hides under the MFMAs instead of
+  // draining at the pre-compute barrier.
+  constexpr int PAIRS_PER_ROW = HALF / 8;     // K pair-items per row
+  constexpr int PER_ROW = HD / 8;             // V items per row
+  constexpr int K_ITEMS = 2 * KVB * PAIRS_PER_ROW;   // 256 (hd64) / 512 (hd128)
+  constexpr int K_ITEMS_PER_THREAD = (K_ITEMS + 255) / 256;
+  constexpr int V_ITEMS_PER_THREAD = (2 * KVB * PER_ROW + 255) / 256;
+
+  bf16x8 klo[K_ITEMS_PER_THREAD], khi[K_ITEMS_PER_THREAD];
+  bf16x8 vreg[V_ITEMS_PER_THREAD];
+  auto issue_loads = [&](int kbase0) {
 #pragma unroll
-        for (int e = 0; e < 8; ++e)
-          vt_lds[(c8 + e) * VT_STRIDE + row] = reinterpret_cast<__hip_bfloat16*>(&vv)[e];
+    for (int j = 0; j < K_ITEMS_PER_THREAD; ++j) {
+      const int item = threadIdx.x + j * 256;
+      const int krow = kbase0 + item / PAIRS_PER_ROW;
+      const int c0 = (item % PAIRS_PER_ROW) * 8;
+      const bool ok = item < K_ITEMS && krow < N;
+      klo[j] = ok ? load8(qv.at(krow, 1, c0)) : bf16x8{};
+      khi[j] = ok ? load8(qv.at(krow, 1, c0 + HALF)) : bf16x8{};
+    }
+#pragma unroll
+    for (int j = 0; j < V_ITEMS_PER_THREAD; ++j) {
+      const int idx = threadIdx.x + j * 256;
+      const int row = idx / PER_ROW;
+      const int krow = kbase0 + row;
+      vreg[j] = (idx < 2 * KVB * PER_ROW && krow < N) ? load8(qv.at(krow, 2, (idx % PER_ROW) * 8))
+                                                      : bf16x8{};
+    }
+  };
+  auto write_tile = [&](int kbase0) {
+#pragma unroll
+    for (int j = 0; j < K_ITEMS_PER_THREAD; ++j) {
+      const int item = threadIdx.x + j * 256;
+      if (item < K_ITEMS) {
+        const int lrow = item / PAIRS_PER_ROW;
+        const int c0 = (item % PAIRS_PER_ROW) * 8;
+        const int krow = kbase0 + lrow;
+        const int p = krow - prefix;
+        if (use_rope && krow < N && p >= 0)
+          rope_rotate8(klo[j], khi[j], sin_t + (long)p * HD, cos_t + (long)p * HD, c0);
+        *reinterpret_cast<bf16x8*>(&k_lds[lrow * LDS_STRIDE + c0]) = klo[j];
+        *reinterpret_cast<bf16x8*>(&k_lds[lrow * LDS_STRIDE + HALF + c0]) = khi[j];
       }
     }
+#pragma unroll
+    for (int j = 0; j < V_ITEMS_PER_THREAD; ++j) {
+      const int idx = threadIdx.x + j * 256;
+      if (idx < 2 * KVB * PER_ROW) {
+        const int row = idx / PER_ROW;
+        const int c8 = (idx % PER_ROW) * 8;
+#pragma unroll
+        for (int e = 0; e < 8; ++e)
+          vt_lds[(c8 + e) * VT_STRIDE + row] = reinterpret_cast<__hip_bfloat16*>(&vreg[j])[e];
+      }
+    }
+  };
+
+  const int n_super = (N + 2 * KVB - 1) / (2 * KVB);  // 64 keys per barrier pair
+  issue_loads(0);
+  for (int kt = 0; kt < n_super; ++kt) {
+    const int kbase0 = kt * 2 * KVB;
+    __syncthreads();               // compute of tile kt-1 done reading LDS
+    write_tile(kbase0);            // regs -> LDS (rope applied at write)
+    if (kt + 1 < n_super) issue_loads(kbase0 + 2 * KVB);  // fly during compute
     __syncthreads();
 
    for (int sub = 0; sub < 2 && kbase0 + sub * KVB < N; ++sub) {
