@@ -359,41 +359,65 @@ __global__ __launch_bounds__(256) void bwd_dq_kernel(
 #pragma unroll
     for (int r = 0; r < 16; ++r) dq_acc[t][r] = 0.f;
 
+  constexpr int PAIRS_PER_ROW = HALF / 8;
+  constexpr int PER_ROW = HD / 8;
+  constexpr int K_ITEMS = 2 * KVB * PAIRS_PER_ROW;
+  constexpr int K_IPT = (K_ITEMS + 255) / 256;
+  constexpr int V_IPT = (2 * KVB * PER_ROW + 255) / 256;
+  bf16x8 klo[K_IPT], khi[K_IPT], vreg[V_IPT];
+  auto issue_loads = [&](int kbase0) {
+#pragma unroll
+    for (int j = 0; j < K_IPT; ++j) {
+      const int item = threadIdx.x + j * 256;
+      const int krow = kbase0 + item / PAIRS_PER_ROW;
+      const int c0 = (item % PAIRS_PER_ROW) * 8;
+      const bool ok = item < K_ITEMS && krow < N;
+      klo[j] = ok ? load8(qv.at(krow, 1, c0)) : bf16x8{};
+      khi[j] = ok ? load8(qv.at(krow, 1, c0 + HALF)) : bf16x8{};
+    }
+#pragma unroll
+    for (int j = 0; j < V_IPT; ++j) {
+      const int idx = threadIdx.x + j * 256;
+      const int krow = kbase0 + idx / PER_ROW;
+      vreg[j] = (idx < 2 * KVB * PER_ROW && krow < N)
+                    ? load8(qv.at(krow, 2, (idx % PER_ROW) * 8)) : bf16x8{};
+    }
+  };
+  auto write_tile = [&](int kbase0) {
+#pragma unroll
+    for (int j = 0; j < K_IPT; ++j) {
+      const int item = threadIdx.x + j * 256;
+      if (item < K_ITEMS) {
+        const int lrow = item / PAIRS_PER_ROW;
+        const int c0 = (item % PAIRS_PER_ROW) * 8;
+        const int krow = kbase0 + lrow;
+        const int p = krow - prefix;
+        if (use_rope && krow < N && p >= 0)
+          rope_rotate8(klo[j], khi[j], sin_t + (long)p * HD, cos_t + (long)p * HD, c0);
+        *reinterpret_cast<bf16x8*>(&k_lds[lrow * LDS_STRIDE + c0]) = klo[j];
+        *reinterpret_cast<bf16x8*>(&k_lds[lrow * LDS_STRIDE + HALF + c0]) = khi[j];
+#pragma unroll
+        for (int e = 0; e < 8; ++e) {
+          kt_lds[(c0 + e) * KT_STRIDE + lrow] = reinterpret_cast<__hip_bfloat16*>(&klo[j])[e];
+          kt_lds[(c0 + HALF + e) * KT_STRIDE + lrow] = reinterpret_cast<__hip_bfloat16*>(&khi[j])[e];
+        }
+      }
+    }
+#pragma unroll
+    for (int j = 0; j < V_IPT; ++j) {
+      const int idx = threadIdx.x + j * 256;
+      if (idx < 2 * KVB * PER_ROW)
+        *reinterpret_cast<bf16x8*>(&v_lds[(idx / PER_ROW) * LDS_STRIDE + (idx % PER_ROW) * 8]) = vreg[j];
+    }
+  };
+
   const int n_super = (N + 2 * KVB - 1) / (2 * KVB);
+  issue_loads(0);
   for (int kt = 0; kt < n_super; ++kt) {
     const int kbase0 = kt * 2 * KVB;
     __syncthreads();
-    {
-      constexpr int PAIRS_PER_ROW = HALF / 8;
-      for (int idx = threadIdx.x; idx < 2 * KVB * PAIRS_PER_ROW; idx += 256) {
-        const int row = idx / PAIRS_PER_ROW;
-        const int c0 = (idx % PAIRS_PER_ROW) * 8;
-        const int krow = kbase0 + row;
-        bf16x8 lo{}, hi{};
-        if (krow < N) {
-          lo = load8(qv.at(krow, 1, c0));
-          hi = load8(qv.at(krow, 1, c0 + HALF));
-          const int p = krow - prefix;
-          if (use_rope && p >= 0)
-            rope_rotate8(lo, hi, sin_t + (long)p * HD, cos_t + (long)p * HD, c0);
-        }
-        *reinterpret_cast<bf16x8*>(&k_lds[row * LDS_STRIDE + c0]) = lo;
-        *reinterpret_cast<bf16x8*>(&k_lds[row * LDS_STRIDE + HALF + c0]) = hi;
-#pragma unroll
-        for (int e = 0; e < 8; ++e) {
-          kt_lds[(c0 + e) * KT_STRIDE + row] = reinterpret_cast<__hip_bfloat16*>(&lo)[e];
-          kt_lds[(c0 + HALF + e) * KT_STRIDE + row] = reinterpret_cast<__hip_bfloat16*>(&hi)[e];
-        }
-      }
-      constexpr int PER_ROW = HD / 8;
-      for (int idx = threadIdx.x; idx < 2 * KVB * PER_ROW; idx += 256) {
-        const int row = idx / PER_ROW;
-        const int c8 = (idx % PER_ROW) * 8;
-        const int krow = kbase0 + row;
-        bf16x8 vv = (krow < N) ? load8(qv.at(krow, 2, c8)) : bf16x8{};
-        *reinterpret_cast<bf16x8*>(&v_lds[row * LDS_STRIDE + c8]) = vv;
-      }
-    }
+    write_tile(kbase0);
+    if (kt + 1 < n_super) issue_loads(kbase0 + 2 * KVB);
     __syncthreads();
 
    for (int sub = 0; sub < 2 && kbase0 + sub * KVB < N; ++sub) {
@@ -526,47 +550,73 @@ __global__ __launch_bounds__(256) void bwd_dkv_kernel(
       dv_acc[t][r] = 0.f;
     }
 
+  constexpr int PAIRS_PER_ROW = HALF / 8;
+  constexpr int PER_ROW = HD / 8;
+  constexpr int QK_ITEMS = QB * PAIRS_PER_ROW;   // <= 256
+  constexpr int DO_ITEMS = QB * PER_ROW;         // <= 512
+  constexpr int DO_IPT = (DO_ITEMS + 255) / 256;
+  bf16x8 qlo{}, qhi{}, doreg[DO_IPT];
+  float lse_reg = INFINITY, d_reg = 0.f;
+  auto issue_loads = [&](int qbase0) {
+    if (threadIdx.x < QK_ITEMS) {
+      const int qrow = qbase0 + threadIdx.x / PAIRS_PER_ROW;
+      const int c0 = (threadIdx.x % PAIRS_PER_ROW) * 8;
+      const bool ok = qrow < N;
+      qlo = ok ? load8(qv.at(qrow, 0, c0)) : bf16x8{};
+      qhi = ok ? load8(qv.at(qrow, 0, c0 + HALF)) : bf16x8{};
+    }
+#pragma unroll
+    for (int j = 0; j < DO_IPT; ++j) {
+      const int idx = threadIdx.x + j * 256;
+      const int qrow = qbase0 + idx / PER_ROW;
+      doreg[j] = (idx < DO_ITEMS && qrow < N)
+                     ? load8(do_base + (long)qrow * do_stride + (idx % PER_ROW) * 8)
+                     : bf16x8{};
+    }
+    if (threadIdx.x < QB) {
+      const int qrow = qbase0 + threadIdx.x;
+      lse_reg = (qrow < N) ? lse[((long)b * H + h) * N + qrow] : INFINITY;
+      d_reg = (qrow < N) ? D[((long)b * H + h) * N + qrow] : 0.f;
+    }
+  };
+  auto write_tile = [&](int qbase0) {
+    if (threadIdx.x < QK_ITEMS) {
+      const int lrow = threadIdx.x / PAIRS_PER_ROW;
+      const int c0 = (threadIdx.x % PAIRS_PER_ROW) * 8;
+      const int qrow = qbase0 + lrow;
+      const int p = qrow - prefix;
+      if (use_rope && qrow < N && p >= 0)
+        rope_rotate8(qlo, qhi, sin_t + (long)p * HD, cos_t + (long)p * HD, c0);
+#pragma unroll
+      for (int e = 0; e < 8; ++e) {
+        qt_lds[(c0 + e) * QT_STRIDE + lrow] = reinterpret_cast<__hip_bfloat16*>(&qlo)[e];
+        qt_lds[(c0 + HALF + e) * QT_STRIDE + lrow] = reinterpret_cast<__hip_bfloat16*>(&qhi)[e];
+      }
+    }
+#pragma unroll
+    for (int j = 0; j < DO_IPT; ++j) {
+      const int idx = threadIdx.x + j * 256;
+      if (idx < DO_ITEMS) {
+        const int lrow = idx / PER_ROW;
+        const int c8 = (idx % PER_ROW) * 8;
+#pragma unroll
+        for (int e = 0; e < 8; ++e)
+          dot_lds[(c8 + e) * QT_STRIDE + lrow] = reinterpret_cast<__hip_bfloat16*>(&doreg[j])[e];
+      }
+    }
+    if (threadIdx.x < QB) {
+      lse_lds[threadIdx.x] = lse_reg;
+      d_lds[threadIdx.x] = d_reg;
+    }
+  };
+
   const int n_q = (N + QB - 1) / QB;
+  issue_loads(0);
   for (int qt = 0; qt < n_q; ++qt) {
     const int qbase0 = qt * QB;
     __syncthreads();
-    {
-      // Q^T staging with rope (pair chunks), dO^T staging
-      constexpr int PAIRS_PER_ROW = HALF / 8;
-      for (int idx = threadIdx.x; idx < QB * PAIRS_PER_ROW; idx += 256) {
-        const int row = idx / PAIRS_PER_ROW;
-        const int c0 = (idx % PAIRS_PER_ROW) * 8;
-        const int qrow = qbase0 + row;
-        bf16x8 lo{}, hi{};
-        if (qrow < N) {
-          lo = load8(qv.at(qrow, 0, c0));
-          hi = load8(qv.at(qrow, 0, c0 + HALF));
-          const int p = qrow - prefix;
-          if (use_rope && p >= 0)
-            rope_rotate8(lo, hi, sin_t + (long)p * HD, cos_t + (long)p * HD, c0);
-        }
-#pragma unroll
-        for (int e = 0; e < 8; ++e) {
-          qt_lds[(c0 + e) * QT_STRIDE + row] = reinterpret_cast<__hip_bfloat16*>(&lo)[e];
-          qt_lds[(c0 + HALF + e) * QT_STRIDE + row] = reinterpret_cast<__hip_bfloat16*>(&hi)[e];
-        }
-      }
-      constexpr int PER_ROW = HD / 8;
-      for (int idx = threadIdx.x; idx < QB * PER_ROW; idx += 256) {
-        const int row = idx / PER_ROW;
-        const int c8 = (idx % PER_ROW) * 8;
-        const int qrow = qbase0 + row;
-        bf16x8 dov = (qrow < N) ? load8(do_base + (long)qrow * do_stride + c8) : bf16x8{};
-#pragma unroll
-        for (int e = 0; e < 8; ++e)
-          dot_lds[(c8 + e) * QT_STRIDE + row] = reinterpret_cast<__hip_bfloat16*>(&dov)[e];
-      }
-      for (int row = threadIdx.x; row < QB; row += 256) {
-        const int qrow = qbase0 + row;
-        lse_lds[row] = (qrow < N) ? lse[((long)b * H + h) * N + qrow] : INFINITY;
-        d_lds[row] = (qrow < N) ? D[((long)b * H + h) * N + qrow] : 0.f;
-      }
-    }
+    write_tile(qbase0);
+    if (qt + 1 < n_q) issue_loads(qbase0 + QB);
     __syncthreads();
 
    {
