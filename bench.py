@@ -17,8 +17,6 @@ import json
 import os
 import sys
 import time
-import types
-
 import torch
 
 REPO_ROOT = os.path.dirname(os.path.abspath(__file__))
@@ -105,10 +103,8 @@ def main():
     os.environ.setdefault("MASTER_ADDR", "127.0.0.1")
 
     from dinov3_amd import parallel
-    from dinov3_amd.parallel.ddp import GradReducer, all_reduce_scalar_sums
     from dinov3_amd.train.ssl_meta_arch import SSLMetaArch
-    from dinov3_amd.train.optim import FusedAdamW
-    from dinov3_amd.train.train import DTYPE_MAP
+    from dinov3_amd.train.train import DTYPE_MAP, build_training_engine
 
     parallel.enable_distributed()
     rank = parallel.get_rank()
@@ -127,8 +123,6 @@ def main():
     model.train()
 
     groups = model.get_params_groups()
-    from dinov3_amd.train.train import build_training_engine
-
     optimizer, finalize_backward = build_training_engine(cfg, groups)
 
     batches = make_synthetic_batch(cfg, device, dtype)

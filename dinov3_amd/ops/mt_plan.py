@@ -67,11 +67,14 @@ def adamw_planned(plan: MultiTensorPlan, lr_mult: torch.Tensor, wd_mult: torch.T
     )
 
 
-def l2norm_planned(grad_plan: MultiTensorPlan, sub_id: torch.Tensor,
-                   n_submodels: int) -> torch.Tensor:
+def l2norm_planned(plan: MultiTensorPlan, sub_id: torch.Tensor, n_submodels: int,
+                   list_index: int = 0) -> torch.Tensor:
+    """Per-submodel sum of squared values over plan list `list_index`."""
     from . import hip_ops
 
-    grad_plan.check_pointers()
-    return hip_ops().multi_tensor_l2norm_planned(grad_plan.ptrs, grad_plan.sizes,
-                                                 grad_plan.ct, grad_plan.co, sub_id,
-                                                 n_submodels, grad_plan.is_bf16)
+    plan.check_pointers()
+    n = plan.n_tensors
+    ptrs = plan.ptrs[list_index * n: (list_index + 1) * n]
+    is_bf16 = plan.lists[list_index][0].dtype == torch.bfloat16
+    return hip_ops().multi_tensor_l2norm_planned(ptrs, plan.sizes, plan.ct, plan.co,
+                                                 sub_id, n_submodels, is_bf16)

@@ -90,19 +90,7 @@ class FusedAdamW:
             from ..ops.mt_plan import l2norm_planned
 
             plan = self._ensure_plan()
-            n = plan.n_tensors
-
-            class _GradView:
-                pass
-
-            gv = _GradView()
-            gv.ptrs = plan.ptrs[n: 2 * n]
-            gv.sizes = plan.sizes
-            gv.ct = plan.ct
-            gv.co = plan.co
-            gv.is_bf16 = plan.is_bf16
-            gv.check_pointers = plan.check_pointers
-            return l2norm_planned(gv, self._sub_id, len(self.submodels))
+            return l2norm_planned(plan, self._sub_id, len(self.submodels), list_index=1)
         sums = torch.zeros(len(self.submodels), dtype=torch.float32)
         for g in self.groups:
             grads = [p.grad for p in g["params"] if p.grad is not None]

@@ -13,11 +13,9 @@ from __future__ import annotations
 
 import argparse
 import logging
-import math
 import sys
-import time
 from functools import partial
-from typing import Dict, Optional
+from typing import Dict
 
 import torch
 
@@ -33,7 +31,7 @@ from ..data import (
     make_dataset,
 )
 from ..logging import MetricLogger, setup_logging
-from ..parallel.ddp import GradReducer, all_reduce_scalar_sums
+from ..parallel.ddp import GradReducer
 from .cosine_lr_scheduler import CosineScheduler, linear_warmup_cosine_decay
 from .optim import FusedAdamW
 from .ssl_meta_arch import SSLMetaArch
