@@ -99,8 +99,8 @@ struct QkvView {
 // ---------------------------------------------------------------------------
 // Forward
 // ---------------------------------------------------------------------------
-template <int HD>
-__global__ __launch_bounds__(256) void fwd_kernel(
+template <int HD, int NT = 256>
+__global__ __launch_bounds__(NT) void fwd_kernel(
     const __hip_bfloat16* __restrict__ qkv, const float* __restrict__ sin_t,
     const float* __restrict__ cos_t, __hip_bfloat16* __restrict__ o,
     float* __restrict__ lse, int B, int H, int N, int P, float scale) {
@@ -118,7 +118,7 @@ __global__ __launch_bounds__(256) void fwd_kernel(
   const int lane = threadIdx.x & 63;
   const int hhalf = lane >> 5;
   const int l31 = lane & 31;
-  const int q0 = blockIdx.y * 128 + wave * 32;
+  const int q0 = blockIdx.y * (NT / 2) + wave * 32;
   const int prefix = N - P;
   const bool use_rope = (sin_t != nullptr);
 
@@ -163,15 +163,15 @@ __global__ __launch_bounds__(256) void fwd_kernel(
   constexpr int PAIRS_PER_ROW = HALF / 8;     // K pair-items per row
   constexpr int PER_ROW = HD / 8;             // V items per row
   constexpr int K_ITEMS = 2 * KVB * PAIRS_PER_ROW;   // 256 (hd64) / 512 (hd128)
-  constexpr int K_ITEMS_PER_THREAD = (K_ITEMS + 255) / 256;
-  constexpr int V_ITEMS_PER_THREAD = (2 * KVB * PER_ROW + 255) / 256;
+  constexpr int K_ITEMS_PER_THREAD = (K_ITEMS + NT - 1) / NT;
+  constexpr int V_ITEMS_PER_THREAD = (2 * KVB * PER_ROW + NT - 1) / NT;
 
   bf16x8 klo[K_ITEMS_PER_THREAD], khi[K_ITEMS_PER_THREAD];
   bf16x8 vreg[V_ITEMS_PER_THREAD];
   auto issue_loads = [&](int kbase0) {
 #pragma unroll
     for (int j = 0; j < K_ITEMS_PER_THREAD; ++j) {
-      const int item = threadIdx.x + j * 256;
+      const int item = threadIdx.x + j * NT;
       const int krow = kbase0 + item / PAIRS_PER_ROW;
       const int c0 = (item % PAIRS_PER_ROW) * 8;
       const bool ok = item < K_ITEMS && krow < N;
@@ -180,7 +180,7 @@ __global__ __launch_bounds__(256) void fwd_kernel(
     }
 #pragma unroll
     for (int j = 0; j < V_ITEMS_PER_THREAD; ++j) {
-      const int idx = threadIdx.x + j * 256;
+      const int idx = threadIdx.x + j * NT;
       const int row = idx / PER_ROW;
       const int krow = kbase0 + row;
       vreg[j] = (idx < 2 * KVB * PER_ROW && krow < N) ? load8(qv.at(krow, 2, (idx % PER_ROW) * 8))
@@ -190,7 +190,7 @@ __global__ __launch_bounds__(256) void fwd_kernel(
   auto write_tile = [&](int kbase0) {
 #pragma unroll
     for (int j = 0; j < K_ITEMS_PER_THREAD; ++j) {
-      const int item = threadIdx.x + j * 256;
+      const int item = threadIdx.x + j * NT;
       if (item < K_ITEMS) {
         const int lrow = item / PAIRS_PER_ROW;
         const int c0 = (item % PAIRS_PER_ROW) * 8;
@@ -204,7 +204,7 @@ __global__ __launch_bounds__(256) void fwd_kernel(
     }
 #pragma unroll
     for (int j = 0; j < V_ITEMS_PER_THREAD; ++j) {
-      const int idx = threadIdx.x + j * 256;
+      const int idx = threadIdx.x + j * NT;
       if (idx < 2 * KVB * PER_ROW) {
         const int row = idx / PER_ROW;
         const int c8 = (idx % PER_ROW) * 8;
@@ -296,8 +296,8 @@ __global__ __launch_bounds__(256) void fwd_kernel(
 // ---------------------------------------------------------------------------
 // Backward dQ
 // ---------------------------------------------------------------------------
-template <int HD>
-__global__ __launch_bounds__(256) void bwd_dq_kernel(
+template <int HD, int NT = 256>
+__global__ __launch_bounds__(NT) void bwd_dq_kernel(
     const __hip_bfloat16* __restrict__ qkv, const __hip_bfloat16* __restrict__ dout,
     const float* __restrict__ sin_t, const float* __restrict__ cos_t,
     const float* __restrict__ lse, const float* __restrict__ D,
@@ -316,7 +316,7 @@ __global__ __launch_bounds__(256) void bwd_dq_kernel(
   const int lane = threadIdx.x & 63;
   const int hhalf = lane >> 5;
   const int l31 = lane & 31;
-  const int q0 = blockIdx.y * 128 + wave * 32;
+  const int q0 = blockIdx.y * (NT / 2) + wave * 32;
   const int prefix = N - P;
   const bool use_rope = (sin_t != nullptr);
 
@@ -362,13 +362,13 @@ __global__ __launch_bounds__(256) void bwd_dq_kernel(
   constexpr int PAIRS_PER_ROW = HALF / 8;
   constexpr int PER_ROW = HD / 8;
   constexpr int K_ITEMS = 2 * KVB * PAIRS_PER_ROW;
-  constexpr int K_IPT = (K_ITEMS + 255) / 256;
-  constexpr int V_IPT = (2 * KVB * PER_ROW + 255) / 256;
+  constexpr int K_IPT = (K_ITEMS + NT - 1) / NT;
+  constexpr int V_IPT = (2 * KVB * PER_ROW + NT - 1) / NT;
   bf16x8 klo[K_IPT], khi[K_IPT], vreg[V_IPT];
   auto issue_loads = [&](int kbase0) {
 #pragma unroll
     for (int j = 0; j < K_IPT; ++j) {
-      const int item = threadIdx.x + j * 256;
+      const int item = threadIdx.x + j * NT;
       const int krow = kbase0 + item / PAIRS_PER_ROW;
       const int c0 = (item % PAIRS_PER_ROW) * 8;
       const bool ok = item < K_ITEMS && krow < N;
@@ -377,7 +377,7 @@ __global__ __launch_bounds__(256) void bwd_dq_kernel(
     }
 #pragma unroll
     for (int j = 0; j < V_IPT; ++j) {
-      const int idx = threadIdx.x + j * 256;
+      const int idx = threadIdx.x + j * NT;
       const int krow = kbase0 + idx / PER_ROW;
       vreg[j] = (idx < 2 * KVB * PER_ROW && krow < N)
                     ? load8(qv.at(krow, 2, (idx % PER_ROW) * 8)) : bf16x8{};
@@ -386,7 +386,7 @@ __global__ __launch_bounds__(256) void bwd_dq_kernel(
   auto write_tile = [&](int kbase0) {
 #pragma unroll
     for (int j = 0; j < K_IPT; ++j) {
-      const int item = threadIdx.x + j * 256;
+      const int item = threadIdx.x + j * NT;
       if (item < K_ITEMS) {
         const int lrow = item / PAIRS_PER_ROW;
         const int c0 = (item % PAIRS_PER_ROW) * 8;
@@ -405,7 +405,7 @@ __global__ __launch_bounds__(256) void bwd_dq_kernel(
     }
 #pragma unroll
     for (int j = 0; j < V_IPT; ++j) {
-      const int idx = threadIdx.x + j * 256;
+      const int idx = threadIdx.x + j * NT;
       if (idx < 2 * KVB * PER_ROW)
         *reinterpret_cast<bf16x8*>(&v_lds[(idx / PER_ROW) * LDS_STRIDE + (idx % PER_ROW) * 8]) = vreg[j];
     }
@@ -486,8 +486,8 @@ __global__ __launch_bounds__(256) void bwd_dq_kernel(
 // ---------------------------------------------------------------------------
 // Backward dK/dV
 // ---------------------------------------------------------------------------
-template <int HD>
-__global__ __launch_bounds__(256) void bwd_dkv_kernel(
+template <int HD, int NT = 256>
+__global__ __launch_bounds__(NT) void bwd_dkv_kernel(
     const __hip_bfloat16* __restrict__ qkv, const __hip_bfloat16* __restrict__ dout,
     const float* __restrict__ sin_t, const float* __restrict__ cos_t,
     const float* __restrict__ lse, const float* __restrict__ D,
@@ -554,7 +554,7 @@ __global__ __launch_bounds__(256) void bwd_dkv_kernel(
   constexpr int PER_ROW = HD / 8;
   constexpr int QK_ITEMS = QB * PAIRS_PER_ROW;   // <= 256
   constexpr int DO_ITEMS = QB * PER_ROW;         // <= 512
-  constexpr int DO_IPT = (DO_ITEMS + 255) / 256;
+  constexpr int DO_IPT = (DO_ITEMS + NT - 1) / NT;
   bf16x8 qlo{}, qhi{}, doreg[DO_IPT];
   float lse_reg = INFINITY, d_reg = 0.f;
   auto issue_loads = [&](int qbase0) {
@@ -567,7 +567,7 @@ __global__ __launch_bounds__(256) void bwd_dkv_kernel(
     }
 #pragma unroll
     for (int j = 0; j < DO_IPT; ++j) {
-      const int idx = threadIdx.x + j * 256;
+      const int idx = threadIdx.x + j * NT;
       const int qrow = qbase0 + idx / PER_ROW;
       doreg[j] = (idx < DO_ITEMS && qrow < N)
                      ? load8(do_base + (long)qrow * do_stride + (idx % PER_ROW) * 8)
@@ -595,7 +595,7 @@ __global__ __launch_bounds__(256) void bwd_dkv_kernel(
     }
 #pragma unroll
     for (int j = 0; j < DO_IPT; ++j) {
-      const int idx = threadIdx.x + j * 256;
+      const int idx = threadIdx.x + j * NT;
       if (idx < DO_ITEMS) {
         const int lrow = idx / PER_ROW;
         const int c8 = (idx % PER_ROW) * 8;
@@ -750,15 +750,25 @@ __global__ void bwd_pre_tm_kernel(const __hip_bfloat16* __restrict__ dout,
 void launch_fmha_rope_fwd(const __hip_bfloat16* qkv, const float* sin_t,
                           const float* cos_t, __hip_bfloat16* o, float* lse, int B, int H,
                           int N, int P, int HD, float scale, hipStream_t stream) {
-  dim3 grid(B * H, (N + 127) / 128);
+  // short sequences (local crops, N <= 64): 128-thread blocks, 64 q rows each
+  const bool small = N <= 64;
+  dim3 grid(B * H, small ? 1 : (N + 127) / 128);
   if (HD == 64) {
     size_t shmem = (64 * 72 + 64 * 72) * sizeof(__hip_bfloat16);
-    hipLaunchKernelGGL((fmha_rope::fwd_kernel<64>), grid, dim3(256), shmem, stream, qkv,
-                       sin_t, cos_t, o, lse, B, H, N, P, scale);
+    if (small)
+      hipLaunchKernelGGL(HIP_KERNEL_NAME(fmha_rope::fwd_kernel<64, 128>), grid, dim3(128),
+                         shmem, stream, qkv, sin_t, cos_t, o, lse, B, H, N, P, scale);
+    else
+      hipLaunchKernelGGL(HIP_KERNEL_NAME(fmha_rope::fwd_kernel<64, 256>), grid, dim3(256),
+                         shmem, stream, qkv, sin_t, cos_t, o, lse, B, H, N, P, scale);
   } else if (HD == 128) {
     size_t shmem = (64 * 136 + 128 * 72) * sizeof(__hip_bfloat16);
-    hipLaunchKernelGGL((fmha_rope::fwd_kernel<128>), grid, dim3(256), shmem, stream, qkv,
-                       sin_t, cos_t, o, lse, B, H, N, P, scale);
+    if (small)
+      hipLaunchKernelGGL(HIP_KERNEL_NAME(fmha_rope::fwd_kernel<128, 128>), grid, dim3(128),
+                         shmem, stream, qkv, sin_t, cos_t, o, lse, B, H, N, P, scale);
+    else
+      hipLaunchKernelGGL(HIP_KERNEL_NAME(fmha_rope::fwd_kernel<128, 256>), grid, dim3(256),
+                         shmem, stream, qkv, sin_t, cos_t, o, lse, B, H, N, P, scale);
   }
 }
 
@@ -774,15 +784,28 @@ void launch_fmha_rope_bwd_dq(const __hip_bfloat16* qkv, const __hip_bfloat16* do
                              const float* sin_t, const float* cos_t, const float* lse,
                              const float* D, __hip_bfloat16* dqkv, int B, int H, int N,
                              int P, int HD, float scale, hipStream_t stream) {
-  dim3 grid(B * H, (N + 127) / 128);
+  const bool small = N <= 64;
+  dim3 grid(B * H, small ? 1 : (N + 127) / 128);
   if (HD == 64) {
     size_t shmem = (2 * 64 * 72 + 64 * 72) * sizeof(__hip_bfloat16);
-    hipLaunchKernelGGL((fmha_rope::bwd_dq_kernel<64>), grid, dim3(256), shmem, stream, qkv,
-                       dout, sin_t, cos_t, lse, D, dqkv, B, H, N, P, scale);
+    if (small)
+      hipLaunchKernelGGL(HIP_KERNEL_NAME(fmha_rope::bwd_dq_kernel<64, 128>), grid,
+                         dim3(128), shmem, stream, qkv, dout, sin_t, cos_t, lse, D, dqkv,
+                         B, H, N, P, scale);
+    else
+      hipLaunchKernelGGL(HIP_KERNEL_NAME(fmha_rope::bwd_dq_kernel<64, 256>), grid,
+                         dim3(256), shmem, stream, qkv, dout, sin_t, cos_t, lse, D, dqkv,
+                         B, H, N, P, scale);
   } else if (HD == 128) {
     size_t shmem = (2 * 64 * 136 + 128 * 72) * sizeof(__hip_bfloat16);
-    hipLaunchKernelGGL((fmha_rope::bwd_dq_kernel<128>), grid, dim3(256), shmem, stream,
-                       qkv, dout, sin_t, cos_t, lse, D, dqkv, B, H, N, P, scale);
+    if (small)
+      hipLaunchKernelGGL(HIP_KERNEL_NAME(fmha_rope::bwd_dq_kernel<128, 128>), grid,
+                         dim3(128), shmem, stream, qkv, dout, sin_t, cos_t, lse, D, dqkv,
+                         B, H, N, P, scale);
+    else
+      hipLaunchKernelGGL(HIP_KERNEL_NAME(fmha_rope::bwd_dq_kernel<128, 256>), grid,
+                         dim3(256), shmem, stream, qkv, dout, sin_t, cos_t, lse, D, dqkv,
+                         B, H, N, P, scale);
   }
 }
 
@@ -790,14 +813,27 @@ void launch_fmha_rope_bwd_dkv(const __hip_bfloat16* qkv, const __hip_bfloat16* d
                               const float* sin_t, const float* cos_t, const float* lse,
                               const float* D, __hip_bfloat16* dqkv, int B, int H, int N,
                               int P, int HD, float scale, hipStream_t stream) {
-  dim3 grid(B * H, (N + 127) / 128);
+  const bool small = N <= 64;
+  dim3 grid(B * H, small ? 1 : (N + 127) / 128);
   if (HD == 64) {
     size_t shmem = 2 * 64 * 72 * sizeof(__hip_bfloat16) + 128 * sizeof(float);
-    hipLaunchKernelGGL((fmha_rope::bwd_dkv_kernel<64>), grid, dim3(256), shmem, stream,
-                       qkv, dout, sin_t, cos_t, lse, D, dqkv, B, H, N, P, scale);
+    if (small)
+      hipLaunchKernelGGL(HIP_KERNEL_NAME(fmha_rope::bwd_dkv_kernel<64, 128>), grid,
+                         dim3(128), shmem, stream, qkv, dout, sin_t, cos_t, lse, D, dqkv,
+                         B, H, N, P, scale);
+    else
+      hipLaunchKernelGGL(HIP_KERNEL_NAME(fmha_rope::bwd_dkv_kernel<64, 256>), grid,
+                         dim3(256), shmem, stream, qkv, dout, sin_t, cos_t, lse, D, dqkv,
+                         B, H, N, P, scale);
   } else if (HD == 128) {
     size_t shmem = 2 * 128 * 72 * sizeof(__hip_bfloat16) + 128 * sizeof(float);
-    hipLaunchKernelGGL((fmha_rope::bwd_dkv_kernel<128>), grid, dim3(256), shmem, stream,
-                       qkv, dout, sin_t, cos_t, lse, D, dqkv, B, H, N, P, scale);
+    if (small)
+      hipLaunchKernelGGL(HIP_KERNEL_NAME(fmha_rope::bwd_dkv_kernel<128, 128>), grid,
+                         dim3(128), shmem, stream, qkv, dout, sin_t, cos_t, lse, D, dqkv,
+                         B, H, N, P, scale);
+    else
+      hipLaunchKernelGGL(HIP_KERNEL_NAME(fmha_rope::bwd_dkv_kernel<128, 256>), grid,
+                         dim3(256), shmem, stream, qkv, dout, sin_t, cos_t, lse, D, dqkv,
+                         B, H, N, P, scale);
   }
 }
