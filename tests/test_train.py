@@ -107,3 +107,40 @@ def test_ema_update_moves_teacher():
     ema_update_(t, s, momentum=0.9)
     assert torch.allclose(t[0], torch.full((4,), 0.9))
     assert torch.allclose(t[1], torch.full((3,), 0.1))
+
+
+def test_do_train_resume(tmp_path, smoke_cfg):
+    import copy
+
+    from dinov3_amd.checkpointer import find_latest_checkpoint
+    from dinov3_amd.train.ssl_meta_arch import SSLMetaArch
+    from dinov3_amd.train.train import do_train
+
+    cfg = copy.deepcopy(smoke_cfg)
+    cfg.train.output_dir = str(tmp_path)
+    cfg.checkpointing.period = 2
+    cfg.checkpointing.max_to_keep = 2
+    torch.manual_seed(0)
+    model = SSLMetaArch(cfg)
+    do_train(cfg, model, resume=False, max_iterations=3)
+    latest = find_latest_checkpoint(str(tmp_path))
+    assert latest is not None
+    it_after_first = int(latest.name)
+    # resume continues from the saved iteration
+    torch.manual_seed(0)
+    model2 = SSLMetaArch(cfg)
+    do_train(cfg, model2, resume=True, max_iterations=2)
+    latest2 = find_latest_checkpoint(str(tmp_path))
+    assert int(latest2.name) > it_after_first
+
+
+def test_scheduler_consistency_with_config(smoke_cfg):
+    from dinov3_amd.train.train import build_schedulers
+
+    s = build_schedulers(smoke_cfg)
+    total = smoke_cfg.optim.epochs * smoke_cfg.train.OFFICIAL_EPOCH_LENGTH
+    assert s["total_iterations"] == total
+    assert s["lr"][total - 1] <= smoke_cfg.optim.lr
+    assert 0.0 <= s["momentum"][0] <= 1.0
+    # teacher temp warms from warmup_teacher_temp to teacher_temp
+    assert abs(s["teacher_temp"][total + 10] - smoke_cfg.teacher.teacher_temp) < 1e-9
