@@ -29,18 +29,24 @@ class PatchEmbed(nn.Module):
         self.norm = norm_layer if norm_layer is not None else nn.Identity()
 
     def patchify(self, x: torch.Tensor) -> Tuple[torch.Tensor, int, int]:
-        """[B, C, H, W] -> [B, H'/p * W'/p, p*p*C] rows in (ph, pw, c) order."""
+        """[B, C, H, W] -> [B, H'/p * W'/p, C*p*p] rows in (c, dy, dx) order —
+        the conv-weight-native flattening, so Meta conv weights map by a plain
+        reshape and each 16-element k-slice is one contiguous pixel run."""
         B, C, H, W = x.shape
         p = self.patch_size
         assert H % p == 0 and W % p == 0, f"input {H}x{W} not divisible by patch {p}"
         hp, wp = H // p, W // p
         x = x.reshape(B, C, hp, p, wp, p)
-        x = x.permute(0, 2, 4, 3, 5, 1)  # B, hp, wp, p, p, C
-        return x.reshape(B, hp * wp, p * p * C), hp, wp
+        x = x.permute(0, 2, 4, 1, 3, 5)  # B, hp, wp, C, p, p
+        return x.reshape(B, hp * wp, C * p * p), hp, wp
 
     def forward(self, x: torch.Tensor) -> Tuple[torch.Tensor, int, int]:
-        rows, hp, wp = self.patchify(x)
-        out = self.proj(rows.to(self.proj.weight.dtype))
+        from ..ops.patch_embed_op import patch_embed_gemm
+
+        B, C, H, W = x.shape
+        p = self.patch_size
+        hp, wp = H // p, W // p
+        out = patch_embed_gemm(x, self.proj.weight, self.proj.bias, p, self.patchify)
         out = self.norm(out)
         if not self.flatten_embedding:
             out = out.reshape(out.shape[0], hp, wp, self.embed_dim)

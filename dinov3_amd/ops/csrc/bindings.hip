@@ -93,6 +93,9 @@ void launch_fmha_rope_bwd_dq(const __hip_bfloat16*, const __hip_bfloat16*, const
 void launch_fmha_rope_bwd_dkv(const __hip_bfloat16*, const __hip_bfloat16*, const float*,
                               const float*, const float*, const float*, __hip_bfloat16*,
                               int, int, int, int, int, float, hipStream_t);
+void launch_patch_embed_fwd(const __hip_bfloat16*, const __hip_bfloat16*,
+                            const __hip_bfloat16*, __hip_bfloat16*, int, int, int, int,
+                            int, int, hipStream_t);
 void launch_probe_mfma(const __hip_bfloat16*, const __hip_bfloat16*, float*, hipStream_t);
 void launch_fmha_fwd(const __hip_bfloat16*, const __hip_bfloat16*, const __hip_bfloat16*,
                      __hip_bfloat16*, float*, int, int, int, float, hipStream_t);
@@ -632,6 +635,26 @@ std::vector<torch::Tensor> fmha_bwd(torch::Tensor dout, torch::Tensor q, torch::
   return {dq, dk, dv};
 }
 
+// ----------------------------- patch embed ------------------------------
+
+torch::Tensor patch_embed_fwd(torch::Tensor x, torch::Tensor w, torch::Tensor bias,
+                              int64_t patch) {
+  CHECK_INPUT(x);
+  CHECK_INPUT(w);
+  TORCH_CHECK(x.dim() == 4 && x.scalar_type() == at::ScalarType::BFloat16);
+  const int B = x.size(0), C = x.size(1), H = x.size(2), W = x.size(3);
+  const int D = w.size(0);
+  TORCH_CHECK(patch == 16 && C == 3, "patch_embed_fwd kernel supports P=16, C=3");
+  const long rows = (long)B * (H / patch) * (W / patch);
+  auto out = torch::empty({(long)B, rows / B, (long)D}, x.options());
+  launch_patch_embed_fwd((const __hip_bfloat16*)x.data_ptr(),
+                         (const __hip_bfloat16*)w.data_ptr(),
+                         (const __hip_bfloat16*)bias.data_ptr(),
+                         (__hip_bfloat16*)out.data_ptr(), B, C, H, W, (int)patch, D,
+                         current_stream());
+  return out;
+}
+
 // ----------------------------- fmha + rope ------------------------------
 
 std::vector<torch::Tensor> fmha_rope_fwd_out(torch::Tensor qkv, torch::Tensor sin_t,
@@ -930,6 +953,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, mod) {
   mod.def("swiglu_bwd", &swiglu_bwd);
   mod.def("rope_fwd", &rope_fwd);
   mod.def("probe_mfma", &probe_mfma);
+  mod.def("patch_embed_fwd", &patch_embed_fwd);
   mod.def("dino_ce_fwd", &dino_ce_fwd);
   mod.def("dino_ce_bwd", &dino_ce_bwd);
   mod.def("ibot_ce_fwd", &ibot_ce_fwd);

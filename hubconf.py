@@ -27,9 +27,9 @@ def convert_meta_state_dict(sd: dict, patch_size: int = 16, in_chans: int = 3) -
         if key.endswith("bias_mask"):  # mask_k_bias NaN-buffer: ours is derived
             continue
         if key == "patch_embed.proj.weight" and value.ndim == 4:
-            # conv [D, C, ph, pw] -> linear [D, ph*pw*C] with (ph, pw, c) rows
-            d = value.shape[0]
-            value = value.permute(0, 2, 3, 1).reshape(d, -1)
+            # conv [D, C, ph, pw] -> linear [D, C*ph*pw]: plain reshape (our
+            # patch rows use the conv-native (c, dy, dx) flattening)
+            value = value.reshape(value.shape[0], -1)
         if key.endswith("last_layer.weight_v"):
             # collapse weight-norm parametrization if present
             g = sd.get(key.replace("weight_v", "weight_g"))

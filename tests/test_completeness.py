@@ -75,8 +75,8 @@ def test_hubconf_roundtrip(tmp_path):
     m = hubconf.dinov3_vits16()
     # simulate a Meta-style state dict: conv-shaped patch embed
     sd = m.state_dict()
-    w = sd["patch_embed.proj.weight"]  # [D, p*p*C]
-    conv_w = w.reshape(384, 16, 16, 3).permute(0, 3, 1, 2).contiguous()
+    w = sd["patch_embed.proj.weight"]  # [D, C*p*p], conv-native flattening
+    conv_w = w.reshape(384, 3, 16, 16).contiguous()
     meta_sd = dict(sd)
     meta_sd["patch_embed.proj.weight"] = conv_w
     converted = hubconf.convert_meta_state_dict(meta_sd)

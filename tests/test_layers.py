@@ -25,9 +25,8 @@ def test_patch_embed_matches_conv():
     assert out.shape == (2, 4, 64) and hp == wp == 2
     # equivalence with a stride-p conv carrying the same weights
     conv = torch.nn.Conv2d(3, 64, 16, stride=16)
-    # our rows are (ph, pw, c) ordered; conv weight is [D, C, ph, pw]
-    w = pe.proj.weight.reshape(64, 16, 16, 3).permute(0, 3, 1, 2)
-    conv.weight.data.copy_(w)
+    # conv-native (c, dy, dx) row flattening: conv weight is a plain reshape
+    conv.weight.data.copy_(pe.proj.weight.reshape(64, 3, 16, 16))
     conv.bias.data.copy_(pe.proj.bias)
     ref = conv(x).flatten(2).transpose(1, 2)
     assert torch.allclose(out, ref, atol=1e-4)

@@ -291,3 +291,29 @@ def test_gather_rows_backward(ops):
     ref = torch.zeros_like(flat, dtype=torch.float32)
     ref[idx] = 2 * flat.detach()[idx].float()
     _assert_close(flat.grad, ref, atol=0.05, what="gather dflat")
+
+
+def test_patch_embed_gemm_matches_reference(ops):
+    torch.manual_seed(13)
+    B, C, H, W, D, P = 3, 3, 224, 224, 1024, 16
+    x = torch.randn(B, C, H, W, device=DEV).bfloat16()
+    w = (torch.randn(D, C * P * P, device=DEV) * 0.02).bfloat16()
+    b = torch.randn(D, device=DEV).bfloat16()
+    out = ops.patch_embed_fwd(x, w, b, P)
+    # fp32 reference with the conv-native (c, dy, dx) flattening
+    xr = x.float().reshape(B, C, H // P, P, W // P, P).permute(0, 2, 4, 1, 3, 5)
+    rows = xr.reshape(B, (H // P) * (W // P), C * P * P)
+    ref = rows @ w.float().T + b.float()
+    _assert_close(out, ref, atol=0.15, rtol=2e-2, what="patch_embed gemm")
+
+
+def test_patch_embed_autograd_wgrad(ops):
+    torch.manual_seed(14)
+    from dinov3_amd.layers.patch_embed import PatchEmbed
+
+    pe = PatchEmbed(img_size=64, patch_size=16, embed_dim=128).to(DEV).bfloat16()
+    x = torch.randn(2, 3, 64, 64, device=DEV).bfloat16()
+    out, hp, wp = pe(x)
+    out.float().pow(2).sum().backward()
+    assert pe.proj.weight.grad is not None and torch.isfinite(pe.proj.weight.grad).all()
+    assert pe.proj.bias.grad is not None
