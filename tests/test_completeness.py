@@ -127,3 +127,11 @@ def test_do_test_eval(smoke_cfg, tmp_path):
     model = SSLMetaArch(cfg)
     results = do_test(cfg, model, iteration=0)
     assert "knn_top1" in results and "linear_top1" in results
+
+
+def test_clip_tokenizer_roundtrip():
+    from dinov3_amd.thirdparty import SimpleTokenizer
+
+    t = SimpleTokenizer()
+    for text in ("a photo of a cat", "hello, world!"):
+        assert t.decode(t.encode(text)) == text
