@@ -230,7 +230,7 @@ def batch_to_device(data: dict, device: torch.device) -> dict:
 
 
 def do_train(cfg, model: SSLMetaArch, resume: bool = True, max_iterations: int = -1,
-             record_losses_to: str = "", compare_losses_to: str = ""):
+             record_losses_to: str = "", compare_losses_to: str = "", profiling: bool = False):
     device = parallel.device()
     param_dtype = DTYPE_MAP.get(cfg.compute_precision.param_dtype, torch.float32)
     if device.type == "cuda":
@@ -298,6 +298,8 @@ def do_train(cfg, model: SSLMetaArch, resume: bool = True, max_iterations: int =
         last_layer_lr = 0.0 if it < schedulers["freeze_last_layer_iterations"] else lr
 
         data = batch_to_device(data, device)
+        if profiling and device.type == "cuda":
+            torch.cuda.nvtx.range_push(f"step_{it}")  # roctx range on ROCm
         try:
             loss, loss_dict = model(data, teacher_temp=teacher_temp, iteration=it)
         except torch.OutOfMemoryError:
@@ -313,7 +315,7 @@ def do_train(cfg, model: SSLMetaArch, resume: bool = True, max_iterations: int =
 
         if not torch.isfinite(loss):
             nan_count += 1
-            nan_logger = logging.getLogger("nan")
+            nan_logger = logging.getLogger("dinov3.nan")
             nan_logger.error("NaN/Inf loss at iteration %d (%d consecutive)", it, nan_count)
             if nan_count > 2:
                 raise FloatingPointError(f"aborting: >{nan_count - 1} consecutive non-finite losses")
@@ -340,6 +342,8 @@ def do_train(cfg, model: SSLMetaArch, resume: bool = True, max_iterations: int =
         optimizer.step(lr=lr, weight_decay=wd, last_layer_lr=last_layer_lr, clip_scales=clip_scales)
         optimizer.zero_grad()
         model.update_ema(mom)
+        if profiling and device.type == "cuda":
+            torch.cuda.nvtx.range_pop()
 
         if model.gram_use_loss and cfg.gram.rep_update and it >= cfg.gram.it_first_update:
             if (it - cfg.gram.it_first_update) % cfg.gram.update_frequency == 0:
@@ -364,6 +368,16 @@ def do_train(cfg, model: SSLMetaArch, resume: bool = True, max_iterations: int =
                     if k in ref and abs(v - ref[k]) > 0.05 * max(1.0, abs(ref[k])):
                         logger.warning("loss parity drift at it=%d %s: %g vs ref %g",
                                        it, k, v, ref[k])
+
+        eval_period = cfg.evaluation.eval_period_iterations
+        if eval_period and eval_period > 0 and (it + 1) % eval_period == 0:
+            try:
+                results = do_test(cfg, model, it)
+                metric_logger.update(**{f"eval_{k}": v for k, v in results.items()
+                                        if isinstance(v, (int, float))})
+            except Exception as e:  # eval must never kill training
+                logger.warning("periodic eval failed at it=%d: %s", it, e)
+            model.train()
 
         if output_dir and ckpt_period > 0 and (it + 1) % ckpt_period == 0:
             save_checkpoint(
@@ -445,7 +459,8 @@ def main(argv=None):
         record_to = os.path.join(args.output_dir or ".", "ref_losses.json")
     t0 = time.time()
     result = do_train(cfg, model, resume=not args.no_resume, max_iterations=args.max_iterations,
-                      record_losses_to=record_to, compare_losses_to=args.ref_losses_path)
+                      record_losses_to=record_to, compare_losses_to=args.ref_losses_path,
+                      profiling=args.profiling)
     if args.benchmark_codebase:
         elapsed = time.time() - t0
         logger.info("benchmark: total wall %.1f s (use bench.py for the timed-step metric)", elapsed)
