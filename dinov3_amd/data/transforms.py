@@ -89,36 +89,24 @@ def adjust_saturation(img: torch.Tensor, factor: float) -> torch.Tensor:
 
 
 def adjust_hue(img: torch.Tensor, hue_shift: float) -> torch.Tensor:
-    """hue_shift in [-0.5, 0.5] turns of the hue wheel."""
+    """hue_shift in [-0.5, 0.5] turns of the hue wheel (branch-free HSV trip)."""
     r, g, b = img[0], img[1], img[2]
     maxc, _ = img.max(dim=0)
     minc, _ = img.min(dim=0)
     v = maxc
     deltac = maxc - minc
-    s = torch.where(maxc > 0, deltac / maxc.clamp_min(1e-8), torch.zeros_like(maxc))
+    s = deltac / maxc.clamp_min(1e-8)
     dz = deltac.clamp_min(1e-8)
     rc = (maxc - r) / dz
     gc = (maxc - g) / dz
     bc = (maxc - b) / dz
     h = torch.where(r == maxc, bc - gc, torch.where(g == maxc, 2.0 + rc - bc, 4.0 + gc - rc))
-    h = (h / 6.0) % 1.0
-    h = torch.where(deltac > 0, h, torch.zeros_like(h))
-    h = (h + hue_shift) % 1.0
-    # hsv -> rgb
-    i = (h * 6.0).floor()
-    f = h * 6.0 - i
-    p = v * (1.0 - s)
-    q = v * (1.0 - f * s)
-    t = v * (1.0 - (1.0 - f) * s)
-    i = i.long() % 6
+    h6 = (h % 6.0) + hue_shift * 6.0
+    # hsv -> rgb without per-sector masks: c(n) = v - v*s*clamp(min(k, 4-k), 0, 1)
     out = torch.empty_like(img)
-    masks = [i == k for k in range(6)]
-    rgb_cases = [(v, t, p), (q, v, p), (p, v, t), (p, q, v), (t, p, v), (v, p, q)]
-    for ch in range(3):
-        acc = torch.zeros_like(v)
-        for k in range(6):
-            acc = torch.where(masks[k], rgb_cases[k][ch], acc)
-        out[ch] = acc
+    for ch, n in enumerate((5.0, 3.0, 1.0)):
+        k = (n + h6) % 6.0
+        out[ch] = v - v * s * torch.clamp(torch.minimum(k, 4.0 - k), 0.0, 1.0)
     return out.clamp(0.0, 1.0)
 
 
