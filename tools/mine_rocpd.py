@@ -1,0 +1,71 @@
+"""Mine a rocprofv3 --output-format rocpd SQLite DB into a per-kernel stats
+table (name, calls, total ms, mean us, % of kernel time), schema-agnostically:
+finds the kernel-dispatch table by column inspection so it works across
+rocprofv3 minor versions.
+
+Usage: python tools/mine_rocpd.py <db-path-or-dir> [top_n]
+"""
+
+import glob
+import os
+import sqlite3
+import sys
+
+
+def find_db(path):
+    if os.path.isfile(path):
+        return path
+    hits = sorted(glob.glob(os.path.join(path, "**", "*.db"), recursive=True))
+    if not hits:
+        raise SystemExit(f"no .db under {path}")
+    return hits[0]
+
+
+def main():
+    db = find_db(sys.argv[1])
+    top_n = int(sys.argv[2]) if len(sys.argv) > 2 else 40
+    con = sqlite3.connect(db)
+    cur = con.cursor()
+    tables = [r[0] for r in cur.execute(
+        "SELECT name FROM sqlite_master WHERE type IN ('table','view')")]
+
+    # Prefer a ready-made kernel view if this rocprofv3 ships one.
+    for t in tables:
+        cols = [c[1].lower() for c in cur.execute(f"PRAGMA table_info('{t}')")]
+        if "kernel" not in t.lower():
+            continue
+        name_col = next((c for c in cols if "name" in c and "display" not in c), None)
+        start = next((c for c in cols if c in ("start", "start_timestamp", "begin")), None)
+        end = next((c for c in cols if c in ("end", "end_timestamp")), None)
+        dur = next((c for c in cols if "duration" in c), None)
+        if name_col and (dur or (start and end)):
+            expr = dur if dur else f"({end} - {start})"
+            rows = cur.execute(
+                f"SELECT {name_col}, COUNT(*), SUM({expr}), AVG({expr}) "
+                f"FROM '{t}' GROUP BY {name_col} ORDER BY SUM({expr}) DESC"
+            ).fetchall()
+            if rows and rows[0][2]:
+                report(rows, top_n, t)
+                return
+    raise SystemExit(f"no kernel table found; tables: {tables}")
+
+
+def short(name, width=86):
+    name = name.split("(")[0].strip()
+    for p in ("void ", "at::native::", "dinov3::"):
+        name = name.replace(p, "")
+    return name[:width]
+
+
+def report(rows, top_n, table):
+    total = sum(r[2] for r in rows)
+    print(f"# kernel table: {table}; {sum(r[1] for r in rows)} dispatches, "
+          f"{total / 1e6:.3f} ms total kernel time")
+    print(f"{'kernel':<88} {'calls':>6} {'total_ms':>9} {'mean_us':>8} {'%':>6}")
+    for name, calls, tot, avg in rows[:top_n]:
+        print(f"{short(name):<88} {calls:>6} {tot / 1e6:>9.3f} {avg / 1e3:>8.1f} "
+              f"{100.0 * tot / total:>6.2f}")
+
+
+if __name__ == "__main__":
+    main()
