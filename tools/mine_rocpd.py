@@ -45,9 +45,29 @@ def main():
                 f"FROM '{t}' GROUP BY {name_col} ORDER BY SUM({expr}) DESC"
             ).fetchall()
             if rows and rows[0][2]:
+                if isinstance(rows[0][0], int):  # interned name -> strings table
+                    lut = string_table(cur, tables)
+                    rows = [(lut.get(r[0], f"id:{r[0]}"),) + r[1:] for r in rows]
                 report(rows, top_n, t)
                 return
     raise SystemExit(f"no kernel table found; tables: {tables}")
+
+
+def string_table(cur, tables):
+    """Map id -> string from whichever interning table this DB carries."""
+    for t in tables:
+        if "string" not in t.lower() and "symbol" not in t.lower() and "info" not in t.lower():
+            continue
+        cols = [c[1].lower() for c in cur.execute(f"PRAGMA table_info('{t}')")]
+        idc = next((c for c in cols if c in ("id", "string_id", "rowid")), None)
+        sc = next((c for c in cols if c in ("string", "value", "name", "display_name",
+                                            "kernel_name", "formatted_kernel_name")), None)
+        if idc and sc:
+            lut = {r[0]: r[1] for r in cur.execute(f"SELECT {idc}, {sc} FROM '{t}'")
+                   if isinstance(r[1], str)}
+            if lut:
+                return lut
+    return {}
 
 
 def short(name, width=86):
