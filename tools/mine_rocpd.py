@@ -29,28 +29,41 @@ def main():
     tables = [r[0] for r in cur.execute(
         "SELECT name FROM sqlite_master WHERE type IN ('table','view')")]
 
-    # Prefer a ready-made kernel view if this rocprofv3 ships one.
-    for t in tables:
+    # Prefer the ready-made `kernels` view (rocprofv3 ships one that joins the
+    # dispatch rows to the symbol names); fall back to anything kernel-shaped.
+    cands = sorted((t for t in tables if "kernel" in t.lower()),
+                   key=lambda t: (t.lower() != "kernels", len(t)))
+    for t in cands:
         cols = [c[1].lower() for c in cur.execute(f"PRAGMA table_info('{t}')")]
-        if "kernel" not in t.lower():
-            continue
-        name_col = next((c for c in cols if "name" in c and "display" not in c), None)
+        name_col = next((c for c in cols if c in ("name", "display_name", "kernel_name")), None)
         start = next((c for c in cols if c in ("start", "start_timestamp", "begin")), None)
         end = next((c for c in cols if c in ("end", "end_timestamp")), None)
         dur = next((c for c in cols if "duration" in c), None)
-        if name_col and (dur or (start and end)):
-            expr = dur if dur else f"({end} - {start})"
-            rows = cur.execute(
-                f"SELECT {name_col}, COUNT(*), SUM({expr}), AVG({expr}) "
-                f"FROM '{t}' GROUP BY {name_col} ORDER BY SUM({expr}) DESC"
-            ).fetchall()
-            if rows and rows[0][2]:
-                if isinstance(rows[0][0], int):  # interned name -> strings table
-                    lut = string_table(cur, tables)
-                    rows = [(lut.get(r[0], f"id:{r[0]}"),) + r[1:] for r in rows]
-                report(rows, top_n, t)
-                return
-    raise SystemExit(f"no kernel table found; tables: {tables}")
+        if not name_col or not (dur or (start and end)):
+            continue
+        sample = cur.execute(f'SELECT "{name_col}" FROM `{t}` LIMIT 1').fetchone()
+        if not sample or not isinstance(sample[0], str) or not sample[0]:
+            continue  # empty/interned names: this is not the joined view
+        expr = f'"{dur}"' if dur else f'("{end}" - "{start}")'
+        rows = cur.execute(
+            f'SELECT "{name_col}", COUNT(*), SUM({expr}), AVG({expr}) '
+            f"FROM `{t}` GROUP BY \"{name_col}\" ORDER BY SUM({expr}) DESC"
+        ).fetchall()
+        if rows and rows[0][2]:
+            report(rows, top_n, t)
+            return
+    # Last resort: join the raw dispatch rows to the symbol table ourselves.
+    if "rocpd_kernel_dispatch" in tables and "rocpd_info_kernel_symbol" in tables:
+        rows = cur.execute(
+            'SELECT S.display_name, COUNT(*), SUM(K."end" - K."start"), '
+            'AVG(K."end" - K."start") FROM rocpd_kernel_dispatch K '
+            "JOIN rocpd_info_kernel_symbol S ON S.id = K.kernel_id "
+            'GROUP BY S.display_name ORDER BY SUM(K."end" - K."start") DESC'
+        ).fetchall()
+        if rows and rows[0][2]:
+            report(rows, top_n, "dispatch x symbol join")
+            return
+    raise SystemExit(f"no kernel table with names found; candidates: {cands}")
 
 
 def string_table(cur, tables):
