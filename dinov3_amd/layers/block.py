@@ -22,6 +22,38 @@ from .ffn_layers import Mlp
 from .norms import LayerScale
 
 
+class DropPathPlan:
+    """Batched stochastic-depth bookkeeping for a whole forward pass.
+
+    The per-sublayer `_subset_rows` costs a randperm (rand + radix sort) plus
+    four index kernels per crop group — ~1000 tiny launches per ViT-L step.
+    This plan draws EVERY sublayer's row subset up front: one rand + one
+    batched argsort + one fused index build per crop group per step; each
+    sublayer then just slices a row of the precomputed table. Stateless by
+    sublayer index, so activation-checkpoint recompute replays identically.
+    """
+
+    def __init__(self, metas: List[GroupMeta], keep_ratio: float, n_sublayers: int, device):
+        rows_parts = []
+        scales = []
+        new_metas: List[GroupMeta] = []
+        new_off = 0
+        for (off, B, N, sin, cos, prefix) in metas:
+            keep = max(int(B * keep_ratio), 1)
+            perm = torch.argsort(torch.rand(n_sublayers, B, device=device), dim=1)[:, :keep]
+            r = off + perm.unsqueeze(2) * N + torch.arange(N, device=device)
+            rows_parts.append(r.reshape(n_sublayers, keep * N))
+            new_metas.append((new_off, keep, N, sin, cos, prefix))
+            scales.append(torch.full((keep * N,), B / keep, device=device))
+            new_off += keep * N
+        self.rows = torch.cat(rows_parts, dim=1)  # [n_sublayers, total_keep_rows]
+        self.scale = torch.cat(scales)
+        self.metas = new_metas
+
+    def take(self, sublayer: int) -> Tuple[torch.Tensor, List[GroupMeta], torch.Tensor]:
+        return self.rows[sublayer], self.metas, self.scale
+
+
 def _subset_rows(metas: List[GroupMeta], keep_ratio: float, device) -> Tuple[torch.Tensor, List[GroupMeta], torch.Tensor]:
     """Random per-group sample subsets -> (row indices, new metas, per-row scale)."""
     rows = []
@@ -81,7 +113,8 @@ class SelfAttentionBlock(nn.Module):
         return flat + res
 
     # ------------------------------------------------------------------
-    def forward_flat(self, flat: torch.Tensor, metas: List[GroupMeta]) -> torch.Tensor:
+    def forward_flat(self, flat: torch.Tensor, metas: List[GroupMeta],
+                     plan: Optional[DropPathPlan] = None, block_idx: int = 0) -> torch.Tensor:
         if not (self.training and self.sample_drop_ratio > 0.0):
             flat = self._add_scaled(flat, self.attn.forward_flat(self.norm1(flat), metas), self.ls1)
             flat = self._add_scaled(flat, self.mlp(self.norm2(flat)), self.ls2)
@@ -90,12 +123,18 @@ class SelfAttentionBlock(nn.Module):
         from ..ops.row_ops import gather_rows, scatter_add_rows
 
         keep_ratio = 1.0 - self.sample_drop_ratio
-        rows1, metas1, scale1 = _subset_rows(metas, keep_ratio, flat.device)
+        if plan is not None:
+            rows1, metas1, scale1 = plan.take(2 * block_idx)
+        else:
+            rows1, metas1, scale1 = _subset_rows(metas, keep_ratio, flat.device)
         sub = gather_rows(flat, rows1)
         res = self.ls1(self.attn.forward_flat(self.norm1(sub), metas1))
         flat = scatter_add_rows(flat, rows1, res, scale1)
 
-        rows2, _, scale2 = _subset_rows(metas, keep_ratio, flat.device)
+        if plan is not None:
+            rows2, _, scale2 = plan.take(2 * block_idx + 1)
+        else:
+            rows2, _, scale2 = _subset_rows(metas, keep_ratio, flat.device)
         sub = gather_rows(flat, rows2)
         res = self.ls2(self.mlp(self.norm2(sub)))
         return scatter_add_rows(flat, rows2, res, scale2)

@@ -188,14 +188,22 @@ class DinoVisionTransformer(nn.Module):
             metas.append(SelfAttention._meta_for(t, rope_tables[key], off))
             off += t.shape[0] * t.shape[1]
 
+        plan = None
+        if self.training and self.blocks[0].sample_drop_ratio > 0.0:
+            from ..layers.block import DropPathPlan
+
+            keep_ratio = 1.0 - self.blocks[0].sample_drop_ratio
+            plan = DropPathPlan(metas, keep_ratio, 2 * len(self.blocks), flat.device)
+
         if self.grad_checkpointing and self.training and torch.is_grad_enabled():
             from torch.utils.checkpoint import checkpoint
 
-            for block in self.blocks:
-                flat = checkpoint(block.forward_flat, flat, metas, use_reentrant=False)
+            for i, block in enumerate(self.blocks):
+                flat = checkpoint(block.forward_flat, flat, metas, plan, i,
+                                  use_reentrant=False)
         else:
-            for block in self.blocks:
-                flat = block.forward_flat(flat, metas)
+            for i, block in enumerate(self.blocks):
+                flat = block.forward_flat(flat, metas, plan, i)
         tokens = uncat_with_shapes(flat, shapes, counts)
 
         output = []

@@ -131,3 +131,43 @@ def test_get_intermediate_layers():
     assert len(outs) == 2
     patches, cls = outs[0]
     assert patches.shape == (1, 4, 384) and cls.shape == (1, 384)
+
+
+def test_droppath_plan_mechanics():
+    from dinov3_amd.layers.block import DropPathPlan
+
+    torch.manual_seed(0)
+    metas = [(0, 4, 5, None, None, 1), (20, 6, 3, None, None, 1)]
+    plan = DropPathPlan(metas, keep_ratio=0.5, n_sublayers=6, device="cpu")
+    assert plan.rows.shape[0] == 6
+    for s in range(6):
+        rows, new_metas, scale = plan.take(s)
+        # group 0: keep 2 of 4 samples -> 10 rows in [0, 20); group 1: 3 of 6 -> 9 rows in [20, 38)
+        assert rows.shape[0] == 2 * 5 + 3 * 3
+        g0, g1 = rows[:10], rows[10:]
+        assert g0.min() >= 0 and g0.max() < 20 and g1.min() >= 20 and g1.max() < 38
+        assert len(set(rows.tolist())) == rows.shape[0]  # no duplicate rows
+        assert torch.allclose(scale[:10], torch.full((10,), 2.0))
+        assert torch.allclose(scale[10:], torch.full((9,), 2.0))
+    # different sublayers draw different subsets (overwhelmingly likely)
+    assert not torch.equal(plan.rows[0], plan.rows[1])
+
+
+def test_droppath_plan_keepall_matches_dense():
+    """keep == B makes drop-path a no-op: the planned subset path must equal
+    the plain dense forward bit-for-bit (scatter-add is order-independent)."""
+    from dinov3_amd.layers.attention import SelfAttention
+    from dinov3_amd.layers.block import DropPathPlan
+    from dinov3_amd.utils.utils import cat_keep_shapes
+
+    torch.manual_seed(0)
+    blk = SelfAttentionBlock(dim=32, num_heads=2, drop_path=0.5, init_values=1e-5)
+    blk.train()
+    x = torch.randn(6, 5, 32)
+    flat, _, _ = cat_keep_shapes([x])
+    metas = [SelfAttention._meta_for(x, None, 0)]
+    plan = DropPathPlan(metas, keep_ratio=1.0, n_sublayers=2, device="cpu")
+    out_plan = blk.forward_flat(flat.clone(), metas, plan, 0)
+    blk.eval()
+    out_dense = blk.forward_flat(flat.clone(), metas)
+    assert torch.allclose(out_plan, out_dense, atol=1e-5)
