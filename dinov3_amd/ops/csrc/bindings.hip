@@ -935,7 +935,17 @@ torch::Tensor multi_tensor_l2norm_sq(std::vector<torch::Tensor> g) {
 
 }  // namespace
 
+// hipBLASLt fused-epilogue entry points (blaslt_ext.hip)
+std::vector<torch::Tensor> blaslt_gemm_bias_gelu_fwd(torch::Tensor x, torch::Tensor w,
+                                                     torch::Tensor bias);
+std::vector<torch::Tensor> blaslt_gemm_dgelu_bgrad(torch::Tensor dy, torch::Tensor w2,
+                                                   torch::Tensor pre);
+torch::Tensor blaslt_gemm_bias(torch::Tensor x, torch::Tensor w, torch::Tensor bias);
+
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, mod) {
+  mod.def("blaslt_gemm_bias_gelu_fwd", &blaslt_gemm_bias_gelu_fwd);
+  mod.def("blaslt_gemm_dgelu_bgrad", &blaslt_gemm_dgelu_bgrad);
+  mod.def("blaslt_gemm_bias", &blaslt_gemm_bias);
   mod.def("layernorm_fwd", &layernorm_fwd);
   mod.def("layernorm_bwd", &layernorm_bwd);
   mod.def("rmsnorm_fwd", &rmsnorm_fwd);

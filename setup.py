@@ -31,6 +31,7 @@ setup(
                 "cxx": ["-O3", "-std=c++17"],
                 "nvcc": ["-O3", "-std=c++17", "--offload-arch=gfx950"],
             },
+            libraries=["hipblaslt"],
         )
     ],
     cmdclass={"build_ext": BuildExtension.with_options(use_ninja=True)},
