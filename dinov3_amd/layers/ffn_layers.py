@@ -25,15 +25,11 @@ class Mlp(nn.Module):
         self.drop = nn.Dropout(drop) if drop > 0 else nn.Identity()
 
     def forward(self, x: torch.Tensor) -> torch.Tensor:
-        from ..ops.blaslt_mlp import blaslt_mlp, blaslt_mlp_enabled
-
-        if (blaslt_mlp_enabled(x) and self.fc1.bias is not None
-                and not isinstance(self.drop, nn.Dropout)):
-            # DINOV3_BLASLT_MLP=1: bias+GELU / dGELU+bias-grad fused into the
-            # fc1-fwd and fc2-dgrad GEMM epilogues.
-            return blaslt_mlp(x, self.fc1.weight, self.fc1.bias,
-                              self.fc2.weight, self.fc2.bias)
-        # default: fc1 GEMM without bias epilogue; bias fused into the GELU kernel.
+        # fc1 GEMM without bias epilogue; bias fused into the GELU kernel.
+        # (hipBLASLt epilogue fusion was probed on gfx950 and rejected: only
+        # BIAS/GELU_BIAS exist — no GELU_AUX / DGELU_BGRAD — so a GEMM-fused
+        # forward would leave backward without the pre-activation. See
+        # ops/csrc/blaslt_ext.hip and docs/KERNELS.md.)
         h = F.linear(x, self.fc1.weight)
         h = bias_gelu(h, self.fc1.bias)
         h = self.drop(h)

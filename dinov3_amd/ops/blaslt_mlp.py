@@ -1,14 +1,15 @@
-"""MLP forward/backward on hipBLASLt fused epilogues (round-2 path, gated).
+"""MLP forward/backward on hipBLASLt fused epilogues — PROBED AND REJECTED.
 
-Replaces the standalone bias+GELU kernels with GEMM epilogues:
-  fc1 fwd:  GELU_AUX_BIAS  — one GEMM computes h = gelu(x@W1^T + b1) and
-            saves the pre-activation as aux (no bias_gelu_fwd launch).
-  fc2 bwd:  DGELU_BGRAD    — the dy@W2 dgrad GEMM applies gelu'(aux) and
-            reduces db1 in its epilogue (no bias_gelu_bwd launch).
-
-Enable with DINOV3_BLASLT_MLP=1. Off by default until the epilogue semantics
-are validated on hardware (tests/test_blaslt_gpu.py runs under the same
-flag); the default path is the hand-written bias_gelu kernels.
+Kept as evidence/experiment scaffolding only. On this stack (ROCm 7.2
+hipBLASLt, gfx950) the heuristic returns ZERO algorithms for every aux /
+gradient epilogue (GELU_AUX, GELU_AUX_BIAS, DGELU, DGELU_BGRAD, BGRADA,
+BGRADB) — only BIAS and GELU_BIAS exist (probed on hardware with
+blaslt_probe_epilogue, 2026-09; 8 algorithms each for BIAS/GELU_BIAS, 0 for
+all others). Without GELU_AUX the fused forward cannot save the
+pre-activation that backward needs, so GEMM-epilogue MLP fusion is not
+viable here and the hand-written bias_gelu kernels (ops/csrc/elementwise.hip)
+remain the production path. This module raises if the entry points hit an
+unsupported epilogue.
 
 Reference behavior: dinov3_jax/layers/ffn_layers.py:24-49 (minus the §8 B4
 extra activation after fc2, deliberately not reproduced).

@@ -941,11 +941,14 @@ std::vector<torch::Tensor> blaslt_gemm_bias_gelu_fwd(torch::Tensor x, torch::Ten
 std::vector<torch::Tensor> blaslt_gemm_dgelu_bgrad(torch::Tensor dy, torch::Tensor w2,
                                                    torch::Tensor pre);
 torch::Tensor blaslt_gemm_bias(torch::Tensor x, torch::Tensor w, torch::Tensor bias);
+int64_t blaslt_probe_epilogue(int64_t epi, int64_t aux_type, int64_t bias_type,
+                              int64_t m, int64_t n, int64_t k);
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, mod) {
   mod.def("blaslt_gemm_bias_gelu_fwd", &blaslt_gemm_bias_gelu_fwd);
   mod.def("blaslt_gemm_dgelu_bgrad", &blaslt_gemm_dgelu_bgrad);
   mod.def("blaslt_gemm_bias", &blaslt_gemm_bias);
+  mod.def("blaslt_probe_epilogue", &blaslt_probe_epilogue);
   mod.def("layernorm_fwd", &layernorm_fwd);
   mod.def("layernorm_bwd", &layernorm_bwd);
   mod.def("rmsnorm_fwd", &rmsnorm_fwd);

@@ -203,6 +203,56 @@ torch::Tensor gemm_bias(torch::Tensor x, torch::Tensor w, torch::Tensor bias) {
   return y;
 }
 
+// Probe: how many heuristic algorithms exist for a given epilogue/dtype combo
+// on this hardware (no matmul is run). aux_type/bias_type: hipDataType ints,
+// -1 = leave unset. Returns n_results (0 = unsupported combo).
+int probe_epilogue(int64_t epi, int64_t aux_type, int64_t bias_type,
+                   int64_t m, int64_t n, int64_t k) {
+  hipblasLtMatmulDesc_t op;
+  if (hipblasLtMatmulDescCreate(&op, HIPBLAS_COMPUTE_32F, HIP_R_32F) !=
+      HIPBLAS_STATUS_SUCCESS)
+    return -1;
+  hipblasOperation_t ta = HIPBLAS_OP_T, tb = HIPBLAS_OP_N;
+  hipblasLtMatmulDescSetAttribute(op, HIPBLASLT_MATMUL_DESC_TRANSA, &ta, sizeof(ta));
+  hipblasLtMatmulDescSetAttribute(op, HIPBLASLT_MATMUL_DESC_TRANSB, &tb, sizeof(tb));
+  hipblasLtEpilogue_t e = (hipblasLtEpilogue_t)epi;
+  hipblasLtMatmulDescSetAttribute(op, HIPBLASLT_MATMUL_DESC_EPILOGUE, &e, sizeof(e));
+  if (bias_type >= 0) {
+    int32_t bt = (int32_t)bias_type;
+    hipblasLtMatmulDescSetAttribute(op, HIPBLASLT_MATMUL_DESC_BIAS_DATA_TYPE, &bt,
+                                    sizeof(bt));
+  }
+  if (aux_type >= 0) {
+    int32_t at = (int32_t)aux_type;
+    hipblasLtMatmulDescSetAttribute(op, HIPBLASLT_MATMUL_DESC_EPILOGUE_AUX_DATA_TYPE,
+                                    &at, sizeof(at));
+    int64_t ld = m;
+    hipblasLtMatmulDescSetAttribute(op, HIPBLASLT_MATMUL_DESC_EPILOGUE_AUX_LD, &ld,
+                                    sizeof(ld));
+  }
+  hipblasLtMatrixLayout_t a, b, c, d;
+  hipblasLtMatrixLayoutCreate(&a, HIP_R_16BF, k, m, k);
+  hipblasLtMatrixLayoutCreate(&b, HIP_R_16BF, k, n, k);
+  hipblasLtMatrixLayoutCreate(&c, HIP_R_16BF, m, n, m);
+  hipblasLtMatrixLayoutCreate(&d, HIP_R_16BF, m, n, m);
+  hipblasLtMatmulPreference_t pref;
+  hipblasLtMatmulPreferenceCreate(&pref);
+  size_t ws = kWorkspaceBytes;
+  hipblasLtMatmulPreferenceSetAttribute(pref, HIPBLASLT_MATMUL_PREF_MAX_WORKSPACE_BYTES,
+                                        &ws, sizeof(ws));
+  hipblasLtMatmulHeuristicResult_t results[8];
+  int n_results = 0;
+  hipblasStatus_t st = hipblasLtMatmulAlgoGetHeuristic(handle(), op, a, b, c, d, pref,
+                                                       8, results, &n_results);
+  hipblasLtMatmulPreferenceDestroy(pref);
+  hipblasLtMatrixLayoutDestroy(a);
+  hipblasLtMatrixLayoutDestroy(b);
+  hipblasLtMatrixLayoutDestroy(c);
+  hipblasLtMatrixLayoutDestroy(d);
+  hipblasLtMatmulDescDestroy(op);
+  return st == HIPBLAS_STATUS_SUCCESS ? n_results : -(int)st;
+}
+
 }  // namespace blaslt
 
 std::vector<torch::Tensor> blaslt_gemm_bias_gelu_fwd(torch::Tensor x, torch::Tensor w,
@@ -215,4 +265,8 @@ std::vector<torch::Tensor> blaslt_gemm_dgelu_bgrad(torch::Tensor dy, torch::Tens
 }
 torch::Tensor blaslt_gemm_bias(torch::Tensor x, torch::Tensor w, torch::Tensor bias) {
   return blaslt::gemm_bias(x, w, bias);
+}
+int64_t blaslt_probe_epilogue(int64_t epi, int64_t aux_type, int64_t bias_type,
+                              int64_t m, int64_t n, int64_t k) {
+  return blaslt::probe_epilogue(epi, aux_type, bias_type, m, n, k);
 }

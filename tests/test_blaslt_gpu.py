@@ -9,11 +9,15 @@ import os
 import pytest
 import torch
 
-pytestmark = [
-    pytest.mark.gpu,
-    pytest.mark.skipif(os.environ.get("DINOV3_BLASLT_MLP", "0") != "1",
-                       reason="set DINOV3_BLASLT_MLP=1 to test the hipBLASLt epilogue path"),
-]
+pytestmark = pytest.mark.gpu
+
+# hipBLASLt on ROCm 7.2 / gfx950 exposes only BIAS and GELU_BIAS epilogues
+# (AUX/DGELU/BGRAD probed on hardware: 0 algorithms) — the fused-MLP tests
+# stay gated until a stack ships those epilogues.
+needs_aux_epilogues = pytest.mark.skipif(
+    os.environ.get("DINOV3_BLASLT_MLP", "0") != "1",
+    reason="hipBLASLt aux/bgrad epilogues unsupported on this stack "
+           "(set DINOV3_BLASLT_MLP=1 to re-probe)")
 
 DEV = "cuda"
 
@@ -41,6 +45,22 @@ def test_gemm_bias_layout(ops):
     _close(y, ref, 0.25, "blaslt gemm+bias")
 
 
+def test_epilogue_support_matrix(ops):
+    """Pin the probed support matrix: BIAS/GELU_BIAS exist, aux/grad don't.
+    If this ever flips on a new ROCm, the fused-MLP path becomes viable."""
+    F32, BF16 = 0, 14
+    assert ops.blaslt_probe_epilogue(4, -1, BF16, 512, 384, 128) > 0    # BIAS
+    assert ops.blaslt_probe_epilogue(36, -1, BF16, 512, 384, 128) > 0   # GELU_BIAS
+    aux_like = [(160, BF16, -1), (164, BF16, BF16), (192, BF16, -1),
+                (208, BF16, F32), (512, -1, F32)]
+    supported = [e for (e, at, bt) in aux_like
+                 if ops.blaslt_probe_epilogue(e, at, bt, 512, 384, 128) > 0]
+    assert not supported, (
+        f"hipBLASLt now supports epilogues {supported} — revisit the fused "
+        "MLP path in ops/blaslt_mlp.py")
+
+
+@needs_aux_epilogues
 def test_gemm_bias_gelu_aux(ops):
     torch.manual_seed(1)
     M, K, N = 384, 128, 512
@@ -53,6 +73,7 @@ def test_gemm_bias_gelu_aux(ops):
     _close(h, torch.nn.functional.gelu(pre_ref, approximate="tanh"), 0.25, "gelu output")
 
 
+@needs_aux_epilogues
 def test_gemm_dgelu_bgrad(ops):
     torch.manual_seed(2)
     M, N1, N2 = 256, 512, 128
@@ -67,6 +88,7 @@ def test_gemm_dgelu_bgrad(ops):
     _close(db1, p.grad.sum(dim=0), p.grad.abs().sum(0).max().item() * 2e-2 + 0.5, "bgrad")
 
 
+@needs_aux_epilogues
 def test_blaslt_mlp_autograd():
     torch.manual_seed(3)
     from dinov3_amd.ops.blaslt_mlp import blaslt_mlp
