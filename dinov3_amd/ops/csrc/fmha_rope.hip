@@ -11,6 +11,7 @@
 // see that file for the fragment-layout contract (probe-verified).
 
 #include "common.h"
+#include <cstdlib>
 
 typedef __attribute__((ext_vector_type(8))) __bf16 bf16x8;
 typedef __attribute__((ext_vector_type(16))) float f32x16;
@@ -707,6 +708,246 @@ __global__ __launch_bounds__(NT) void bwd_dkv_kernel(
   }
 }
 
+// ---------------------------------------------------------------------------
+// Backward dK/dV, 64-row super-tile variant (gated: DINOV3_FMHA_DKV64=1).
+//
+// Stages TWO 32-row q-tiles per barrier pair (half the __syncthreads and
+// lse/D loads of bwd_dkv_kernel). The round-1 attempt of this NaN'd because
+// the 64 staged rows kept the 32-row column stride (QB+8=40): rows 40..63 of
+// every d-column overwrote the next column, and lse/d overlapped (d_lds was
+// offset by QB). Fixed here: QT_STRIDE = 2*QB+8 = 72 and 2*QB-sized lse/d —
+// matching the stride the fwd/dq kernels already use. Validate on hardware
+// (tests/test_fmha_rope_gpu.py::test_dkv64_variant) before enabling.
+// ---------------------------------------------------------------------------
+template <int HD, int NT = 256>
+__global__ __launch_bounds__(NT) void bwd_dkv64_kernel(
+    const __hip_bfloat16* __restrict__ qkv, const __hip_bfloat16* __restrict__ dout,
+    const float* __restrict__ sin_t, const float* __restrict__ cos_t,
+    const float* __restrict__ lse, const float* __restrict__ D,
+    __hip_bfloat16* __restrict__ dqkv, int B, int H, int N, int P, float scale) {
+  constexpr int KSLICES = HD / 16;
+  constexpr int DTILES = HD / 32;
+  constexpr int QB = 32;
+  constexpr int SQB = 2 * QB;            // staged rows per barrier pair
+  constexpr int QT_STRIDE = SQB + 8;     // 72: column stride sized for SQB rows
+  constexpr int HALF = HD / 2;
+
+  const int bh = blockIdx.x;
+  const int b = bh / H;
+  const int h = bh % H;
+  const int wave = threadIdx.x / 64;
+  const int lane = threadIdx.x & 63;
+  const int hhalf = lane >> 5;
+  const int l31 = lane & 31;
+  const int k0 = blockIdx.y * 128 + wave * 32;
+  const int prefix = N - P;
+  const bool use_rope = (sin_t != nullptr);
+
+  QkvView qv;
+  qv.base = qkv + ((long)b * N * 3 * H + h) * HD;
+  qv.row_stride = (long)3 * H * HD;
+  qv.which_stride = (long)H * HD;
+  const __hip_bfloat16* do_base = dout + ((long)b * N * H + h) * HD;
+  const long do_stride = (long)H * HD;
+
+  extern __shared__ __attribute__((aligned(16))) char smem_raw[];
+  __hip_bfloat16* qt_lds = reinterpret_cast<__hip_bfloat16*>(smem_raw);  // rope'd Q^T [HD][72]
+  __hip_bfloat16* dot_lds = qt_lds + HD * QT_STRIDE;                     // dO^T [HD][72]
+  float* lse_lds = reinterpret_cast<float*>(dot_lds + HD * QT_STRIDE);   // [SQB]
+  float* d_lds = lse_lds + SQB;                                          // [SQB]
+
+  const int krow = k0 + l31;
+  const int safe = krow < N ? krow : (N - 1);
+  bf16x8 kf[KSLICES], vf[KSLICES];
+  {
+#pragma unroll
+    for (int s = 0; s < KSLICES; ++s) {
+      kf[s] = load8(qv.at(safe, 1, s * 16 + hhalf * 8));
+      vf[s] = load8(qv.at(safe, 2, s * 16 + hhalf * 8));
+    }
+    const int p = safe - prefix;
+    if (use_rope && p >= 0) {
+      const float* srow = sin_t + (long)p * HD;
+      const float* crow = cos_t + (long)p * HD;
+#pragma unroll
+      for (int s2 = 0; s2 < KSLICES / 2; ++s2)
+        rope_rotate8(kf[s2], kf[s2 + KSLICES / 2], srow, crow, s2 * 16 + hhalf * 8);
+    }
+  }
+
+  float dk_acc[DTILES][16], dv_acc[DTILES][16];
+#pragma unroll
+  for (int t = 0; t < DTILES; ++t)
+#pragma unroll
+    for (int r = 0; r < 16; ++r) {
+      dk_acc[t][r] = 0.f;
+      dv_acc[t][r] = 0.f;
+    }
+
+  constexpr int PAIRS_PER_ROW = HALF / 8;
+  constexpr int PER_ROW = HD / 8;
+  constexpr int QK_ITEMS = SQB * PAIRS_PER_ROW;  // up to 512 (HD=128)
+  constexpr int DO_ITEMS = SQB * PER_ROW;        // up to 1024 (HD=128)
+  constexpr int QK_IPT = (QK_ITEMS + NT - 1) / NT;
+  constexpr int DO_IPT = (DO_ITEMS + NT - 1) / NT;
+  bf16x8 qlo[QK_IPT], qhi[QK_IPT], doreg[DO_IPT];
+  float lse_reg = INFINITY, d_reg = 0.f;
+  auto issue_loads = [&](int qbase0) {
+#pragma unroll
+    for (int j = 0; j < QK_IPT; ++j) {
+      const int idx = threadIdx.x + j * NT;
+      const int qrow = qbase0 + idx / PAIRS_PER_ROW;
+      const int c0 = (idx % PAIRS_PER_ROW) * 8;
+      const bool ok = idx < QK_ITEMS && qrow < N;
+      qlo[j] = ok ? load8(qv.at(qrow, 0, c0)) : bf16x8{};
+      qhi[j] = ok ? load8(qv.at(qrow, 0, c0 + HALF)) : bf16x8{};
+    }
+#pragma unroll
+    for (int j = 0; j < DO_IPT; ++j) {
+      const int idx = threadIdx.x + j * NT;
+      const int qrow = qbase0 + idx / PER_ROW;
+      doreg[j] = (idx < DO_ITEMS && qrow < N)
+                     ? load8(do_base + (long)qrow * do_stride + (idx % PER_ROW) * 8)
+                     : bf16x8{};
+    }
+    if (threadIdx.x < SQB) {
+      const int qrow = qbase0 + threadIdx.x;
+      lse_reg = (qrow < N) ? lse[((long)b * H + h) * N + qrow] : INFINITY;
+      d_reg = (qrow < N) ? D[((long)b * H + h) * N + qrow] : 0.f;
+    }
+  };
+  auto write_tile = [&](int qbase0) {
+#pragma unroll
+    for (int j = 0; j < QK_IPT; ++j) {
+      const int idx = threadIdx.x + j * NT;
+      if (idx < QK_ITEMS) {
+        const int lrow = idx / PAIRS_PER_ROW;
+        const int c0 = (idx % PAIRS_PER_ROW) * 8;
+        const int qrow = qbase0 + lrow;
+        const int p = qrow - prefix;
+        if (use_rope && qrow < N && p >= 0)
+          rope_rotate8(qlo[j], qhi[j], sin_t + (long)p * HD, cos_t + (long)p * HD, c0);
+#pragma unroll
+        for (int e = 0; e < 8; ++e) {
+          qt_lds[(c0 + e) * QT_STRIDE + lrow] = reinterpret_cast<__hip_bfloat16*>(&qlo[j])[e];
+          qt_lds[(c0 + HALF + e) * QT_STRIDE + lrow] = reinterpret_cast<__hip_bfloat16*>(&qhi[j])[e];
+        }
+      }
+    }
+#pragma unroll
+    for (int j = 0; j < DO_IPT; ++j) {
+      const int idx = threadIdx.x + j * NT;
+      if (idx < DO_ITEMS) {
+        const int lrow = idx / PER_ROW;
+        const int c8 = (idx % PER_ROW) * 8;
+#pragma unroll
+        for (int e = 0; e < 8; ++e)
+          dot_lds[(c8 + e) * QT_STRIDE + lrow] = reinterpret_cast<__hip_bfloat16*>(&doreg[j])[e];
+      }
+    }
+    if (threadIdx.x < SQB) {
+      lse_lds[threadIdx.x] = lse_reg;
+      d_lds[threadIdx.x] = d_reg;
+    }
+  };
+
+  const int n_super = (N + SQB - 1) / SQB;
+  issue_loads(0);
+  for (int qt = 0; qt < n_super; ++qt) {
+    const int qbase0 = qt * SQB;
+    __syncthreads();
+    write_tile(qbase0);
+    if (qt + 1 < n_super) issue_loads(qbase0 + SQB);
+    __syncthreads();
+
+    for (int sub = 0; sub < 2 && qbase0 + sub * QB < N; ++sub) {
+      const int qbase = qbase0 + sub * QB;
+      const int qrow_off = sub * QB;
+      f32x16 s_acc = {}, dp_acc = {};
+#pragma unroll
+      for (int s = 0; s < KSLICES; ++s) {
+        union { unsigned u[4]; bf16x8 v8; } aq, ad;
+#pragma unroll
+        for (int e = 0; e < 8; ++e) {
+          const int d = s * 16 + hhalf * 8 + e;
+          reinterpret_cast<__hip_bfloat16*>(&aq)[e] = qt_lds[d * QT_STRIDE + qrow_off + l31];
+          reinterpret_cast<__hip_bfloat16*>(&ad)[e] = dot_lds[d * QT_STRIDE + qrow_off + l31];
+        }
+        s_acc = MFMA32(aq.v8, kf[s], s_acc);
+        dp_acc = MFMA32(ad.v8, vf[s], dp_acc);
+      }
+      float pv[16], ds[16];
+#pragma unroll
+      for (int r = 0; r < 16; ++r) {
+        const int qrow = qbase + c_row(r, hhalf);
+        const bool valid = (qrow < N) && (krow < N);
+        const float l = lse_lds[qrow_off + c_row(r, hhalf)];
+        float p = valid ? __expf(s_acc[r] * scale - l) : 0.f;
+        pv[r] = p;
+        ds[r] = p * (dp_acc[r] - d_lds[qrow_off + c_row(r, hhalf)]) * scale;
+      }
+      bf16x8 p0 = pack_fragment(pv, 0);
+      bf16x8 p1 = pack_fragment(pv, 8);
+      bf16x8 s0 = pack_fragment(ds, 0);
+      bf16x8 s1 = pack_fragment(ds, 8);
+#pragma unroll
+      for (int t = 0; t < DTILES; ++t) {
+        union { unsigned u[4]; bf16x8 v8; } ado0, ado1, aq0, aq1;
+#pragma unroll
+        for (int e = 0; e < 8; ++e) {
+          reinterpret_cast<__hip_bfloat16*>(&ado0)[e] = dot_lds[(t * 32 + l31) * QT_STRIDE + qrow_off + hhalf * 8 + e];
+          reinterpret_cast<__hip_bfloat16*>(&ado1)[e] = dot_lds[(t * 32 + l31) * QT_STRIDE + qrow_off + 16 + hhalf * 8 + e];
+          reinterpret_cast<__hip_bfloat16*>(&aq0)[e] = qt_lds[(t * 32 + l31) * QT_STRIDE + qrow_off + hhalf * 8 + e];
+          reinterpret_cast<__hip_bfloat16*>(&aq1)[e] = qt_lds[(t * 32 + l31) * QT_STRIDE + qrow_off + 16 + hhalf * 8 + e];
+        }
+        f32x16 accv, acck;
+#pragma unroll
+        for (int r = 0; r < 16; ++r) {
+          accv[r] = dv_acc[t][r];
+          acck[r] = dk_acc[t][r];
+        }
+        accv = MFMA32(ado0.v8, p0, accv);
+        accv = MFMA32(ado1.v8, p1, accv);
+        acck = MFMA32(aq0.v8, s0, acck);
+        acck = MFMA32(aq1.v8, s1, acck);
+#pragma unroll
+        for (int r = 0; r < 16; ++r) {
+          dv_acc[t][r] = accv[r];
+          dk_acc[t][r] = acck[r];
+        }
+      }
+    }
+  }
+
+  if (krow < N) {
+    const int p = krow - prefix;
+    if (use_rope && p >= 0) {
+      const float* srow = sin_t + (long)p * HD;
+      const float* crow = cos_t + (long)p * HD;
+#pragma unroll
+      for (int t = 0; t < DTILES / 2; ++t)
+#pragma unroll
+        for (int r = 0; r < 16; ++r) {
+          const int d = t * 32 + c_row(r, hhalf);
+          const float c = crow[d], s = srow[d];
+          const float glo = dk_acc[t][r], ghi = dk_acc[t + DTILES / 2][r];
+          dk_acc[t][r] = glo * c + ghi * s;
+          dk_acc[t + DTILES / 2][r] = ghi * c - glo * s;
+        }
+    }
+    __hip_bfloat16* dkp = dqkv + ((((long)b * N + krow) * 3 + 1) * H + h) * HD;
+    __hip_bfloat16* dvp = dqkv + ((((long)b * N + krow) * 3 + 2) * H + h) * HD;
+#pragma unroll
+    for (int t = 0; t < DTILES; ++t)
+#pragma unroll
+      for (int r = 0; r < 16; ++r) {
+        const int d = t * 32 + c_row(r, hhalf);
+        *reinterpret_cast<short*>(dkp + d) = f32_to_bf16(dk_acc[t][r]);
+        *reinterpret_cast<short*>(dvp + d) = f32_to_bf16(dv_acc[t][r]);
+      }
+  }
+}
+
 // preprocess on token-major dO/O: D[b,h,n] = sum_d dO*O
 __global__ void bwd_pre_tm_kernel(const __hip_bfloat16* __restrict__ dout,
                                   const __hip_bfloat16* __restrict__ o,
@@ -814,12 +1055,21 @@ void launch_fmha_rope_bwd_dkv(const __hip_bfloat16* qkv, const __hip_bfloat16* d
                               const float* D, __hip_bfloat16* dqkv, int B, int H, int N,
                               int P, int HD, float scale, hipStream_t stream) {
   const bool small = N <= 64;
+  // Gated 64-row super-tile variant (see bwd_dkv64_kernel); only useful when
+  // there is more than one 32-row q-tile. getenv per launch is ~ns vs the
+  // 100 us kernel and keeps the flag flippable from tests.
+  const char* dkv64 = getenv("DINOV3_FMHA_DKV64");
+  const bool use64 = !small && dkv64 && dkv64[0] == '1';
   dim3 grid(B * H, small ? 1 : (N + 127) / 128);
   if (HD == 64) {
     size_t shmem = 2 * 64 * 72 * sizeof(__hip_bfloat16) + 128 * sizeof(float);
     if (small)
       hipLaunchKernelGGL(HIP_KERNEL_NAME(fmha_rope::bwd_dkv_kernel<64, 128>), grid,
                          dim3(128), shmem, stream, qkv, dout, sin_t, cos_t, lse, D, dqkv,
+                         B, H, N, P, scale);
+    else if (use64)
+      hipLaunchKernelGGL(HIP_KERNEL_NAME(fmha_rope::bwd_dkv64_kernel<64, 256>), grid,
+                         dim3(256), shmem, stream, qkv, dout, sin_t, cos_t, lse, D, dqkv,
                          B, H, N, P, scale);
     else
       hipLaunchKernelGGL(HIP_KERNEL_NAME(fmha_rope::bwd_dkv_kernel<64, 256>), grid,
@@ -830,6 +1080,10 @@ void launch_fmha_rope_bwd_dkv(const __hip_bfloat16* qkv, const __hip_bfloat16* d
     if (small)
       hipLaunchKernelGGL(HIP_KERNEL_NAME(fmha_rope::bwd_dkv_kernel<128, 128>), grid,
                          dim3(128), shmem, stream, qkv, dout, sin_t, cos_t, lse, D, dqkv,
+                         B, H, N, P, scale);
+    else if (use64)
+      hipLaunchKernelGGL(HIP_KERNEL_NAME(fmha_rope::bwd_dkv64_kernel<128, 256>), grid,
+                         dim3(256), shmem, stream, qkv, dout, sin_t, cos_t, lse, D, dqkv,
                          B, H, N, P, scale);
     else
       hipLaunchKernelGGL(HIP_KERNEL_NAME(fmha_rope::bwd_dkv_kernel<128, 256>), grid,

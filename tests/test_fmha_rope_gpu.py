@@ -114,3 +114,32 @@ def test_flat_multi_group():
         got = out[off: off + B * N].view(B, N, H, hd).float()
         err = (got - ref).abs().max().item()
         assert err < 0.03, f"group err {err}"
+
+
+@pytest.mark.parametrize("N,hd,prefix", [(197, 64, 1), (197, 128, 1), (300, 64, 2), (128, 64, 0)])
+def test_dkv64_variant(N, hd, prefix, monkeypatch):
+    """Gated 64-row super-tile dkv kernel vs the fp32 reference (the round-1
+    attempt NaN'd from a 32-row LDS stride; this variant fixes the staging
+    layout — flip DINOV3_FMHA_DKV64 on by default once this passes)."""
+    monkeypatch.setenv("DINOV3_FMHA_DKV64", "1")
+    from dinov3_amd.ops.flat_attention import flat_multi_fmha
+
+    torch.manual_seed(4)
+    B, H = 2, 3
+    D = H * hd
+    qkv_flat = torch.randn(B * N, 3 * D, device=DEV).bfloat16().requires_grad_(True)
+    P = N - prefix
+    angles = torch.rand(P, hd // 2, device=DEV) * 6.28
+    angles = torch.cat([angles, angles], dim=-1)
+    sin, cos = angles.sin().contiguous(), angles.cos().contiguous()
+    out = flat_multi_fmha(qkv_flat, H, [(0, B, N, sin, cos, prefix)])
+    dy = torch.randn_like(out)
+    out.backward(dy)
+
+    ref_in = qkv_flat.detach().float().requires_grad_(True)
+    o_ref = _ref_group(ref_in.view(B, N, 3, H, hd), sin, cos, prefix).reshape(B * N, D)
+    o_ref.backward(dy.float().reshape(B * N, D))
+    err = (qkv_flat.grad.float() - ref_in.grad).abs().max().item()
+    scale = ref_in.grad.abs().max().item()
+    assert torch.isfinite(qkv_flat.grad).all(), "dkv64: non-finite grads"
+    assert err < 0.02 + 0.02 * scale, f"dkv64 bwd err {err} (scale {scale})"
