@@ -60,10 +60,18 @@ class SelfAttention(nn.Module):
         self.proj = nn.Linear(dim, dim, bias=proj_bias)
         self.proj_drop = nn.Dropout(proj_drop) if proj_drop > 0 else nn.Identity()
 
-    def forward_flat(self, flat: torch.Tensor, metas: List[GroupMeta]) -> torch.Tensor:
-        """flat: [R, D] concatenated tokens of all crop groups."""
+    def forward_flat(self, flat: torch.Tensor, metas: List[GroupMeta],
+                     skip_proj_bias: bool = False) -> torch.Tensor:
+        """flat: [R, D] concatenated tokens of all crop groups.
+
+        skip_proj_bias: leave the out-projection bias off the GEMM — the
+        caller folds it into the residual kernel (ops/fused_residual.py)."""
         qkv_flat = self.qkv(flat)
         ctx_flat = flat_multi_fmha(qkv_flat, self.num_heads, metas)
+        if skip_proj_bias:
+            import torch.nn.functional as F
+
+            return self.proj_drop(F.linear(ctx_flat, self.proj.weight))
         return self.proj_drop(self.proj(ctx_flat))
 
     @staticmethod

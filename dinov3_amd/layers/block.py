@@ -112,14 +112,43 @@ class SelfAttentionBlock(nn.Module):
             return ls_axpy(flat, res, ls.gamma)
         return flat + res
 
+    def _fusable(self) -> Tuple[bool, bool]:
+        """Which sublayers can fold (gamma, GEMM-bias) into the residual
+        kernel (DINOV3_FUSED_RESIDUAL=1, ops/fused_residual.py)."""
+        from ..ops.fused_residual import fused_residual_enabled
+
+        if not fused_residual_enabled():
+            return False, False
+        fa = (isinstance(self.ls1, LayerScale) and self.attn.proj.bias is not None
+              and not isinstance(self.attn.proj_drop, nn.Dropout))
+        fm = (isinstance(self.ls2, LayerScale) and isinstance(self.mlp, Mlp)
+              and self.mlp.fc2.bias is not None
+              and not isinstance(self.mlp.drop, nn.Dropout))
+        return fa, fm
+
     # ------------------------------------------------------------------
     def forward_flat(self, flat: torch.Tensor, metas: List[GroupMeta],
-                     plan: Optional[DropPathPlan] = None, block_idx: int = 0) -> torch.Tensor:
+                     plan: Optional[DropPathPlan] = None, block_idx: int = 0,
+                     inplace_ok: bool = True) -> torch.Tensor:
+        fa, fm = self._fusable()
         if not (self.training and self.sample_drop_ratio > 0.0):
-            flat = self._add_scaled(flat, self.attn.forward_flat(self.norm1(flat), metas), self.ls1)
-            flat = self._add_scaled(flat, self.mlp(self.norm2(flat)), self.ls2)
+            if fa:
+                from ..ops.fused_residual import ls_axpy_bias
+
+                res = self.attn.forward_flat(self.norm1(flat), metas, skip_proj_bias=True)
+                flat = ls_axpy_bias(flat, res, self.ls1.gamma, self.attn.proj.bias)
+            else:
+                flat = self._add_scaled(flat, self.attn.forward_flat(self.norm1(flat), metas), self.ls1)
+            if fm:
+                from ..ops.fused_residual import ls_axpy_bias
+
+                res = self.mlp(self.norm2(flat), skip_out_bias=True)
+                flat = ls_axpy_bias(flat, res, self.ls2.gamma, self.mlp.fc2.bias)
+            else:
+                flat = self._add_scaled(flat, self.mlp(self.norm2(flat)), self.ls2)
             return flat
 
+        from ..ops.fused_residual import ls_scatter_add_rows
         from ..ops.row_ops import gather_rows, scatter_add_rows
 
         keep_ratio = 1.0 - self.sample_drop_ratio
@@ -128,14 +157,25 @@ class SelfAttentionBlock(nn.Module):
         else:
             rows1, metas1, scale1 = _subset_rows(metas, keep_ratio, flat.device)
         sub = gather_rows(flat, rows1)
-        res = self.ls1(self.attn.forward_flat(self.norm1(sub), metas1))
-        flat = scatter_add_rows(flat, rows1, res, scale1)
+        if fa:
+            if not inplace_ok:
+                flat = flat.clone()
+            res = self.attn.forward_flat(self.norm1(sub), metas1, skip_proj_bias=True)
+            flat = ls_scatter_add_rows(flat, rows1, res, self.ls1.gamma,
+                                       self.attn.proj.bias, scale1)
+        else:
+            res = self.ls1(self.attn.forward_flat(self.norm1(sub), metas1))
+            flat = scatter_add_rows(flat, rows1, res, scale1)
 
         if plan is not None:
             rows2, _, scale2 = plan.take(2 * block_idx + 1)
         else:
             rows2, _, scale2 = _subset_rows(metas, keep_ratio, flat.device)
         sub = gather_rows(flat, rows2)
+        if fm:
+            res = self.mlp(self.norm2(sub), skip_out_bias=True)
+            return ls_scatter_add_rows(flat, rows2, res, self.ls2.gamma,
+                                       self.mlp.fc2.bias, scale2)
         res = self.ls2(self.mlp(self.norm2(sub)))
         return scatter_add_rows(flat, rows2, res, scale2)
 

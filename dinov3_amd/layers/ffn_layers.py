@@ -24,15 +24,18 @@ class Mlp(nn.Module):
         self.fc2 = nn.Linear(hidden_features, out_features, bias=use_bias)
         self.drop = nn.Dropout(drop) if drop > 0 else nn.Identity()
 
-    def forward(self, x: torch.Tensor) -> torch.Tensor:
+    def forward(self, x: torch.Tensor, skip_out_bias: bool = False) -> torch.Tensor:
         # fc1 GEMM without bias epilogue; bias fused into the GELU kernel.
         # (hipBLASLt epilogue fusion was probed on gfx950 and rejected: only
         # BIAS/GELU_BIAS exist — no GELU_AUX / DGELU_BGRAD — so a GEMM-fused
         # forward would leave backward without the pre-activation. See
         # ops/csrc/blaslt_ext.hip and docs/KERNELS.md.)
+        # skip_out_bias: caller folds fc2's bias into the residual kernel.
         h = F.linear(x, self.fc1.weight)
         h = bias_gelu(h, self.fc1.bias)
         h = self.drop(h)
+        if skip_out_bias:
+            return self.drop(F.linear(h, self.fc2.weight))
         return self.drop(self.fc2(h))
 
 

@@ -199,7 +199,9 @@ class DinoVisionTransformer(nn.Module):
             from torch.utils.checkpoint import checkpoint
 
             for i, block in enumerate(self.blocks):
-                flat = checkpoint(block.forward_flat, flat, metas, plan, i,
+                # inplace_ok=False: the checkpoint saves `flat` for recompute,
+                # so the fused residual scatter must not mutate it in place.
+                flat = checkpoint(block.forward_flat, flat, metas, plan, i, False,
                                   use_reentrant=False)
         else:
             for i, block in enumerate(self.blocks):

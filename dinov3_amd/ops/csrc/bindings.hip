@@ -36,6 +36,18 @@ void launch_ls_axpy_fwd(const T*, const T*, const T*, T*, long, int, hipStream_t
 template <typename T>
 void launch_ls_axpy_bwd(const T*, const T*, const T*, T*, float*, long, int, hipStream_t);
 template <typename T>
+void launch_ls_axpy_bias_fwd(const T*, const T*, const T*, const T*, T*, long, int,
+                             hipStream_t);
+template <typename T>
+void launch_ls_axpy_bias_bwd(const T*, const T*, const T*, const T*, T*, float*, float*,
+                             long, int, hipStream_t);
+template <typename T>
+void launch_ls_scatter_add(T*, const long*, const T*, const T*, const T*, const float*,
+                           long, int, hipStream_t);
+template <typename T>
+void launch_ls_scatter_bwd(const T*, const long*, const T*, const T*, const T*,
+                           const float*, T*, float*, float*, long, int, hipStream_t);
+template <typename T>
 void launch_row_gather(const T*, const long*, T*, long, int, hipStream_t);
 template <typename T>
 void launch_row_scatter_add(T*, const long*, const T*, const float*, long, int,
@@ -291,6 +303,23 @@ torch::Tensor ls_axpy_fwd(torch::Tensor x, torch::Tensor res, torch::Tensor gamm
   return out;
 }
 
+torch::Tensor ls_axpy_bias_fwd(torch::Tensor x, torch::Tensor res, torch::Tensor gamma,
+                               torch::Tensor bias) {
+  CHECK_INPUT(x);
+  CHECK_INPUT(res);
+  const int D = x.size(-1);
+  TORCH_CHECK(D % 8 == 0, "ls_axpy_bias: D must be a multiple of 8");
+  const long rows = x.numel() / D;
+  auto out = torch::empty_like(x);
+  DISPATCH_FLOAT_BF16(x.scalar_type(), "ls_axpy_bias_fwd", [&] {
+    launch_ls_axpy_bias_fwd<scalar_t>(
+        (const scalar_t*)x.data_ptr(), (const scalar_t*)res.data_ptr(),
+        (const scalar_t*)gamma.data_ptr(), (const scalar_t*)bias.data_ptr(),
+        (scalar_t*)out.data_ptr(), rows, D, current_stream());
+  });
+  return out;
+}
+
 std::vector<torch::Tensor> ls_axpy_bwd(torch::Tensor dout, torch::Tensor res,
                                        torch::Tensor gamma) {
   CHECK_INPUT(dout);
@@ -306,6 +335,25 @@ std::vector<torch::Tensor> ls_axpy_bwd(torch::Tensor dout, torch::Tensor res,
                                  rows, D, current_stream());
   });
   return {dres, dgamma.to(dout.scalar_type())};
+}
+
+std::vector<torch::Tensor> ls_axpy_bias_bwd(torch::Tensor dout, torch::Tensor res,
+                                            torch::Tensor gamma, torch::Tensor bias) {
+  CHECK_INPUT(dout);
+  const int D = dout.size(-1);
+  const long rows = dout.numel() / D;
+  auto dres = torch::empty_like(dout);
+  auto fopt = dout.options().dtype(torch::kFloat);
+  auto dgamma = torch::zeros({D}, fopt);
+  auto dbias = torch::zeros({D}, fopt);
+  DISPATCH_FLOAT_BF16(dout.scalar_type(), "ls_axpy_bias_bwd", [&] {
+    launch_ls_axpy_bias_bwd<scalar_t>(
+        (const scalar_t*)dout.data_ptr(), (const scalar_t*)res.data_ptr(),
+        (const scalar_t*)gamma.data_ptr(), (const scalar_t*)bias.data_ptr(),
+        (scalar_t*)dres.data_ptr(), dgamma.data_ptr<float>(), dbias.data_ptr<float>(),
+        rows, D, current_stream());
+  });
+  return {dres, dgamma.to(dout.scalar_type()), dbias.to(dout.scalar_type())};
 }
 
 torch::Tensor row_gather(torch::Tensor src, torch::Tensor idx) {
@@ -347,6 +395,49 @@ torch::Tensor row_gather_scaled(torch::Tensor src, torch::Tensor idx, torch::Ten
                                        M, D, current_stream());
   });
   return out;
+}
+
+void ls_scatter_add_(torch::Tensor dst, torch::Tensor idx, torch::Tensor src,
+                     torch::Tensor gamma, torch::Tensor bias, torch::Tensor scale) {
+  CHECK_INPUT(dst);
+  CHECK_INPUT(src);
+  const int D = dst.size(-1);
+  TORCH_CHECK(D % 8 == 0, "ls_scatter: D must be a multiple of 8");
+  const long M = idx.numel();
+  DISPATCH_FLOAT_BF16(dst.scalar_type(), "ls_scatter_add_", [&] {
+    launch_ls_scatter_add<scalar_t>(
+        (scalar_t*)dst.data_ptr(), idx.data_ptr<long>(), (const scalar_t*)src.data_ptr(),
+        gamma.defined() ? (const scalar_t*)gamma.data_ptr() : nullptr,
+        bias.defined() ? (const scalar_t*)bias.data_ptr() : nullptr,
+        scale.defined() && scale.numel() > 0 ? scale.data_ptr<float>() : nullptr, M, D,
+        current_stream());
+  });
+}
+
+std::vector<torch::Tensor> ls_scatter_bwd(torch::Tensor dy, torch::Tensor idx,
+                                          torch::Tensor src, torch::Tensor gamma,
+                                          torch::Tensor bias, torch::Tensor scale) {
+  CHECK_INPUT(dy);
+  CHECK_INPUT(src);
+  const int D = dy.size(-1);
+  const long M = idx.numel();
+  auto dres = torch::empty_like(src);
+  auto fopt = dy.options().dtype(torch::kFloat);
+  auto dgamma = gamma.defined() ? torch::zeros({D}, fopt) : torch::Tensor();
+  auto dbias = bias.defined() ? torch::zeros({D}, fopt) : torch::Tensor();
+  DISPATCH_FLOAT_BF16(dy.scalar_type(), "ls_scatter_bwd", [&] {
+    launch_ls_scatter_bwd<scalar_t>(
+        (const scalar_t*)dy.data_ptr(), idx.data_ptr<long>(),
+        (const scalar_t*)src.data_ptr(),
+        gamma.defined() ? (const scalar_t*)gamma.data_ptr() : nullptr,
+        bias.defined() ? (const scalar_t*)bias.data_ptr() : nullptr,
+        scale.defined() && scale.numel() > 0 ? scale.data_ptr<float>() : nullptr,
+        (scalar_t*)dres.data_ptr(),
+        dgamma.defined() ? dgamma.data_ptr<float>() : nullptr,
+        dbias.defined() ? dbias.data_ptr<float>() : nullptr, M, D, current_stream());
+  });
+  return {dres, dgamma.defined() ? dgamma.to(dy.scalar_type()) : dgamma,
+          dbias.defined() ? dbias.to(dy.scalar_type()) : dbias};
 }
 
 torch::Tensor swiglu_fwd(torch::Tensor x12) {
@@ -961,6 +1052,10 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, mod) {
   mod.def("row_scatter_add_", &row_scatter_add_);
   mod.def("row_gather_scaled", &row_gather_scaled);
   mod.def("ls_axpy_fwd", &ls_axpy_fwd);
+  mod.def("ls_axpy_bias_fwd", &ls_axpy_bias_fwd);
+  mod.def("ls_axpy_bias_bwd", &ls_axpy_bias_bwd);
+  mod.def("ls_scatter_add_", &ls_scatter_add_);
+  mod.def("ls_scatter_bwd", &ls_scatter_bwd);
   mod.def("ls_axpy_bwd", &ls_axpy_bwd);
   mod.def("swiglu_fwd", &swiglu_fwd);
   mod.def("swiglu_bwd", &swiglu_bwd);

@@ -161,6 +161,149 @@ __global__ void ls_axpy_bwd_kernel(const T* __restrict__ dout, const T* __restri
   for (int e = 0; e < 8; ++e) atomicAdd(dgamma + col8 + e, dg[e]);
 }
 
+// ---- bias-fused variants (DINOV3_FUSED_RESIDUAL): out = x + gamma*(res+bias).
+// Separate kernels so the validated ls_axpy binaries above stay untouched.
+
+template <typename T>
+__global__ void ls_axpy_bias_fwd_kernel(const T* __restrict__ x, const T* __restrict__ res,
+                                        const T* __restrict__ gamma,
+                                        const T* __restrict__ bias, T* __restrict__ out,
+                                        long rows, int D) {
+  const long total8 = rows * (long)(D / 8);
+  for (long idx = blockIdx.x * (long)blockDim.x + threadIdx.x; idx < total8;
+       idx += (long)gridDim.x * blockDim.x) {
+    const int col8 = (int)(idx % (D / 8)) * 8;
+    T xb[8], rb[8], gb[8], bb[8], ob[8];
+    Vec8<T>::load(xb, x + idx * 8);
+    Vec8<T>::load(rb, res + idx * 8);
+    Vec8<T>::load(gb, gamma + col8);
+    Vec8<T>::load(bb, bias + col8);
+#pragma unroll
+    for (int e = 0; e < 8; ++e) {
+      float v = ScalarOps<T>::load(xb + e) +
+                ScalarOps<T>::load(gb + e) *
+                    (ScalarOps<T>::load(rb + e) + ScalarOps<T>::load(bb + e));
+      ScalarOps<T>::store(ob + e, v);
+    }
+    Vec8<T>::store(out + idx * 8, ob);
+  }
+}
+
+template <typename T>
+__global__ void ls_axpy_bias_bwd_kernel(const T* __restrict__ dout,
+                                        const T* __restrict__ res,
+                                        const T* __restrict__ gamma,
+                                        const T* __restrict__ bias, T* __restrict__ dres,
+                                        float* __restrict__ dgamma,
+                                        float* __restrict__ dbias, long rows, int D) {
+  const int col8 = (blockIdx.y * blockDim.x + threadIdx.x) * 8;
+  if (col8 >= D) return;
+  T gb[8], bb[8];
+  Vec8<T>::load(gb, gamma + col8);
+  Vec8<T>::load(bb, bias + col8);
+  float dg[8] = {0.f, 0.f, 0.f, 0.f, 0.f, 0.f, 0.f, 0.f};
+  float db_acc[8] = {0.f, 0.f, 0.f, 0.f, 0.f, 0.f, 0.f, 0.f};
+  for (long row = blockIdx.x; row < rows; row += gridDim.x) {
+    const long off = row * (long)D + col8;
+    T db[8], rb[8], ob[8];
+    Vec8<T>::load(db, dout + off);
+    Vec8<T>::load(rb, res + off);
+#pragma unroll
+    for (int e = 0; e < 8; ++e) {
+      const float g = ScalarOps<T>::load(db + e);
+      const float gm = ScalarOps<T>::load(gb + e);
+      dg[e] += g * (ScalarOps<T>::load(rb + e) + ScalarOps<T>::load(bb + e));
+      db_acc[e] += g * gm;
+      ScalarOps<T>::store(ob + e, g * gm);
+    }
+    Vec8<T>::store(dres + off, ob);
+  }
+#pragma unroll
+  for (int e = 0; e < 8; ++e) {
+    atomicAdd(dgamma + col8 + e, dg[e]);
+    atomicAdd(dbias + col8 + e, db_acc[e]);
+  }
+}
+
+// ---------------- fused LayerScale scatter-add (K10 + K11) ----------------
+// Stochastic-depth residual with the LayerScale gamma and the producing
+// GEMM's bias folded in (kills the standalone gamma multiply and the bias
+// epilogue + bias-grad reduction of the proj/fc2 Linears):
+//   dst[idx[r], c] += scale[r] * gamma[c] * (src[r, c] + bias[c])
+// Rows are disjoint (drop-path subsets), so no atomics on dst.
+
+template <typename T>
+__global__ void ls_scatter_add_kernel(T* __restrict__ dst, const long* __restrict__ idx,
+                                      const T* __restrict__ src,
+                                      const T* __restrict__ gamma,
+                                      const T* __restrict__ bias,
+                                      const float* __restrict__ scale, long M, int D) {
+  const int d8 = D / 8;
+  long total = M * (long)d8;
+  for (long i = blockIdx.x * (long)blockDim.x + threadIdx.x; i < total;
+       i += (long)gridDim.x * blockDim.x) {
+    const long r = i / d8;
+    const int c8 = (int)(i % d8) * 8;
+    const long drow = idx[r];
+    const float s = scale != nullptr ? scale[r] : 1.0f;
+    T sb[8], db[8], gb[8], bb[8];
+    Vec8<T>::load(sb, src + r * (long)D + c8);
+    Vec8<T>::load(db, dst + drow * (long)D + c8);
+    if (gamma != nullptr) Vec8<T>::load(gb, gamma + c8);
+    if (bias != nullptr) Vec8<T>::load(bb, bias + c8);
+#pragma unroll
+    for (int e = 0; e < 8; ++e) {
+      const float g = gamma != nullptr ? ScalarOps<T>::load(gb + e) : 1.0f;
+      const float bv = bias != nullptr ? ScalarOps<T>::load(bb + e) : 0.0f;
+      ScalarOps<T>::store(
+          db + e, ScalarOps<T>::load(db + e) + s * g * (ScalarOps<T>::load(sb + e) + bv));
+    }
+    Vec8<T>::store(dst + drow * (long)D + c8, db);
+  }
+}
+
+// backward: dres[r,c] = dy[idx[r],c] * gamma[c] * scale[r];
+//           dgamma[c] += sum_r dy[idx[r],c] * (src[r,c]+bias[c]) * scale[r];
+//           dbias[c]  += sum_r dy[idx[r],c] * gamma[c] * scale[r]
+template <typename T>
+__global__ void ls_scatter_bwd_kernel(const T* __restrict__ dy, const long* __restrict__ idx,
+                                      const T* __restrict__ src,
+                                      const T* __restrict__ gamma,
+                                      const T* __restrict__ bias,
+                                      const float* __restrict__ scale,
+                                      T* __restrict__ dres, float* __restrict__ dgamma,
+                                      float* __restrict__ dbias, long M, int D) {
+  const int col8 = (blockIdx.y * blockDim.x + threadIdx.x) * 8;
+  if (col8 >= D) return;
+  T gb[8], bb[8];
+  if (gamma != nullptr) Vec8<T>::load(gb, gamma + col8);
+  if (bias != nullptr) Vec8<T>::load(bb, bias + col8);
+  float dg[8] = {0.f, 0.f, 0.f, 0.f, 0.f, 0.f, 0.f, 0.f};
+  float db_acc[8] = {0.f, 0.f, 0.f, 0.f, 0.f, 0.f, 0.f, 0.f};
+  for (long r = blockIdx.x; r < M; r += gridDim.x) {
+    const long srow = idx[r];
+    const float s = scale != nullptr ? scale[r] : 1.0f;
+    T dyb[8], rb[8], ob[8];
+    Vec8<T>::load(dyb, dy + srow * (long)D + col8);
+    Vec8<T>::load(rb, src + r * (long)D + col8);
+#pragma unroll
+    for (int e = 0; e < 8; ++e) {
+      const float gv = ScalarOps<T>::load(dyb + e) * s;
+      const float gm = gamma != nullptr ? ScalarOps<T>::load(gb + e) : 1.0f;
+      const float bv = bias != nullptr ? ScalarOps<T>::load(bb + e) : 0.0f;
+      dg[e] += gv * (ScalarOps<T>::load(rb + e) + bv);
+      db_acc[e] += gv * gm;
+      ScalarOps<T>::store(ob + e, gv * gm);
+    }
+    Vec8<T>::store(dres + r * (long)D + col8, ob);
+  }
+#pragma unroll
+  for (int e = 0; e < 8; ++e) {
+    if (dgamma != nullptr) atomicAdd(dgamma + col8 + e, dg[e]);
+    if (dbias != nullptr) atomicAdd(dbias + col8 + e, db_acc[e]);
+  }
+}
+
 // ------------------------------ swiglu ---------------------------------
 // x12 = [rows, 2H] as [x1 | x2]; y = silu(x1) * x2.
 
@@ -343,6 +486,48 @@ void launch_ls_axpy_bwd(const T* dout, const T* res, const T* gamma, T* dres,
 }
 
 template <typename T>
+void launch_ls_axpy_bias_fwd(const T* x, const T* res, const T* gamma, const T* bias,
+                             T* out, long rows, int D, hipStream_t stream) {
+  long total8 = rows * (long)(D / 8);
+  int grid = (int)min((total8 + EW_BLOCK - 1) / EW_BLOCK, (long)2048);
+  hipLaunchKernelGGL((ls_axpy_bias_fwd_kernel<T>), dim3(grid), dim3(EW_BLOCK), 0, stream,
+                     x, res, gamma, bias, out, rows, D);
+}
+
+template <typename T>
+void launch_ls_axpy_bias_bwd(const T* dout, const T* res, const T* gamma, const T* bias,
+                             T* dres, float* dgamma, float* dbias, long rows, int D,
+                             hipStream_t stream) {
+  const int col_tiles = (D / 8 + EW_BLOCK - 1) / EW_BLOCK;
+  int row_grid = (int)min(rows, (long)(2048 / col_tiles + 1));
+  hipLaunchKernelGGL((ls_axpy_bias_bwd_kernel<T>), dim3(row_grid, col_tiles),
+                     dim3(EW_BLOCK), 0, stream, dout, res, gamma, bias, dres, dgamma,
+                     dbias, rows, D);
+}
+
+template <typename T>
+void launch_ls_scatter_add(T* dst, const long* idx, const T* src, const T* gamma,
+                           const T* bias, const float* scale, long M, int D,
+                           hipStream_t stream) {
+  long total = M * (long)(D / 8);
+  int grid = (int)min((total + EW_BLOCK - 1) / EW_BLOCK, (long)2048);
+  if (grid == 0) return;
+  hipLaunchKernelGGL((ls_scatter_add_kernel<T>), dim3(grid), dim3(EW_BLOCK), 0, stream,
+                     dst, idx, src, gamma, bias, scale, M, D);
+}
+
+template <typename T>
+void launch_ls_scatter_bwd(const T* dy, const long* idx, const T* src, const T* gamma,
+                           const T* bias, const float* scale, T* dres, float* dgamma,
+                           float* dbias, long M, int D, hipStream_t stream) {
+  const int col_tiles = (D / 8 + EW_BLOCK - 1) / EW_BLOCK;
+  int row_grid = (int)min(M > 0 ? M : 1, (long)(2048 / col_tiles + 1));
+  hipLaunchKernelGGL((ls_scatter_bwd_kernel<T>), dim3(row_grid, col_tiles), dim3(EW_BLOCK),
+                     0, stream, dy, idx, src, gamma, bias, scale, dres, dgamma, dbias, M,
+                     D);
+}
+
+template <typename T>
 void launch_row_gather(const T* src, const long* idx, T* out, long M, int D,
                        hipStream_t stream) {
   long total = M * (long)(D / 8);
@@ -405,6 +590,17 @@ void launch_rope_fwd(const T* x, const float* sin_t, const float* cos_t, T* y, l
                                       hipStream_t);                                  \
   template void launch_ls_axpy_bwd<T>(const T*, const T*, const T*, T*, float*,      \
                                       long, int, hipStream_t);                       \
+  template void launch_ls_axpy_bias_fwd<T>(const T*, const T*, const T*, const T*,   \
+                                           T*, long, int, hipStream_t);              \
+  template void launch_ls_axpy_bias_bwd<T>(const T*, const T*, const T*, const T*,   \
+                                           T*, float*, float*, long, int,            \
+                                           hipStream_t);                             \
+  template void launch_ls_scatter_add<T>(T*, const long*, const T*, const T*,        \
+                                         const T*, const float*, long, int,          \
+                                         hipStream_t);                               \
+  template void launch_ls_scatter_bwd<T>(const T*, const long*, const T*, const T*,  \
+                                         const T*, const float*, T*, float*, float*, \
+                                         long, int, hipStream_t);                    \
   template void launch_row_gather<T>(const T*, const long*, T*, long, int,           \
                                      hipStream_t);                                   \
   template void launch_row_scatter_add<T>(T*, const long*, const T*, const float*,   \

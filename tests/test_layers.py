@@ -171,3 +171,42 @@ def test_droppath_plan_keepall_matches_dense():
     blk.eval()
     out_dense = blk.forward_flat(flat.clone(), metas)
     assert torch.allclose(out_plan, out_dense, atol=1e-5)
+
+
+def _block_run(fused, drop_path, monkeypatch, checkpoint=False):
+    import os
+    from dinov3_amd.layers.attention import SelfAttention
+    from dinov3_amd.utils.utils import cat_keep_shapes
+    from dinov3_amd.layers.block import DropPathPlan
+
+    monkeypatch.setenv("DINOV3_FUSED_RESIDUAL", "1" if fused else "0")
+    torch.manual_seed(11)
+    blk = SelfAttentionBlock(dim=32, num_heads=2, qkv_bias=True,
+                             drop_path=drop_path, init_values=1e-2)
+    blk.train()
+    torch.manual_seed(12)
+    x = torch.randn(6, 5, 32)
+    flat, _, _ = cat_keep_shapes([x])
+    flat = flat.clone().requires_grad_(True)
+    metas = [SelfAttention._meta_for(x, None, 0)]
+    plan = None
+    if drop_path > 0:
+        torch.manual_seed(13)
+        plan = DropPathPlan(metas, 1.0 - drop_path, 2, flat.device)
+    out = blk.forward_flat(flat, metas, plan, 0, inplace_ok=not checkpoint)
+    out.float().pow(2).sum().backward()
+    grads = {n: p.grad.clone() for n, p in blk.named_parameters()}
+    return out.detach().clone(), grads, flat.grad.clone()
+
+
+@pytest.mark.parametrize("drop_path", [0.0, 0.5])
+def test_fused_residual_matches_default(drop_path, monkeypatch):
+    """DINOV3_FUSED_RESIDUAL folds gamma+proj/fc2-bias into the residual op;
+    on CPU both paths are plain torch math and must agree to fp tolerance."""
+    out_a, grads_a, dx_a = _block_run(False, drop_path, monkeypatch)
+    out_b, grads_b, dx_b = _block_run(True, drop_path, monkeypatch)
+    assert torch.allclose(out_a, out_b, atol=1e-5), (out_a - out_b).abs().max()
+    assert torch.allclose(dx_a, dx_b, atol=1e-5)
+    for n in grads_a:
+        assert torch.allclose(grads_a[n], grads_b[n], atol=1e-4), \
+            f"{n}: {(grads_a[n] - grads_b[n]).abs().max()}"
