@@ -113,3 +113,82 @@ def test_make_dataset_string_parsing():
     assert img.shape == (3, 224, 224)
     ds2 = make_dataset(dataset_str="ImageNet:split=VAL")
     assert len(ds2) == 50_000
+
+
+class TestGpuAugment:
+    def test_batched_jitter_matches_per_image(self):
+        import random
+        from dinov3_amd.data import transforms as T
+        from dinov3_amd.data.gpu_augment import (
+            batched_brightness, batched_contrast, batched_hue, batched_saturation)
+
+        torch.manual_seed(0)
+        imgs = torch.rand(4, 3, 32, 32)
+        bf = torch.tensor([0.7, 1.0, 1.3, 1.1])
+        cf = torch.tensor([0.8, 1.2, 1.0, 0.9])
+        sf = torch.tensor([1.1, 0.9, 1.0, 1.2])
+        hf = torch.tensor([0.05, -0.08, 0.0, 0.1])
+        outs = {
+            "b": batched_brightness(imgs, bf),
+            "c": batched_contrast(imgs, cf),
+            "s": batched_saturation(imgs, sf),
+            "h": batched_hue(imgs, hf),
+        }
+        for i in range(4):
+            assert torch.allclose(outs["b"][i], T.adjust_brightness(imgs[i], float(bf[i])), atol=1e-6)
+            assert torch.allclose(outs["c"][i], T.adjust_contrast(imgs[i], float(cf[i])), atol=1e-5)
+            assert torch.allclose(outs["s"][i], T.adjust_saturation(imgs[i], float(sf[i])), atol=1e-6)
+            assert torch.allclose(outs["h"][i], T.adjust_hue(imgs[i], float(hf[i])), atol=1e-5)
+
+    def test_batched_blur_matches_per_image(self):
+        from dinov3_amd.data.gpu_augment import GAUSS_KSIZE, batched_gaussian_blur
+
+        torch.manual_seed(1)
+        imgs = torch.rand(3, 3, 24, 24)
+        sigma = torch.tensor([0.5, 1.0, 1.4])
+        apply = torch.tensor([True, True, False])
+        out = batched_gaussian_blur(imgs, sigma, apply)
+        # identity where not applied
+        assert torch.allclose(out[2], imgs[2], atol=1e-6)
+        # per-image reference with the same capped kernel
+        for i in range(2):
+            k = GAUSS_KSIZE
+            x = torch.arange(k, dtype=torch.float32) - k // 2
+            kern = torch.exp(-0.5 * (x / float(sigma[i])) ** 2)
+            kern = kern / kern.sum()
+            kx = kern.view(1, 1, 1, k).expand(3, 1, 1, k)
+            ky = kern.view(1, 1, k, 1).expand(3, 1, k, 1)
+            ref = torch.nn.functional.conv2d(imgs[i:i+1], kx, padding=(0, k // 2), groups=3)
+            ref = torch.nn.functional.conv2d(ref, ky, padding=(k // 2, 0), groups=3)
+            assert torch.allclose(out[i], ref[0], atol=1e-5)
+
+    def test_batched_rrc_flip_box_semantics(self):
+        from dinov3_amd.data.gpu_augment import batched_rrc_flip
+
+        torch.manual_seed(2)
+        img = torch.rand(1, 3, 64, 64)
+        # full-image box, no flip -> bilinear resize of the whole image
+        boxes = torch.tensor([[0.0, 0.0, 64.0, 64.0]])
+        out = batched_rrc_flip(img, boxes, torch.tensor([False]), 32)
+        ref = torch.nn.functional.interpolate(img, size=(32, 32), mode="bilinear",
+                                              align_corners=False)
+        assert (out - ref).abs().max() < 2e-2
+        # flip equivariance: flip(crop(img)) == crop_flipped(img)
+        out_f = batched_rrc_flip(img, boxes, torch.tensor([True]), 32)
+        assert torch.allclose(out_f, out.flip(-1), atol=1e-5)
+
+    def test_full_pipeline_shapes_and_stats(self):
+        import random
+        from dinov3_amd.data.gpu_augment import GpuDataAugmentationDINO
+
+        random.seed(3)
+        torch.manual_seed(3)
+        aug = GpuDataAugmentationDINO(local_crops_number=4, global_crops_size=64,
+                                      local_crops_size=32)
+        imgs = (torch.rand(5, 3, 96, 96) * 255).to(torch.uint8)
+        out = aug(imgs)
+        assert out["global_crops"].shape == (10, 3, 64, 64)
+        assert out["local_crops"].shape == (20, 3, 32, 32)
+        assert torch.isfinite(out["global_crops"]).all()
+        # normalized output: roughly zero-centred
+        assert out["global_crops"].mean().abs() < 2.0
