@@ -184,3 +184,15 @@ def _full_meta_arch_worker(rank, port):
 
 def test_full_meta_arch_sharded_step():
     _run(_full_meta_arch_worker, 29617)
+
+
+def _fused_meta_arch_worker(rank, port):
+    os.environ["DINOV3_FUSED_RESIDUAL"] = "1"
+    _full_meta_arch_worker(rank, port)
+
+
+def test_full_meta_arch_sharded_step_fused_residual():
+    """Same end-to-end 2-rank step with the fused residual path on (CPU
+    fallback math): grad hooks must still fire for proj/fc2 biases whose
+    grads now come from the residual op."""
+    _run(_fused_meta_arch_worker, 29619)
