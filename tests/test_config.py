@@ -42,3 +42,34 @@ def test_scaling_rule_sqrt():
 def test_reference_style_config_loads(smoke_cfg):
     assert smoke_cfg.student.arch == "vit_small"
     assert smoke_cfg.dino.head_n_prototypes == 64
+
+
+def test_reference_recipe_configs_load():
+    """The reference's named train recipes exist here as delta configs and
+    produce complete configs through the merge (unknown recipe-only keys are
+    tolerated by the file merge)."""
+    import types
+
+    from dinov3_amd.configs import setup_config
+
+    for name in [
+        "dinov3_vit7b16_pretrain",
+        "dinov3_vit7b16_gram_anchor",
+        "dinov3_vit7b16_high_res_adapt",
+        "vitl_im1k_lin834_smol",
+        "multi_distillation_test",
+    ]:
+        args = types.SimpleNamespace(
+            config_file=f"dinov3_amd/configs/train/{name}.yaml", opts=[], output_dir="")
+        cfg = setup_config(args, apply_scaling=False)
+        assert cfg.train.batch_size_per_gpu > 0
+        assert cfg.student.arch.startswith("vit") or cfg.multidistillation.enabled
+    # vit7b pretrain recipe specifics survive the merge
+    args = types.SimpleNamespace(
+        config_file="dinov3_amd/configs/train/dinov3_vit7b16_pretrain.yaml",
+        opts=[], output_dir="")
+    cfg = setup_config(args, apply_scaling=False)
+    assert cfg.student.arch == "vit_7b"
+    assert cfg.dino.head_n_prototypes == 262144
+    assert cfg.student.ffn_layer == "swiglu64"
+    assert cfg.crops.global_crops_size == 256
