@@ -112,14 +112,88 @@ class SSLMetaArch(nn.Module):
                 self.student_backbone.set_grad_checkpointing(True)
                 logger.info("activation checkpointing enabled on the student backbone")
 
-        # the teacher tower never takes gradients and starts as a copy of the student
-        self._sync_teacher_from_student()
+        # the teacher tower never takes gradients; EMA mode starts it as a
+        # copy of the student, distillation mode replaces it with a frozen
+        # pretrained teacher (possibly a different architecture)
+        self.is_distillation_enabled = bool(config.distillation.enabled)
+        if self.is_distillation_enabled:
+            self._setup_distillation()
+        else:
+            self._sync_teacher_from_student()
         for module in (self.teacher_backbone, self.teacher_dino_head, self.teacher_ibot_head):
             module.requires_grad_(False)
         if self.gram_backbone is not None:
             self.gram_backbone.requires_grad_(False)
 
     # ------------------------------------------------------------------
+    def _setup_distillation(self) -> None:
+        """Replace the EMA teacher with a frozen teacher built from the
+        distillation config and checkpoint.
+
+        Intended semantics of reference dinov3_jax/train/ssl_meta_arch.py:
+        257-286 — whose version leaves the ibot head in an orphaned local
+        dict, passes head_hidden_dim as the dino head's out_dim and never
+        loads distillation.checkpoint_path; those defects are not
+        reproduced."""
+        from ..configs import get_default_config, load_yaml
+        from ..configs.config import _merge_into
+
+        cfg = self.config
+        logger.info("distillation: teacher config %s", cfg.distillation.full_cfg_path)
+        t_cfg = get_default_config()
+        _merge_into(t_cfg, load_yaml(cfg.distillation.full_cfg_path).to_plain(),
+                    strict=False)
+        assert t_cfg.ibot.separate_head is True
+        assert t_cfg.ibot.head_n_prototypes == cfg.ibot.head_n_prototypes
+        assert t_cfg.dino.head_n_prototypes == cfg.dino.head_n_prototypes
+        assert t_cfg.student.patch_size == cfg.student.patch_size
+
+        self.teacher_backbone, t_embed = build_model_from_cfg(t_cfg, only_teacher=True)
+        self.teacher_dino_head = DINOHead(
+            in_dim=t_embed, out_dim=t_cfg.dino.head_n_prototypes,
+            hidden_dim=t_cfg.dino.head_hidden_dim,
+            bottleneck_dim=t_cfg.dino.head_bottleneck_dim,
+            nlayers=t_cfg.dino.head_nlayers)
+        self.teacher_ibot_head = DINOHead(
+            in_dim=t_embed, out_dim=t_cfg.ibot.head_n_prototypes,
+            hidden_dim=t_cfg.ibot.head_hidden_dim,
+            bottleneck_dim=t_cfg.ibot.head_bottleneck_dim,
+            nlayers=t_cfg.ibot.head_nlayers)
+
+        path = cfg.distillation.checkpoint_path
+        if path and path != "ignore":
+            self._load_distillation_teacher(path)
+
+    def _load_distillation_teacher(self, path: str) -> None:
+        """Load teacher weights from a checkpoint: either a rank file / dir
+        in our checkpointer layout (payload["model"] holding the source
+        run's SSLMetaArch state dict) or a bare state-dict .pth; teacher_*
+        keys are selected, falling back to student_* (distilling from a
+        trained student)."""
+        import os
+
+        p = path
+        if os.path.isdir(p):
+            cand = os.path.join(p, "rank_0.pth")
+            p = cand if os.path.exists(cand) else p
+        payload = torch.load(p, map_location="cpu", weights_only=False)
+        state = payload.get("model", payload) if isinstance(payload, dict) else payload
+        for attr, prefixes in (
+            ("teacher_backbone", ("teacher_backbone.", "student_backbone.")),
+            ("teacher_dino_head", ("teacher_dino_head.", "student_dino_head.")),
+            ("teacher_ibot_head", ("teacher_ibot_head.", "student_ibot_head.")),
+        ):
+            sub = {}
+            for prefix in prefixes:
+                sub = {k[len(prefix):]: v for k, v in state.items() if k.startswith(prefix)}
+                if sub:
+                    break
+            if not sub:  # bare backbone/head state dict
+                sub = state
+            missing, unexpected = getattr(self, attr).load_state_dict(sub, strict=False)
+            logger.info("distillation %s: loaded %d tensors (missing %d, unexpected %d)",
+                        attr, len(sub), len(missing), len(unexpected))
+
     @torch.no_grad()
     def _sync_teacher_from_student(self) -> None:
         self.teacher_backbone.load_state_dict(self.student_backbone.state_dict())
@@ -141,7 +215,10 @@ class SSLMetaArch(nn.Module):
     @torch.no_grad()
     def update_ema(self, momentum: float) -> None:
         """teacher <- m*teacher + (1-m)*student, one fused in-place kernel
-        launch over a cached device-side plan."""
+        launch over a cached device-side plan. No-op in distillation mode:
+        the teacher is a frozen pretrained model, not an EMA."""
+        if getattr(self, "is_distillation_enabled", False):
+            return
         pairs = self._teacher_student_param_pairs()
         if pairs and pairs[0][0].is_cuda:
             from ..ops.mt_plan import MultiTensorPlan, ema_planned

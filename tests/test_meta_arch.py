@@ -76,3 +76,39 @@ def test_gram_loss_path(smoke_cfg):
     assert "gram_loss" in metrics
     assert torch.isfinite(loss)
     loss.backward()
+
+
+def test_distillation_mode(smoke_cfg, tmp_path):
+    """distillation.enabled: frozen teacher built from the distillation
+    config, EMA disabled, checkpoint weights loaded into the teacher."""
+    import copy
+    import torch as _torch
+
+    from dinov3_amd.train.ssl_meta_arch import SSLMetaArch
+
+    # source run: a normal smoke model whose weights we distill from
+    torch.manual_seed(0)
+    src = SSLMetaArch(copy.deepcopy(smoke_cfg))
+    ckpt = tmp_path / "teacher.pth"
+    _torch.save({"model": src.state_dict()}, ckpt)
+
+    cfg = copy.deepcopy(smoke_cfg)
+    cfg.distillation.enabled = True
+    cfg.distillation.full_cfg_path = "dinov3_amd/configs/train/vits_smoke.yaml"
+    cfg.distillation.checkpoint_path = str(ckpt)
+    torch.manual_seed(1)
+    model = SSLMetaArch(cfg)
+    assert model.is_distillation_enabled
+    # teacher came from the checkpoint, not from this student
+    t = dict(model.teacher_backbone.named_parameters())
+    s = dict(model.student_backbone.named_parameters())
+    src_t = dict(src.teacher_backbone.named_parameters())
+    name = next(iter(t))
+    assert torch.equal(t[name], src_t[name])
+    assert not torch.equal(t[name], s[name])
+    # EMA is a no-op
+    before = t[name].clone()
+    model.update_ema(0.5)
+    assert torch.equal(dict(model.teacher_backbone.named_parameters())[name], before)
+    # teacher takes no grad
+    assert not any(p.requires_grad for p in model.teacher_backbone.parameters())
