@@ -3,6 +3,8 @@ checkpointing, eval protocols, hubconf interop, multi-resolution loader."""
 
 import sys
 
+import pytest
+
 import torch
 
 
@@ -136,3 +138,35 @@ def test_clip_tokenizer_roundtrip():
     assert t.decode(t.encode("a photo of a cat")) == "a photo of a cat"
     # punctuation re-spacing is lossy (as in CLIP's own tokenizer)
     assert t.decode(t.encode("hello, world!")).replace(" ", "") == "hello,world!"
+
+
+@pytest.mark.parametrize("fused", [False, True])
+def test_checkpointing_with_droppath_and_fused_residual(fused, monkeypatch):
+    """Activation checkpointing must replay the batched DropPathPlan subsets
+    identically, including with the fused residual path (which must not
+    mutate the checkpoint-saved flat buffer in place)."""
+    from dinov3_amd.models.vision_transformer import vit_small
+
+    monkeypatch.setenv("DINOV3_FUSED_RESIDUAL", "1" if fused else "0")
+    torch.manual_seed(0)
+    m = vit_small(img_size=32, layerscale_init=1e-5, drop_path_rate=0.4)
+    m.train()
+    x = torch.randn(4, 3, 32, 32)
+
+    torch.manual_seed(7)
+    out1 = m.forward_features(x)["x_norm_clstoken"]
+    out1.float().pow(2).sum().backward()
+    g1 = {n: p.grad.clone() for n, p in m.named_parameters() if p.grad is not None}
+    m.zero_grad(set_to_none=True)
+
+    m.set_grad_checkpointing(True)
+    torch.manual_seed(7)
+    out2 = m.forward_features(x)["x_norm_clstoken"]
+    out2.float().pow(2).sum().backward()
+    g2 = {n: p.grad.clone() for n, p in m.named_parameters() if p.grad is not None}
+
+    assert torch.allclose(out1, out2, atol=1e-5)
+    assert g1.keys() == g2.keys()
+    for n in g1:
+        assert torch.allclose(g1[n], g2[n], atol=1e-4), \
+            f"{n}: {(g1[n] - g2[n]).abs().max()}"
