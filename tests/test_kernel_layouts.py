@@ -1,0 +1,141 @@
+"""Symbolic (host-side) verification of HIP kernel LDS layouts.
+
+These tests transcribe the exact integer index arithmetic of the staged
+kernels and check, cell by cell, that (a) staging writes never collide or
+leave the allocated region and (b) every compute-phase read sees exactly
+the (row, column) element the math requires. This is how the round-1
+bwd_dkv 64-row staging bug (32-row column stride kept for 64 staged rows —
+see docs/ROADMAP_R2.md) would have been caught without hardware.
+"""
+
+import pytest
+
+
+def _c_row(r, h):
+    return (r & 3) + 8 * (r >> 2) + 4 * h
+
+
+@pytest.mark.parametrize("HD", [64, 128])
+@pytest.mark.parametrize("N", [65, 128, 197, 300])
+def test_dkv64_staging_layout(HD, N):
+    NT = 256
+    QB, SQB = 32, 64
+    QT_STRIDE = SQB + 8
+    HALF = HD // 2
+    PAIRS_PER_ROW = HALF // 8
+    PER_ROW = HD // 8
+    QK_ITEMS = SQB * PAIRS_PER_ROW
+    DO_ITEMS = SQB * PER_ROW
+    QK_IPT = (QK_ITEMS + NT - 1) // NT
+    DO_IPT = (DO_ITEMS + NT - 1) // NT
+    qt_size = HD * QT_STRIDE
+
+    for qbase0 in range(0, N, SQB):
+        qt, dot, lse, dl = {}, {}, {}, {}
+        writes = set()
+        for tid in range(NT):
+            for j in range(QK_IPT):
+                idx = tid + j * NT
+                if idx < QK_ITEMS:
+                    lrow = idx // PAIRS_PER_ROW
+                    c0 = (idx % PAIRS_PER_ROW) * 8
+                    for e in range(8):
+                        for c in (c0 + e, c0 + HALF + e):
+                            addr = c * QT_STRIDE + lrow
+                            assert addr < qt_size
+                            assert ("q", addr) not in writes
+                            writes.add(("q", addr))
+                            qt[addr] = (qbase0 + lrow, c)
+            for j in range(DO_IPT):
+                idx = tid + j * NT
+                if idx < DO_ITEMS:
+                    lrow = idx // PER_ROW
+                    c8 = (idx % PER_ROW) * 8
+                    for e in range(8):
+                        addr = (c8 + e) * QT_STRIDE + lrow
+                        assert addr < qt_size
+                        assert ("do", addr) not in writes
+                        writes.add(("do", addr))
+                        dot[addr] = (qbase0 + lrow, c8 + e)
+            if tid < SQB:
+                lse[tid] = qbase0 + tid
+                dl[tid] = qbase0 + tid
+
+        for tid in range(NT):
+            lane = tid & 63
+            hhalf = lane >> 5
+            l31 = lane & 31
+            for sub in range(2):
+                if qbase0 + sub * QB >= N:
+                    continue
+                qrow_off = sub * QB
+                qbase = qbase0 + sub * QB
+                for s in range(HD // 16):
+                    for e in range(8):
+                        d = s * 16 + hhalf * 8 + e
+                        addr = d * QT_STRIDE + qrow_off + l31
+                        assert qt[addr] == (qbase + l31, d)
+                        assert dot[addr] == (qbase + l31, d)
+                for r in range(16):
+                    want = qbase + _c_row(r, hhalf)
+                    assert lse[qrow_off + _c_row(r, hhalf)] == want
+                    assert dl[qrow_off + _c_row(r, hhalf)] == want
+                for t in range(HD // 32):
+                    for half16 in (0, 16):
+                        for e in range(8):
+                            row = qrow_off + half16 + hhalf * 8 + e
+                            addr = (t * 32 + l31) * QT_STRIDE + row
+                            want = (qbase0 + row, t * 32 + l31)
+                            assert dot[addr] == want
+                            assert qt[addr] == want
+
+
+@pytest.mark.parametrize("HD,NT", [(64, 128), (64, 256), (128, 128), (128, 256)])
+def test_dkv32_staging_layout(HD, NT):
+    """Same check for the production 32-row kernel (and its buggy-history
+    counterexample: with 64 staged rows this stride would fail)."""
+    QB = 32
+    QT_STRIDE = QB + 8
+    HALF = HD // 2
+    PAIRS_PER_ROW = HALF // 8
+    PER_ROW = HD // 8
+    QK_ITEMS = QB * PAIRS_PER_ROW
+    DO_ITEMS = QB * PER_ROW
+    DO_IPT = (DO_ITEMS + NT - 1) // NT
+    qt_size = HD * QT_STRIDE
+    qt, dot = {}, {}
+    for tid in range(NT):
+        if tid < QK_ITEMS:
+            lrow = tid // PAIRS_PER_ROW
+            c0 = (tid % PAIRS_PER_ROW) * 8
+            for e in range(8):
+                for c in (c0 + e, c0 + HALF + e):
+                    addr = c * QT_STRIDE + lrow
+                    assert addr < qt_size
+                    assert addr not in qt
+                    qt[addr] = (lrow, c)
+        for j in range(DO_IPT):
+            idx = tid + j * NT
+            if idx < DO_ITEMS:
+                lrow = idx // PER_ROW
+                c8 = (idx % PER_ROW) * 8
+                for e in range(8):
+                    addr = (c8 + e) * QT_STRIDE + lrow
+                    assert addr < qt_size
+                    assert addr not in dot
+                    dot[addr] = (lrow, c8 + e)
+    for lane in range(64):
+        hhalf = lane >> 5
+        l31 = lane & 31
+        for s in range(HD // 16):
+            for e in range(8):
+                d = s * 16 + hhalf * 8 + e
+                assert qt[d * QT_STRIDE + l31] == (l31, d)
+                assert dot[d * QT_STRIDE + l31] == (l31, d)
+        for t in range(HD // 32):
+            for half16 in (0, 16):
+                for e in range(8):
+                    row = half16 + hhalf * 8 + e
+                    addr = (t * 32 + l31) * QT_STRIDE + row
+                    assert dot[addr] == (row, t * 32 + l31)
+                    assert qt[addr] == (row, t * 32 + l31)
