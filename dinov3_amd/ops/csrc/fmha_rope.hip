@@ -555,16 +555,23 @@ __global__ __launch_bounds__(NT) void bwd_dkv_kernel(
   constexpr int PER_ROW = HD / 8;
   constexpr int QK_ITEMS = QB * PAIRS_PER_ROW;   // <= 256
   constexpr int DO_ITEMS = QB * PER_ROW;         // <= 512
+  // QK_ITEMS can exceed NT (HD=128, NT=128 -> 256 items): loop like the
+  // dO staging. QK_IPT==1 for every other variant, so those paths are
+  // unchanged. (Single-shot staging here silently dropped half the Q tile
+  // for hd-128 small-N — caught by tests/test_kernel_layouts.py.)
+  constexpr int QK_IPT = (QK_ITEMS + NT - 1) / NT;
   constexpr int DO_IPT = (DO_ITEMS + NT - 1) / NT;
-  bf16x8 qlo{}, qhi{}, doreg[DO_IPT];
+  bf16x8 qlo[QK_IPT], qhi[QK_IPT], doreg[DO_IPT];
   float lse_reg = INFINITY, d_reg = 0.f;
   auto issue_loads = [&](int qbase0) {
-    if (threadIdx.x < QK_ITEMS) {
-      const int qrow = qbase0 + threadIdx.x / PAIRS_PER_ROW;
-      const int c0 = (threadIdx.x % PAIRS_PER_ROW) * 8;
-      const bool ok = qrow < N;
-      qlo = ok ? load8(qv.at(qrow, 0, c0)) : bf16x8{};
-      qhi = ok ? load8(qv.at(qrow, 0, c0 + HALF)) : bf16x8{};
+#pragma unroll
+    for (int j = 0; j < QK_IPT; ++j) {
+      const int idx = threadIdx.x + j * NT;
+      const int qrow = qbase0 + idx / PAIRS_PER_ROW;
+      const int c0 = (idx % PAIRS_PER_ROW) * 8;
+      const bool ok = idx < QK_ITEMS && qrow < N;
+      qlo[j] = ok ? load8(qv.at(qrow, 0, c0)) : bf16x8{};
+      qhi[j] = ok ? load8(qv.at(qrow, 0, c0 + HALF)) : bf16x8{};
     }
 #pragma unroll
     for (int j = 0; j < DO_IPT; ++j) {
@@ -581,17 +588,21 @@ __global__ __launch_bounds__(NT) void bwd_dkv_kernel(
     }
   };
   auto write_tile = [&](int qbase0) {
-    if (threadIdx.x < QK_ITEMS) {
-      const int lrow = threadIdx.x / PAIRS_PER_ROW;
-      const int c0 = (threadIdx.x % PAIRS_PER_ROW) * 8;
-      const int qrow = qbase0 + lrow;
-      const int p = qrow - prefix;
-      if (use_rope && qrow < N && p >= 0)
-        rope_rotate8(qlo, qhi, sin_t + (long)p * HD, cos_t + (long)p * HD, c0);
 #pragma unroll
-      for (int e = 0; e < 8; ++e) {
-        qt_lds[(c0 + e) * QT_STRIDE + lrow] = reinterpret_cast<__hip_bfloat16*>(&qlo)[e];
-        qt_lds[(c0 + HALF + e) * QT_STRIDE + lrow] = reinterpret_cast<__hip_bfloat16*>(&qhi)[e];
+    for (int j = 0; j < QK_IPT; ++j) {
+      const int idx = threadIdx.x + j * NT;
+      if (idx < QK_ITEMS) {
+        const int lrow = idx / PAIRS_PER_ROW;
+        const int c0 = (idx % PAIRS_PER_ROW) * 8;
+        const int qrow = qbase0 + lrow;
+        const int p = qrow - prefix;
+        if (use_rope && qrow < N && p >= 0)
+          rope_rotate8(qlo[j], qhi[j], sin_t + (long)p * HD, cos_t + (long)p * HD, c0);
+#pragma unroll
+        for (int e = 0; e < 8; ++e) {
+          qt_lds[(c0 + e) * QT_STRIDE + lrow] = reinterpret_cast<__hip_bfloat16*>(&qlo[j])[e];
+          qt_lds[(c0 + HALF + e) * QT_STRIDE + lrow] = reinterpret_cast<__hip_bfloat16*>(&qhi[j])[e];
+        }
       }
     }
 #pragma unroll

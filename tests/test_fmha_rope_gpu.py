@@ -62,7 +62,7 @@ def test_fmha_rope_no_rope():
     assert err < 0.03, f"fwd err {err}"
 
 
-@pytest.mark.parametrize("N,hd,prefix", [(37, 64, 1), (197, 64, 1), (197, 128, 1)])
+@pytest.mark.parametrize("N,hd,prefix", [(37, 64, 1), (197, 64, 1), (197, 128, 1), (54, 128, 5), (64, 128, 0)])
 def test_fmha_rope_bwd(N, hd, prefix):
     from dinov3_amd.ops.flat_attention import flat_multi_fmha
 

@@ -101,19 +101,22 @@ def test_dkv32_staging_layout(HD, NT):
     PER_ROW = HD // 8
     QK_ITEMS = QB * PAIRS_PER_ROW
     DO_ITEMS = QB * PER_ROW
+    QK_IPT = (QK_ITEMS + NT - 1) // NT
     DO_IPT = (DO_ITEMS + NT - 1) // NT
     qt_size = HD * QT_STRIDE
     qt, dot = {}, {}
     for tid in range(NT):
-        if tid < QK_ITEMS:
-            lrow = tid // PAIRS_PER_ROW
-            c0 = (tid % PAIRS_PER_ROW) * 8
-            for e in range(8):
-                for c in (c0 + e, c0 + HALF + e):
-                    addr = c * QT_STRIDE + lrow
-                    assert addr < qt_size
-                    assert addr not in qt
-                    qt[addr] = (lrow, c)
+        for j in range(QK_IPT):
+            idx = tid + j * NT
+            if idx < QK_ITEMS:
+                lrow = idx // PAIRS_PER_ROW
+                c0 = (idx % PAIRS_PER_ROW) * 8
+                for e in range(8):
+                    for c in (c0 + e, c0 + HALF + e):
+                        addr = c * QT_STRIDE + lrow
+                        assert addr < qt_size
+                        assert addr not in qt
+                        qt[addr] = (lrow, c)
         for j in range(DO_IPT):
             idx = tid + j * NT
             if idx < DO_ITEMS:
