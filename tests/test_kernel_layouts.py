@@ -142,3 +142,67 @@ def test_dkv32_staging_layout(HD, NT):
                     addr = (t * 32 + l31) * QT_STRIDE + row
                     assert dot[addr] == (row, t * 32 + l31)
                     assert qt[addr] == (row, t * 32 + l31)
+
+
+@pytest.mark.parametrize("HD,NT", [(64, 128), (64, 256), (128, 128), (128, 256)])
+def test_fwd_kv_staging_layout(HD, NT):
+    """fwd_kernel 64-key super-tile: K rows into k_lds [2*KVB][HD+8], V^T
+    into vt_lds [HD][2*KVB+8]; compute reads must see the staged element."""
+    KVB = 32
+    LDS_STRIDE = HD + 8
+    VT_STRIDE = 2 * KVB + 8
+    HALF = HD // 2
+    PAIRS_PER_ROW = HALF // 8
+    PER_ROW = HD // 8
+    K_ITEMS = 2 * KVB * PAIRS_PER_ROW
+    K_IPT = (K_ITEMS + NT - 1) // NT
+    V_ITEMS = 2 * KVB * PER_ROW
+    V_IPT = (V_ITEMS + NT - 1) // NT
+    k_size = 2 * KVB * LDS_STRIDE
+    v_size = HD * VT_STRIDE
+
+    k_lds, vt_lds = {}, {}
+    kw, vw = set(), set()
+    for tid in range(NT):
+        for j in range(K_IPT):
+            item = tid + j * NT
+            if item < K_ITEMS:
+                lrow = item // PAIRS_PER_ROW
+                c0 = (item % PAIRS_PER_ROW) * 8
+                for e in range(8):
+                    for c in (c0 + e, HALF + c0 + e):
+                        addr = lrow * LDS_STRIDE + c
+                        assert addr < k_size
+                        assert addr not in kw
+                        kw.add(addr)
+                        k_lds[addr] = (lrow, c)
+        for j in range(V_IPT):
+            idx = tid + j * NT
+            if idx < V_ITEMS:
+                row = idx // PER_ROW
+                c8 = (idx % PER_ROW) * 8
+                for e in range(8):
+                    addr = (c8 + e) * VT_STRIDE + row
+                    assert addr < v_size
+                    assert addr not in vw
+                    vw.add(addr)
+                    vt_lds[addr] = (row, c8 + e)
+
+    for lane in range(64):
+        hhalf = lane >> 5
+        l31 = lane & 31
+        for sub in range(2):
+            krow_off = sub * KVB
+            # S^T A-fragment: K[krow_off + l31][k = s*16 + hhalf*8 + e]
+            for s in range(HD // 16):
+                for e in range(8):
+                    c = s * 16 + hhalf * 8 + e
+                    addr = (krow_off + l31) * LDS_STRIDE + c
+                    assert k_lds[addr] == (krow_off + l31, c)
+            # O accumulation: V^T[d = t*32+l31][k-rows]
+            for t in range(HD // 32):
+                for half16 in (0, 16):
+                    for e in range(8):
+                        row = krow_off + half16 + hhalf * 8 + e
+                        addr = (t * 32 + l31) * VT_STRIDE + row
+                        assert vt_lds[addr] == (row, t * 32 + l31)
