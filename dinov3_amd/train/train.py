@@ -11,6 +11,7 @@ overlaps backward (parallel/ddp.py or the sharding engine in parallel/fsdp.py).
 
 from __future__ import annotations
 
+import math
 import argparse
 import logging
 import sys
@@ -268,6 +269,15 @@ def do_train(cfg, model: SSLMetaArch, resume: bool = True, max_iterations: int =
 
     nan_count = 0
     iteration = start_iter
+    # reconstruct how many gram refreshes a resumed run has already done
+    # (reference train.py:605-615)
+    num_gram_updates = 0
+    if (model.gram_use_loss and cfg.gram.rep_update and start_iter > 0
+            and start_iter >= cfg.gram.it_first_update):
+        num_gram_updates = math.ceil(
+            (start_iter + 1 - cfg.gram.it_first_update) / cfg.gram.update_frequency)
+        logger.info("gram teacher refreshed %d times before iteration %d",
+                    num_gram_updates, start_iter)
     recorded_losses = []
     reference_losses = None
     if compare_losses_to:
@@ -345,9 +355,16 @@ def do_train(cfg, model: SSLMetaArch, resume: bool = True, max_iterations: int =
         if profiling and device.type == "cuda":
             torch.cuda.nvtx.range_pop()
 
-        if model.gram_use_loss and cfg.gram.rep_update and it >= cfg.gram.it_first_update:
-            if (it - cfg.gram.it_first_update) % cfg.gram.update_frequency == 0:
-                model.update_gram_teacher()
+        # gram-teacher refresh cadence (reference train.py:668-678): absolute
+        # (it+1) phase, capped at gram.max_updates refreshes over the run
+        if (model.gram_use_loss and cfg.gram.rep_update
+                and (it + 1) >= cfg.gram.it_first_update
+                and (it + 1) % cfg.gram.update_frequency == 0
+                and (cfg.gram.max_updates is None
+                     or num_gram_updates < cfg.gram.max_updates)):
+            logger.info("updating gram teacher from EMA teacher after iteration %d", it)
+            model.update_gram_teacher()
+            num_gram_updates += 1
 
         metric_logger.update(
             lr=lr, wd=wd, mom=mom, last_layer_lr=last_layer_lr, teacher_temp=teacher_temp,
