@@ -206,3 +206,64 @@ def test_fwd_kv_staging_layout(HD, NT):
                         row = krow_off + half16 + hhalf * 8 + e
                         addr = (t * 32 + l31) * VT_STRIDE + row
                         assert vt_lds[addr] == (row, t * 32 + l31)
+
+
+@pytest.mark.parametrize("HD,NT", [(64, 128), (64, 256), (128, 128), (128, 256)])
+def test_dq_kv_staging_layout(HD, NT):
+    """bwd_dq_kernel: K rows into k_lds/[2KVB][HD+8] AND transposed into
+    kt_lds/[HD][2KVB+8]; V rows into v_lds; all compute reads verified."""
+    KVB = 32
+    LDS_STRIDE = HD + 8
+    KT_STRIDE = 2 * KVB + 8
+    HALF = HD // 2
+    PAIRS_PER_ROW = HALF // 8
+    PER_ROW = HD // 8
+    K_ITEMS = 2 * KVB * PAIRS_PER_ROW
+    K_IPT = (K_ITEMS + NT - 1) // NT
+    V_ITEMS = 2 * KVB * PER_ROW
+    V_IPT = (V_ITEMS + NT - 1) // NT
+    k_size = 2 * KVB * LDS_STRIDE
+    kt_size = HD * KT_STRIDE
+
+    k_lds, v_lds, kt_lds = {}, {}, {}
+    for tid in range(NT):
+        for j in range(K_IPT):
+            item = tid + j * NT
+            if item < K_ITEMS:
+                lrow = item // PAIRS_PER_ROW
+                c0 = (item % PAIRS_PER_ROW) * 8
+                for e in range(8):
+                    for c in (c0 + e, HALF + c0 + e):
+                        addr = lrow * LDS_STRIDE + c
+                        assert addr < k_size and addr not in k_lds
+                        k_lds[addr] = (lrow, c)
+                        taddr = c * KT_STRIDE + lrow
+                        assert taddr < kt_size and taddr not in kt_lds
+                        kt_lds[taddr] = (lrow, c)
+        for j in range(V_IPT):
+            idx = tid + j * NT
+            if idx < V_ITEMS:
+                row = idx // PER_ROW
+                c8 = (idx % PER_ROW) * 8
+                for e in range(8):
+                    addr = row * LDS_STRIDE + c8 + e
+                    assert addr < k_size and addr not in v_lds
+                    v_lds[addr] = (row, c8 + e)
+
+    for lane in range(64):
+        hhalf = lane >> 5
+        l31 = lane & 31
+        for sub in range(2):
+            krow_off = sub * KVB
+            for s in range(HD // 16):
+                for e in range(8):
+                    c = s * 16 + hhalf * 8 + e
+                    addr = (krow_off + l31) * LDS_STRIDE + c
+                    assert k_lds[addr] == (krow_off + l31, c)
+                    assert v_lds[addr] == (krow_off + l31, c)
+            for t in range(HD // 32):
+                for half16 in (0, 16):
+                    for e in range(8):
+                        row = krow_off + half16 + hhalf * 8 + e
+                        addr = (t * 32 + l31) * KT_STRIDE + row
+                        assert kt_lds[addr] == (row, t * 32 + l31)
