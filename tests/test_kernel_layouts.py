@@ -267,3 +267,32 @@ def test_dq_kv_staging_layout(HD, NT):
                         row = krow_off + half16 + hhalf * 8 + e
                         addr = (t * 32 + l31) * KT_STRIDE + row
                         assert kt_lds[addr] == (row, t * 32 + l31)
+
+
+def test_patch_embed_staging_layout():
+    """patch_embed::fwd_kernel: 256 threads stage a 128x32 A tile and a
+    128x32 W tile per K-step ([row][40] layout); MFMA reads must line up."""
+    A_STRIDE = 40
+    a_lds, b_lds = {}, {}
+    for tid in range(256):
+        s_row = tid // 2
+        s_half = (tid % 2) * 16
+        for e in range(16):
+            addr = s_row * A_STRIDE + s_half + e
+            assert addr < 128 * A_STRIDE
+            assert addr not in a_lds
+            a_lds[addr] = (s_row, s_half + e)   # (tile row, k within step)
+            b_lds[addr] = (s_row, s_half + e)
+    for wave in range(4):
+        for lane in range(64):
+            hhalf = lane >> 5
+            l31 = lane & 31
+            for ks in range(2):
+                for e in range(8):
+                    k = ks * 16 + hhalf * 8 + e
+                    addr = (wave * 32 + l31) * A_STRIDE + k
+                    assert a_lds[addr] == (wave * 32 + l31, k)
+                for t in range(4):
+                    addr = (t * 32 + l31) * A_STRIDE + ks * 16 + hhalf * 8
+                    for e in range(8):
+                        assert b_lds[addr + e] == (t * 32 + l31, ks * 16 + hhalf * 8 + e)
