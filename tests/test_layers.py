@@ -210,3 +210,18 @@ def test_fused_residual_matches_default(drop_path, monkeypatch):
     for n in grads_a:
         assert torch.allclose(grads_a[n], grads_b[n], atol=1e-4), \
             f"{n}: {(grads_a[n] - grads_b[n]).abs().max()}"
+
+
+def test_patch14_forward_backward():
+    """Non-16 patch sizes (e.g. the ViT-g/14 target config) take the
+    patchify+GEMM fallback; full fwd+bwd must work."""
+    from dinov3_amd.models.vision_transformer import vit_small
+
+    torch.manual_seed(0)
+    m = vit_small(img_size=56, patch_size=14, layerscale_init=1e-5)
+    m.train()
+    x = torch.randn(2, 3, 56, 56)
+    out = m.forward_features(x)
+    assert out["x_norm_patchtokens"].shape == (2, 16, 384)
+    out["x_norm_clstoken"].float().pow(2).sum().backward()
+    assert m.patch_embed.proj.weight.grad is not None
