@@ -90,10 +90,13 @@ def save_checkpoint(output_dir: str, iteration: int, model: torch.nn.Module,
 def load_checkpoint(ckpt_dir: os.PathLike, model: torch.nn.Module,
                     optimizer: Optional[Any] = None, strict: bool = True) -> Dict[str, Any]:
     ckpt_dir = Path(ckpt_dir)
-    rank = parallel.get_rank()
-    path = ckpt_dir / f"rank_{rank}.pth"
-    if not path.exists():
-        path = ckpt_dir / "rank_0.pth"
+    if ckpt_dir.is_file():  # MODEL.WEIGHTS may point at a rank file directly
+        path = ckpt_dir
+    else:
+        rank = parallel.get_rank()
+        path = ckpt_dir / f"rank_{rank}.pth"
+        if not path.exists():
+            path = ckpt_dir / "rank_0.pth"
     payload = torch.load(path, map_location="cpu", weights_only=False)
     missing, unexpected = model.load_state_dict(payload["model"], strict=strict)
     if missing or unexpected:

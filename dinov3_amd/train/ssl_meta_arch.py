@@ -88,23 +88,58 @@ class SSLMetaArch(nn.Module):
         self.dino_koleo_loss_weight = config.dino.koleo_loss_weight
         self.ibot_loss_weight = config.ibot.loss_weight
 
-        # gram anchoring
+        # gram anchoring (reference ssl_meta_arch.py:165-254; option semantics
+        # honored incl. the ones the reference only validates/logs)
         self.gram_use_loss = config.gram.use_loss
         self.gram_img_level = config.gram.img_level
         self.gram_loss_weight = config.gram.loss_weight
         self.gram_compute_stats = config.gram.compute_stats
-        self.has_gram_teacher = self.gram_use_loss
+        self.gram_ema_teacher = bool(config.gram.get("ema_teacher", False))
+        self.gram_ckpt = config.gram.get("ckpt", None)
+        self.gram_tokens_used = config.gram.get("tokens_used", "all")
+        self.gram_it_load_ema_teacher = config.gram.get("it_load_ema_teacher", -1)
+        # ema_teacher=True: gram features come straight from the EMA teacher,
+        # no separate frozen gram backbone
+        self.has_gram_teacher = self.gram_use_loss and not self.gram_ema_teacher
+        self.gram_loss_schedule = None
         if self.gram_use_loss:
-            gram_backbone, _ = build_model_from_cfg(config, only_teacher=True)
-            self.gram_backbone = gram_backbone
+            if self.gram_ema_teacher and self.gram_ckpt is not None:
+                raise ValueError("gram.ema_teacher and gram.ckpt are mutually exclusive")
+            if (not self.gram_ema_teacher and self.gram_ckpt is None
+                    and self.gram_it_load_ema_teacher < 0):
+                raise ValueError("without gram.ckpt, gram.it_load_ema_teacher must be >= 0")
+            assert self.gram_tokens_used in ("all", "masked", "unmasked")
+            if self.gram_tokens_used in ("masked", "unmasked"):
+                assert not self.gram_img_level, "token-subset gram needs img_level=False"
+            assert not (self.gram_ema_teacher and config.gram.rep_update)
+            if config.crops.gram_teacher_crops_size is not None and self.gram_ema_teacher:
+                raise ValueError("crops.gram_teacher_crops_size must be unset with gram.ema_teacher")
+            if self.has_gram_teacher:
+                gram_backbone, _ = build_model_from_cfg(config, only_teacher=True)
+                self.gram_backbone = gram_backbone
+                if self.gram_ckpt not in (None, "ignore"):
+                    payload = torch.load(self.gram_ckpt, map_location="cpu", weights_only=False)
+                    state = payload.get("model", payload) if isinstance(payload, dict) else payload
+                    sub = {k[len("teacher_backbone."):]: v for k, v in state.items()
+                           if k.startswith("teacher_backbone.")} or state
+                    self.gram_backbone.load_state_dict(sub, strict=False)
+                    logger.info("gram teacher loaded from %s", self.gram_ckpt)
+            else:
+                self.gram_backbone = None
             self.gram_loss = GramLoss(
                 apply_norm=config.gram.normalized,
                 img_level=config.gram.img_level,
                 remove_neg=config.gram.remove_neg,
                 remove_only_teacher_neg=config.gram.remove_only_teacher_neg,
             )
+            if config.gram.get("loss_weight_schedule"):
+                self.gram_loss_schedule = self._weight_schedule(config.gram.loss_weight_schedule)
         else:
             self.gram_backbone = None
+        self.dino_local_loss_schedule = None
+        if config.dino.get("reweight_dino_local_loss") and config.dino.get("local_loss_weight_schedule"):
+            self.dino_local_loss_schedule = self._weight_schedule(
+                config.dino.local_loss_weight_schedule)
 
         # activation checkpointing (selective per-block recompute) on request
         if config.train.checkpointing or config.train.checkpointing_full:
@@ -126,6 +161,28 @@ class SSLMetaArch(nn.Module):
             self.gram_backbone.requires_grad_(False)
 
     # ------------------------------------------------------------------
+    def _weight_schedule(self, sched_cfg):
+        """Per-iteration loss-weight array from a {start, peak, end,
+        warmup_epochs, cosine_epochs} block (reference ssl_meta_arch.py:
+        150-163, 185-198; missing keys default sensibly)."""
+        from .cosine_lr_scheduler import linear_warmup_cosine_decay
+
+        epoch_len = self.config.train.OFFICIAL_EPOCH_LENGTH
+        total = max(int(self.config.optim.epochs * epoch_len), 1)
+        start = float(sched_cfg.get("start", 0.0))
+        peak = float(sched_cfg.get("peak", start))
+        end = float(sched_cfg.get("end", peak))
+        cosine = sched_cfg.get("cosine_epochs", None)
+        return linear_warmup_cosine_decay(
+            start=start, peak=peak, end=end,
+            warmup_iterations=int(sched_cfg.get("warmup_epochs", 0) * epoch_len),
+            total_iterations=total,
+            cosine_iterations=int(cosine * epoch_len) if cosine is not None else None,
+        )
+
+    def _schedule_at(self, schedule, iteration: int) -> float:
+        return float(schedule[int(iteration)])  # scheduler clamps past the end
+
     def _setup_distillation(self) -> None:
         """Replace the EMA teacher with a frozen teacher built from the
         distillation config and checkpoint.
@@ -416,11 +473,8 @@ class SSLMetaArch(nn.Module):
         )
         loss_dict["dino_local_crops_loss"] = dino_local_crops_loss
         local_weight = 1.0
-        if self.config.dino.reweight_dino_local_loss:
-            from .cosine_lr_scheduler import linear_warmup_cosine_decay  # noqa: F401
-
-            sched = self.config.dino.local_loss_weight_schedule
-            local_weight = float(sched.peak)
+        if self.dino_local_loss_schedule is not None:
+            local_weight = self._schedule_at(self.dino_local_loss_schedule, iteration)
         loss_dict["dino_local_loss_weight"] = local_weight
         loss_accumulator = loss_accumulator + self.dino_loss_weight * dino_local_scale * local_weight * dino_local_crops_loss
 
@@ -447,12 +501,24 @@ class SSLMetaArch(nn.Module):
         loss_accumulator = loss_accumulator + self.ibot_loss_weight * ibot_loss
 
         if self.gram_use_loss and gram_global:
-            gram_loss = self.gram_loss(
-                gram_global["student_patches"], gram_global["teacher_patches"], img_level=self.gram_img_level,
-            )
+            s_patches = gram_global["student_patches"]
+            t_patches = gram_global["teacher_patches"]
+            if self.gram_tokens_used != "all":
+                # masked/unmasked token subset over the flattened global-crop
+                # tokens (intended semantics; the reference validates this
+                # option but never consumes it)
+                flat_mask = masks.reshape(-1)
+                keep = flat_mask if self.gram_tokens_used == "masked" else ~flat_mask
+                s2 = s_patches.reshape(-1, s_patches.shape[-1])[keep]
+                t2 = t_patches.reshape(-1, t_patches.shape[-1])[keep]
+                gram_loss = self.gram_loss(s2.unsqueeze(0), t2.unsqueeze(0), img_level=False)
+            else:
+                gram_loss = self.gram_loss(s_patches, t_patches, img_level=self.gram_img_level)
+            gram_weight = (self._schedule_at(self.gram_loss_schedule, iteration)
+                           if self.gram_loss_schedule is not None else self.gram_loss_weight)
             loss_dict["gram_loss"] = gram_loss
-            loss_dict["gram_loss_weight"] = self.gram_loss_weight
-            loss_accumulator = loss_accumulator + self.gram_loss_weight * gram_loss
+            loss_dict["gram_loss_weight"] = gram_weight
+            loss_accumulator = loss_accumulator + gram_weight * gram_loss
 
         loss_dict["total_loss"] = loss_accumulator
         return loss_accumulator, loss_dict

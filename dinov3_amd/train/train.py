@@ -355,6 +355,13 @@ def do_train(cfg, model: SSLMetaArch, resume: bool = True, max_iterations: int =
         if profiling and device.type == "cuda":
             torch.cuda.nvtx.range_pop()
 
+        # one-time gram init from the EMA teacher (gram.it_load_ema_teacher,
+        # used when no gram checkpoint is configured)
+        if (model.gram_use_loss and model.has_gram_teacher
+                and model.gram_it_load_ema_teacher >= 0
+                and it == model.gram_it_load_ema_teacher):
+            logger.info("loading gram teacher from EMA teacher at iteration %d", it)
+            model.update_gram_teacher()
         # gram-teacher refresh cadence (reference train.py:668-678): absolute
         # (it+1) phase, capped at gram.max_updates refreshes over the run
         if (model.gram_use_loss and cfg.gram.rep_update
@@ -461,13 +468,35 @@ def main(argv=None):
         logger.info("--test-ibot: dino/koleo loss weights zeroed")
     if args.multi_distillation or cfg.multidistillation.enabled:
         cfg.multidistillation.enabled = True
-        from .multidist_meta_arch import MultiDistillationMetaArch
+        assert cfg.MODEL.META_ARCHITECTURE == "MultiDistillationMetaArch", \
+            "multi-distillation runs need MODEL.META_ARCHITECTURE=MultiDistillationMetaArch"
+    # meta-arch dispatch by config (reference train.py:293-299)
+    from .multidist_meta_arch import MultiDistillationMetaArch
 
+    meta_arch_cls = {
+        "SSLMetaArch": SSLMetaArch,
+        "MultiDistillationMetaArch": MultiDistillationMetaArch,
+    }.get(cfg.MODEL.META_ARCHITECTURE)
+    if meta_arch_cls is None:
+        raise ValueError(f"unknown MODEL.META_ARCHITECTURE {cfg.MODEL.META_ARCHITECTURE}")
+    if meta_arch_cls is MultiDistillationMetaArch:
         MultiDistillationMetaArch(cfg)  # validates the subgroup layout
         raise NotImplementedError("multi-distillation training loop is a stub (reference parity)")
-    model = SSLMetaArch(cfg)
+    model = meta_arch_cls(cfg)
     if args.eval_only:
-        return do_test(cfg, model, 0)
+        # load weights for evaluation: latest checkpoint unless MODEL.WEIGHTS
+        # names one explicitly (reference train.py:302-309)
+        weights = cfg.MODEL.get("WEIGHTS", "")
+        iteration = 0
+        if weights:
+            payload = load_checkpoint(weights, model, strict=False)
+            iteration = payload.get("iteration", 0)
+        elif not args.no_resume and args.output_dir:
+            latest = find_latest_checkpoint(args.output_dir)
+            if latest is not None:
+                payload = load_checkpoint(latest, model, strict=False)
+                iteration = payload.get("iteration", 0)
+        return do_test(cfg, model, iteration)
     import os
     import time
 

@@ -66,6 +66,7 @@ def test_loss_weights_scale(smoke_cfg):
 def test_gram_loss_path(smoke_cfg):
     cfg = copy.deepcopy(smoke_cfg)
     cfg.gram.use_loss = True
+    cfg.gram.ckpt = "ignore"
     cfg.gram.img_level = True
     cfg.gram.remove_neg = True
     torch.manual_seed(0)
@@ -112,3 +113,47 @@ def test_distillation_mode(smoke_cfg, tmp_path):
     assert torch.equal(dict(model.teacher_backbone.named_parameters())[name], before)
     # teacher takes no grad
     assert not any(p.requires_grad for p in model.teacher_backbone.parameters())
+
+
+def test_gram_tokens_used_and_schedules(smoke_cfg):
+    """gram.tokens_used masked/unmasked subsets and the gram/dino-local
+    loss-weight schedules (options the reference validates but drops)."""
+    cfg = copy.deepcopy(smoke_cfg)
+    cfg.gram.use_loss = True
+    cfg.gram.ckpt = "ignore"
+    cfg.gram.img_level = False
+    cfg.gram.remove_neg = True
+    cfg.gram.tokens_used = "masked"
+    cfg.gram.loss_weight_schedule = {"start": 0.0, "peak": 0.0, "end": 2.0,
+                                     "warmup_epochs": 0, "cosine_epochs": 1}
+    cfg.dino.reweight_dino_local_loss = True
+    cfg.dino.local_loss_weight_schedule = {"start": 1.0, "peak": 1.0, "end": 0.5,
+                                           "warmup_epochs": 0, "cosine_epochs": 1}
+    torch.manual_seed(0)
+    model = SSLMetaArch(cfg)
+    model.train()
+    batch = _synthetic_batch(cfg)
+    loss0, m0 = model(batch, teacher_temp=0.07, iteration=0)
+    assert torch.isfinite(loss0)
+    total = cfg.train.OFFICIAL_EPOCH_LENGTH * cfg.optim.epochs
+    loss1, m1 = model(batch, teacher_temp=0.07, iteration=total - 1)
+    # schedules move over the run: gram weight 0 -> 2, local weight 1 -> 0.5
+    assert abs(float(m0["gram_loss_weight"]) - 0.0) < 1e-6
+    assert abs(float(m1["gram_loss_weight"]) - 2.0) < 1e-3
+    assert abs(float(m0["dino_local_loss_weight"]) - 1.0) < 1e-6
+    assert abs(float(m1["dino_local_loss_weight"]) - 0.5) < 1e-3
+
+
+def test_gram_ema_teacher_mode(smoke_cfg):
+    cfg = copy.deepcopy(smoke_cfg)
+    cfg.gram.use_loss = True
+    cfg.gram.ema_teacher = True
+    cfg.gram.remove_neg = True
+    cfg.gram.rep_update = False
+    torch.manual_seed(0)
+    model = SSLMetaArch(cfg)
+    assert model.gram_backbone is None and not model.has_gram_teacher
+    model.train()
+    batch = _synthetic_batch(cfg)
+    loss, metrics = model(batch, teacher_temp=0.07, iteration=0)
+    assert "gram_loss" in metrics and torch.isfinite(loss)
