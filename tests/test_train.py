@@ -144,3 +144,28 @@ def test_scheduler_consistency_with_config(smoke_cfg):
     assert 0.0 <= s["momentum"][0] <= 1.0
     # teacher temp warms from warmup_teacher_temp to teacher_temp
     assert abs(s["teacher_temp"][total + 10] - smoke_cfg.teacher.teacher_temp) < 1e-9
+
+
+def test_main_cli_end_to_end(tmp_path):
+    """`python -m dinov3_amd.train.train` CLI: parse args, train a few
+    iterations, write checkpoints, then auto-resume in a second invocation."""
+    from dinov3_amd.train.train import main
+
+    out = str(tmp_path / "run")
+    argv = [
+        "--config-file", "dinov3_amd/configs/train/vits_smoke.yaml",
+        "--output-dir", out,
+        "--max-iterations", "2",
+        "train.batch_size_per_gpu=2",
+        "checkpointing.period=1",
+        "crops.local_crops_number=2",
+    ]
+    main(argv)
+    import os
+
+    ckpts = os.listdir(os.path.join(out, "ckpt"))
+    assert ckpts, "no checkpoint written"
+    # resume picks up from the saved iteration and advances
+    main(argv)
+    ckpts2 = sorted(int(x) for x in os.listdir(os.path.join(out, "ckpt")) if x.isdigit())
+    assert ckpts2 and ckpts2[-1] >= 2
