@@ -102,3 +102,36 @@ def test_gram_loss_batch_level():
     y = torch.randn(2, 5, 8)
     out = g(x, y, img_level=False)
     assert out.ndim == 0 and out > 0
+
+
+def test_factored_sinkhorn_matches_dense_reference():
+    """The factored u/v iteration (what the HIP kernels compute — per-
+    iteration scalars cancel; GPU numerics in tests/test_loss_ops_gpu.py)
+    must equal a dense Sinkhorn-Knopp implementation."""
+    torch.manual_seed(0)
+    M, K, T, iters = 12, 40, 0.5, 3
+    x = torch.randn(M, K)
+    # pure-torch transcription of ops/proto_scores.sinkhorn_knopp_factored
+    E = (x / T).exp()          # [M, K]
+    u = torch.ones(M)
+    for it in range(iters):
+        A = (u[:, None] * E).sum(dim=0)          # colsum of u*E  -> [K]
+        v = 1.0 / (A * K)
+        u = 1.0 / (E * v[None, :]).sum(dim=1)    # rowsum of E*v  -> [M]
+    probs = E * u[:, None] * v[None, :]
+
+    # dense reference following the reference algorithm
+    # (dinov3_jax/loss/dino_clstoken_loss.py:35-62): Q [K, M]
+    Q = (x / T).T.exp()
+    Q = Q / Q.sum()
+    B = Q.shape[1]
+    Kn = Q.shape[0]
+    for _ in range(iters):
+        Q = Q / Q.sum(dim=1, keepdim=True) / Kn
+        Q = Q / Q.sum(dim=0, keepdim=True) / B
+    Q = Q * B
+    ref = Q.T  # [M, K]
+    err = (probs - ref).abs().max().item()
+    assert err < 1e-5, f"factored vs dense mismatch {err}"
+    # rows are probability distributions
+    assert torch.allclose(probs.sum(dim=-1), torch.ones(M), atol=1e-4)
