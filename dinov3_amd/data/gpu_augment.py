@@ -24,9 +24,8 @@ Works identically on CPU tensors, which is how the equivalence tests run.
 
 from __future__ import annotations
 
-import math
 import random
-from typing import Dict, List, Optional, Tuple
+from typing import Dict, Tuple
 
 import torch
 import torch.nn.functional as F

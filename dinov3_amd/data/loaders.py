@@ -9,7 +9,7 @@ from __future__ import annotations
 
 import logging
 from enum import Enum
-from typing import Any, Callable, Optional
+from typing import Callable, Optional
 
 import torch
 

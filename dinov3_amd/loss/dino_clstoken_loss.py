@@ -6,8 +6,6 @@ Parity: dinov3_jax/loss/dino_clstoken_loss.py:14-95. Cross-device reductions
 
 from __future__ import annotations
 
-from typing import Optional
-
 import torch
 import torch.distributed as dist
 import torch.nn as nn

@@ -8,7 +8,7 @@ x_norm_patchtokens dict the SSL meta-arch consumes, sizes tiny..large.
 
 from __future__ import annotations
 
-from typing import List, Optional, Sequence
+from typing import Optional, Sequence
 
 import torch
 import torch.nn as nn

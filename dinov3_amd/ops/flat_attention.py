@@ -11,7 +11,6 @@ buffer in backward.
 
 from __future__ import annotations
 
-import math
 from typing import List, Optional, Tuple
 
 import torch

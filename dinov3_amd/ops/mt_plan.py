@@ -8,7 +8,7 @@ and the plan rebuilt if any tensor re-allocated.
 
 from __future__ import annotations
 
-from typing import List, Optional, Sequence
+from typing import Sequence
 
 import torch
 
