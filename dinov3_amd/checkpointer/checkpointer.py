@@ -63,14 +63,24 @@ def keep_last_n_checkpoints(output_dir: str, n: int, keep_every: int = 0) -> Non
 
 def save_checkpoint(output_dir: str, iteration: int, model: torch.nn.Module,
                     optimizer: Optional[Any] = None, extra: Optional[Dict[str, Any]] = None,
-                    max_to_keep: int = 3, keep_every: int = 0) -> Path:
-    """Each rank writes its own shard file (rank0 also writes metadata)."""
+                    max_to_keep: int = 3, keep_every: int = 0,
+                    skip_prefixes: Optional[Any] = None) -> Path:
+    """Each rank writes its own shard file (rank0 also writes metadata).
+
+    skip_prefixes: state-dict key prefixes to omit — the
+    `register_dont_save_hooks` analogue the reference README promises
+    (dinov3_jax/README.md:35, call site train.py:453-457 is dead); used to
+    skip the frozen distillation teacher, which never changes."""
     ckpt_dir = _ckpt_root(output_dir) / str(iteration)
     ckpt_dir.mkdir(parents=True, exist_ok=True)
     rank = parallel.get_rank()
+    state = model.state_dict()
+    if skip_prefixes:
+        state = {k: v for k, v in state.items()
+                 if not any(k.startswith(p) for p in skip_prefixes)}
     payload: Dict[str, Any] = {
         "iteration": iteration,
-        "model": model.state_dict(),
+        "model": state,
         "world_size": parallel.get_world_size(),
     }
     if optimizer is not None:

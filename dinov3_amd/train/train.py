@@ -269,6 +269,10 @@ def do_train(cfg, model: SSLMetaArch, resume: bool = True, max_iterations: int =
 
     nan_count = 0
     iteration = start_iter
+    # frozen distillation teacher never changes: skip it in checkpoints
+    skip_save_prefixes = (
+        ("teacher_backbone.", "teacher_dino_head.", "teacher_ibot_head.")
+        if getattr(model, "is_distillation_enabled", False) else None)
     # reconstruct how many gram refreshes a resumed run has already done
     # (reference train.py:605-615)
     num_gram_updates = 0
@@ -407,6 +411,7 @@ def do_train(cfg, model: SSLMetaArch, resume: bool = True, max_iterations: int =
             save_checkpoint(
                 output_dir, it, model, optimizer,
                 max_to_keep=cfg.checkpointing.max_to_keep, keep_every=cfg.checkpointing.keep_every,
+                skip_prefixes=skip_save_prefixes,
             )
         iteration += 1
 
@@ -420,6 +425,7 @@ def do_train(cfg, model: SSLMetaArch, resume: bool = True, max_iterations: int =
         save_checkpoint(
             output_dir, iteration - 1, model, optimizer,
             max_to_keep=cfg.checkpointing.max_to_keep, keep_every=cfg.checkpointing.keep_every,
+            skip_prefixes=skip_save_prefixes,
         )
     metric_logger.synchronize_between_processes()
     logger.info("training done at iteration %d", iteration)
