@@ -249,7 +249,11 @@ def do_train(cfg, model: SSLMetaArch, resume: bool = True, max_iterations: int =
     if resume and output_dir:
         latest = find_latest_checkpoint(output_dir)
         if latest is not None:
-            payload = load_checkpoint(latest, model, optimizer, strict=True)
+            # distillation checkpoints omit the frozen teacher (it is rebuilt
+            # from distillation.checkpoint_path at construction)
+            payload = load_checkpoint(
+                latest, model, optimizer,
+                strict=not getattr(model, "is_distillation_enabled", False))
             start_iter = payload["iteration"] + 1
             logger.info("resumed at iteration %d", start_iter)
 

@@ -169,3 +169,36 @@ def test_main_cli_end_to_end(tmp_path):
     main(argv)
     ckpts2 = sorted(int(x) for x in os.listdir(os.path.join(out, "ckpt")) if x.isdigit())
     assert ckpts2 and ckpts2[-1] >= 2
+
+
+def test_distillation_checkpoint_skips_frozen_teacher(tmp_path, smoke_cfg):
+    """Distillation-mode checkpoints omit the frozen teacher and resume
+    cleanly (teacher is rebuilt from the distillation checkpoint)."""
+    import copy
+
+    import torch as _torch
+
+    from dinov3_amd.checkpointer import load_checkpoint, save_checkpoint
+    from dinov3_amd.train.ssl_meta_arch import SSLMetaArch
+
+    torch.manual_seed(0)
+    src = SSLMetaArch(copy.deepcopy(smoke_cfg))
+    teacher_ckpt = tmp_path / "teacher.pth"
+    _torch.save({"model": src.state_dict()}, teacher_ckpt)
+
+    cfg = copy.deepcopy(smoke_cfg)
+    cfg.distillation.enabled = True
+    cfg.distillation.full_cfg_path = "dinov3_amd/configs/train/vits_smoke.yaml"
+    cfg.distillation.checkpoint_path = str(teacher_ckpt)
+    model = SSLMetaArch(cfg)
+    out = str(tmp_path / "run")
+    save_checkpoint(out, 0, model,
+                    skip_prefixes=("teacher_backbone.", "teacher_dino_head.",
+                                   "teacher_ibot_head."))
+    payload = _torch.load(tmp_path / "run" / "ckpt" / "0" / "rank_0.pth",
+                          map_location="cpu", weights_only=False)
+    assert not any(k.startswith("teacher_") for k in payload["model"])
+    assert any(k.startswith("student_") for k in payload["model"])
+    # resume-style partial load works
+    model2 = SSLMetaArch(cfg)
+    load_checkpoint(tmp_path / "run" / "ckpt" / "0", model2, strict=False)
