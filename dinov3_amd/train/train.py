@@ -425,7 +425,8 @@ def do_train(cfg, model: SSLMetaArch, resume: bool = True, max_iterations: int =
         with open(record_losses_to, "w") as f:
             json.dump(recorded_losses, f)
         logger.info("recorded %d iterations of losses to %s", len(recorded_losses), record_losses_to)
-    if output_dir:
+    already_saved = ckpt_period > 0 and iteration % ckpt_period == 0 and iteration > 0
+    if output_dir and not already_saved:
         save_checkpoint(
             output_dir, iteration - 1, model, optimizer,
             max_to_keep=cfg.checkpointing.max_to_keep, keep_every=cfg.checkpointing.keep_every,
