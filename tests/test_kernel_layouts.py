@@ -296,3 +296,32 @@ def test_patch_embed_staging_layout():
                     addr = (t * 32 + l31) * A_STRIDE + ks * 16 + hhalf * 8
                     for e in range(8):
                         assert b_lds[addr + e] == (t * 32 + l31, ks * 16 + hhalf * 8 + e)
+
+
+def test_multi_tensor_plan_chunk_coverage():
+    """MultiTensorPlan chunk table: every element of every tensor covered by
+    exactly one (tensor, offset) chunk of <= CHUNK elements."""
+    import torch
+
+    from dinov3_amd.ops.mt_plan import CHUNK, MultiTensorPlan
+
+    sizes = [1, CHUNK - 1, CHUNK, CHUNK + 1, 3 * CHUNK + 17]
+    tensors = [torch.zeros(s) for s in sizes]
+    plan = MultiTensorPlan([tensors])
+    ct = plan.ct.tolist()
+    co = plan.co.tolist()
+    covered = {i: [] for i in range(len(sizes))}
+    for t, off in zip(ct, co):
+        n = min(CHUNK, sizes[t] - off)
+        assert n > 0
+        covered[t].append((off, off + n))
+    for i, s in enumerate(sizes):
+        spans = sorted(covered[i])
+        assert spans[0][0] == 0 and spans[-1][1] == s
+        for (a0, a1), (b0, b1) in zip(spans, spans[1:]):
+            assert a1 == b0, f"gap/overlap in tensor {i}: {spans}"
+    # pointer-change detection triggers a table refresh
+    old = plan.ptrs.clone()
+    tensors[2].data = torch.zeros(CHUNK)
+    assert plan.check_pointers() is False
+    assert not torch.equal(plan.ptrs, old)
