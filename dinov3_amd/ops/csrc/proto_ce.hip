@@ -432,7 +432,7 @@ void launch_sinkhorn_fact_colsum(const __hip_bfloat16* x, const float* u, float*
                                  int M, long K, float inv_temp, hipStream_t stream) {
   const int kw = (int)((K + CE_BLOCK * 8 - 1) / (CE_BLOCK * 8));
   const int mt = (M + SK_MT - 1) / SK_MT;
-  if (mt > 1) hipMemsetAsync(A, 0, K * sizeof(float), stream);
+  if (mt > 1) (void)hipMemsetAsync(A, 0, K * sizeof(float), stream);
   hipLaunchKernelGGL(sinkhorn_fact_colsum_kernel, dim3(kw, mt), dim3(CE_BLOCK), 0, stream,
                      x, u, A, M, K, inv_temp);
 }
