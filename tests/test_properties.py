@@ -75,3 +75,56 @@ def test_sinkhorn_rows_are_distributions(m, k, temp, iters):
     assert probs.shape == (m, k)
     assert (probs >= 0).all()
     assert torch.allclose(probs.sum(dim=-1), torch.ones(m), atol=1e-4)
+
+
+@settings(max_examples=25, deadline=None)
+@given(b=st.integers(1, 6), nloc=st.integers(1, 4),
+       prob=st.floats(0.0, 1.0), rmin=st.floats(0.05, 0.3), rspan=st.floats(0.05, 0.3))
+def test_collate_mask_invariants(b, nloc, prob, rmin, rspan):
+    from dinov3_amd.data import MaskingGenerator, collate_data_and_cast
+
+    gs, ls, p = 32, 16, 16
+    n_tokens = (gs // p) ** 2
+    gen = MaskingGenerator(input_size=(gs // p, gs // p),
+                           max_num_patches=max(1, n_tokens // 2))
+    samples = [({"global_crops": [torch.randn(3, gs, gs) for _ in range(2)],
+                 "local_crops": [torch.randn(3, ls, ls) for _ in range(nloc)]}, ())
+               for _ in range(b)]
+    out = collate_data_and_cast(samples, mask_ratio_tuple=(rmin, min(0.5, rmin + rspan)),
+                                mask_probability=prob, dtype=torch.float32,
+                                n_tokens=n_tokens, mask_generator=gen)
+    masks = out["collated_masks"]
+    idx = out["mask_indices_list"]
+    w = out["masks_weight"]
+    assert masks.shape == (2 * b, n_tokens)
+    assert out["collated_global_crops"].shape == (2 * b, 3, gs, gs)
+    assert out["collated_local_crops"].shape == (nloc * b, 3, ls, ls)
+    # mask_indices_list is exactly the nonzero positions of the flat mask
+    assert torch.equal(idx, masks.flatten().nonzero().flatten())
+    assert int(out["n_masked_patches"][0]) == int(masks.sum())
+    assert out["upperbound"] >= int(masks.sum())
+    # per masked sample, its weights sum to 1
+    assert w.shape[0] == idx.shape[0]
+    if idx.numel():
+        sample_of = idx // n_tokens
+        for s in sample_of.unique().tolist():
+            assert abs(float(w[sample_of == s].sum()) - 1.0) < 1e-5
+
+
+@settings(max_examples=20, deadline=None)
+@given(st.dictionaries(st.sampled_from(["train", "optim", "dino"]),
+                       st.dictionaries(st.sampled_from(["a_key", "b_key"]),
+                                       st.one_of(st.integers(), st.floats(allow_nan=False),
+                                                 st.text(max_size=5)), max_size=2),
+                       max_size=3))
+def test_config_merge_roundtrip(over):
+    """Merging a delta then re-diffing yields the same delta (tolerant mode)."""
+    from dinov3_amd.configs import get_default_config
+    from dinov3_amd.configs.config import _merge_into
+
+    cfg = get_default_config()
+    _merge_into(cfg, over, strict=False)
+    for section, kv in over.items():
+        for k, v in kv.items():
+            got = cfg[section][k]
+            assert got == v or (got != got and v != v)  # NaN-safe
