@@ -202,3 +202,33 @@ def test_distillation_checkpoint_skips_frozen_teacher(tmp_path, smoke_cfg):
     # resume-style partial load works
     model2 = SSLMetaArch(cfg)
     load_checkpoint(tmp_path / "run" / "ckpt" / "0", model2, strict=False)
+
+
+def test_main_eval_only_with_weights(tmp_path):
+    """--eval-only loads MODEL.WEIGHTS (or the latest checkpoint) before
+    running the eval protocols (reference train.py:302-309)."""
+    import torch as _torch
+
+    from dinov3_amd.train.train import main
+
+    # make a checkpoint with a trained model
+    out = str(tmp_path / "src")
+    main([
+        "--config-file", "dinov3_amd/configs/train/vits_smoke.yaml",
+        "--output-dir", out, "--max-iterations", "1",
+        "train.batch_size_per_gpu=2", "checkpointing.period=1",
+        "crops.local_crops_number=2",
+        "train.dataset_path=Synthetic:split=TRAIN:length=16",
+    ])
+    ckpt = str(tmp_path / "src" / "ckpt" / "0" / "rank_0.pth")
+    import os
+
+    assert os.path.exists(ckpt)
+    results = main([
+        "--config-file", "dinov3_amd/configs/train/vits_smoke.yaml",
+        "--output-dir", str(tmp_path / "eval"), "--eval-only",
+        f"MODEL.WEIGHTS={ckpt}",
+        "train.dataset_path=Synthetic:split=TRAIN:length=16",
+        "crops.local_crops_number=2",
+    ])
+    assert results and "knn_top1" in results
