@@ -28,14 +28,28 @@ class MultiDistillationMetaArch(nn.Module):
         super().__init__()
         self.config = config
         assert config.multidistillation.enabled
-        students = config.multidistillation.get("students", [])
+        students = config.multidistillation.get("students", []) or []
+        # recipe schema: students[i].ranks_range = [lo, hi) (reference
+        # multi_distillation_test.yaml / dinov3_vitl16_lvd1689m_distilled.yaml)
+        ranges = [tuple(s["ranks_range"]) for s in students]
+        if not ranges:
+            raise ValueError("multidistillation.students must not be empty")
+        ranges.sort()
+        if ranges[0][0] != 0:
+            raise ValueError(f"subgroup ranges must start at rank 0, got {ranges}")
+        for (a0, a1), (b0, b1) in zip(ranges, ranges[1:]):
+            if a1 != b0:
+                raise ValueError(f"subgroup ranges must be contiguous, got {ranges}")
+        self.total_ranks = ranges[-1][1]
+        self.subgroup_sizes: List[int] = [hi - lo for lo, hi in ranges]
+        self.rank_ranges = ranges
         world = parallel.get_world_size()
-        self.subgroup_sizes: List[int] = [s.get("ranks", 1) for s in students] or [world]
-        if sum(self.subgroup_sizes) != world:
+        if world > 1 and self.total_ranks != world:
             raise ValueError(
-                f"multidistillation subgroups {self.subgroup_sizes} must cover world size {world}"
-            )
-        logger.info("multidistillation subgroups: %s", self.subgroup_sizes)
+                f"multidistillation layout covers {self.total_ranks} ranks "
+                f"but world size is {world}")
+        logger.info("multidistillation subgroups: %s (total %d ranks)",
+                    self.subgroup_sizes, self.total_ranks)
 
     def forward(self, *args, **kwargs):
         raise NotImplementedError(
