@@ -48,7 +48,7 @@ def main():
     R, D = 2 * 64 * 197 + 8 * 64 * 37, 1024
     H4 = 4 * D
 
-    from dinov3_amd.ops import bias_gelu, hip_ops, l2_normalize, layernorm
+    from dinov3_amd.ops import bias_gelu, hip_ops, l2_normalize, layer_norm
     from dinov3_amd.ops.ls_axpy import ls_axpy
 
     ops = hip_ops()
@@ -57,7 +57,7 @@ def main():
     b = torch.zeros(D, device=dev).bfloat16()
 
     row("layernorm_fwd [R,1024]",
-        timeit(lambda: layernorm(x, w, b), it),
+        timeit(lambda: layer_norm(x, w, b), it),
         timeit(lambda: torch.nn.functional.layer_norm(x, (D,), w, b), it),
         R * D * 2 * 2)
 
