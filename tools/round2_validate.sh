@@ -21,6 +21,10 @@ DINOV3_FUSED_RESIDUAL=1 timeout 180 python -m pytest \
     > "$OUT/pytest_fused.log" 2>&1
 echo "fused_suite=$?" >> "$OUT/summary.txt"
 
+# 2b. per-kernel micro-benchmarks (HIP vs torch)
+timeout 120 python tools/bench_kernels.py --iters 30 > "$OUT/bench_kernels.log" 2>&1
+echo "kernel_bench=$?" >> "$OUT/summary.txt"
+
 # 3. bench: baseline, then each flag, then both
 timeout 200 python bench.py --steps 15 --warmup 4 2>/dev/null | tail -1 > "$OUT/bench_base.json"
 DINOV3_FMHA_DKV64=1 timeout 200 python bench.py --steps 15 --warmup 4 2>/dev/null \
