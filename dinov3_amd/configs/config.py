@@ -161,7 +161,9 @@ def setup_config(args, strict_cfg: bool = True, apply_scaling: bool = True) -> D
         cfg.train.output_dir = output_dir
     if apply_scaling:
         apply_scaling_rules_to_cfg(cfg)
-    if cfg.train.output_dir:
+    if cfg.train.output_dir and cfg.train.output_dir not in (".", "./"):
+        # skip the config echo for the degenerate cwd output dir (the schema
+        # default) so library/test usage doesn't litter the working tree
         try:
             write_config(cfg, cfg.train.output_dir)
         except OSError:
