@@ -128,3 +128,21 @@ def test_config_merge_roundtrip(over):
         for k, v in kv.items():
             got = cfg[section][k]
             assert got == v or (got != got and v != v)  # NaN-safe
+
+
+def test_droppath_plan_uniform_inclusion():
+    """argsort-of-uniforms subsets are uniform: every sample is kept with
+    probability keep/B (chi-square-ish bound over many draws)."""
+    from dinov3_amd.layers.block import DropPathPlan
+
+    torch.manual_seed(123)
+    B, keep_ratio, trials = 8, 0.5, 400
+    counts = torch.zeros(B)
+    metas = [(0, B, 1, None, None, 0)]
+    for _ in range(trials):
+        plan = DropPathPlan(metas, keep_ratio, 1, torch.device("cpu"))
+        rows, _, _ = plan.take(0)
+        counts[rows] += 1
+    expected = trials * 0.5
+    # std of a binomial(400, .5) is 10; allow 4 sigma
+    assert ((counts - expected).abs() < 40).all(), counts
