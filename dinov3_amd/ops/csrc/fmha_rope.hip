@@ -1068,9 +1068,12 @@ void launch_fmha_rope_bwd_dkv(const __hip_bfloat16* qkv, const __hip_bfloat16* d
   const bool small = N <= 64;
   // Gated 64-row super-tile variant (see bwd_dkv64_kernel); only useful when
   // there is more than one 32-row q-tile. getenv per launch is ~ns vs the
-  // 100 us kernel and keeps the flag flippable from tests.
+  // 100 us kernel and keeps the flag flippable from tests. Flip
+  // DKV64_DEFAULT to 1 in round 2 after hardware validation.
+  constexpr bool DKV64_DEFAULT = false;
   const char* dkv64 = getenv("DINOV3_FMHA_DKV64");
-  const bool use64 = !small && dkv64 && dkv64[0] == '1';
+  const bool dkv64_on = dkv64 ? (dkv64[0] == '1') : DKV64_DEFAULT;
+  const bool use64 = !small && dkv64_on;
   dim3 grid(B * H, small ? 1 : (N + 127) / 128);
   if (HD == 64) {
     size_t shmem = 2 * 64 * 72 * sizeof(__hip_bfloat16) + 128 * sizeof(float);
