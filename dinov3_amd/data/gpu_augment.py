@@ -219,7 +219,9 @@ class GpuDataAugmentationDINO:
         B, _, H, W = imgs.shape
         boxes, flips = self._sample_boxes(B, H, W, scale)
         crops = batched_rrc_flip(imgs, boxes.to(imgs.device), flips.to(imgs.device), size)
-        crops = batched_color_jitter(crops, self._sample_jitter(B))
+        jitter = {k: v.to(imgs.device) if torch.is_tensor(v) else v
+                  for k, v in self._sample_jitter(B).items()}
+        crops = batched_color_jitter(crops, jitter)
         sigma, apply = self._blur_params(B, blur_prob)
         crops = batched_gaussian_blur(crops, sigma, apply)
         if solarize_prob > 0:

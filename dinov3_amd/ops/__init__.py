@@ -25,7 +25,7 @@ from .rope import rope_apply
 from .l2norm import l2_normalize
 from .bias_act import bias_gelu
 from .fused_update import ema_update_, multi_tensor_adamw_, grad_l2_norm_sq
-from .proto_scores import dino_softmax_ce, ibot_softmax_ce, sinkhorn_knopp, sinkhorn_rowcol
+from .proto_scores import dino_softmax_ce, ibot_softmax_ce, sinkhorn_knopp
 from .ls_axpy import ls_axpy
 from .row_ops import gather_rows, scatter_add_rows
 from .flat_attention import flat_multi_fmha
@@ -43,7 +43,6 @@ __all__ = [
     "dino_softmax_ce",
     "ibot_softmax_ce",
     "sinkhorn_knopp",
-    "sinkhorn_rowcol",
     "ls_axpy",
     "gather_rows",
     "scatter_add_rows",

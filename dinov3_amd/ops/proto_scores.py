@@ -235,11 +235,6 @@ def sinkhorn_knopp(teacher_logits: torch.Tensor, teacher_temp: float,
     return Q.T.contiguous()
 
 
-def sinkhorn_rowcol(Q: torch.Tensor) -> torch.Tensor:
-    """Placeholder hook for the fused row/col-normalization kernel."""
-    return Q
-
-
 def dino_softmax_ce(student_logits: torch.Tensor, teacher_probs: torch.Tensor,
                     student_temp: float = 0.1, ignore_diagonal: bool = False) -> torch.Tensor:
     """student_logits: [S, B, K]; teacher_probs: [T, B, K].

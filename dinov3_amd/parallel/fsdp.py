@@ -166,6 +166,10 @@ class ShardedEngine:
         self._reset_pending()
 
     # ------------------------------------------------------------------
+    # Shards are disjoint, so the per-rank sums are partial and the trainer
+    # must all-reduce them to get the global squared norm.
+    needs_norm_allreduce = True
+
     @torch.no_grad()
     def grad_norm_sums(self) -> torch.Tensor:
         """Per-submodel sum of squared grads over THIS rank's shards; caller

@@ -46,6 +46,7 @@ class CosineScheduler:
 
 class linear_warmup_cosine_decay:
     def __init__(self, start, peak, end, warmup_iterations, total_iterations, cosine_iterations=None):
+        self.end = np.float64(end)
         linear = np.linspace(start, peak, warmup_iterations, endpoint=False)
         if cosine_iterations is None:
             cosine_iterations = total_iterations - warmup_iterations
@@ -61,6 +62,9 @@ class linear_warmup_cosine_decay:
         return self.schedule
 
     def __getitem__(self, it):
+        # Past the end of the schedule the value is the terminal `end`, even
+        # when a degenerate cosine span (cosine_iterations==1 with
+        # warmup≈total) leaves the last stored sample at `peak`.
         if it >= len(self.schedule):
-            return float(self.schedule[-1])
+            return float(self.end)
         return float(self.schedule[it])

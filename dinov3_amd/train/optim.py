@@ -82,10 +82,15 @@ class FusedAdamW:
         return bool(self.groups) and self.groups[0]["params"][0].is_cuda
 
     # ------------------------------------------------------------------
+    # Grads are fully replicated on every rank after GradReducer.finalize, so
+    # the per-rank sums are already global — the trainer must NOT all-reduce
+    # them (that would inflate squared norms by world_size).
+    needs_norm_allreduce = False
+
     @torch.no_grad()
     def grad_norm_sums(self) -> torch.Tensor:
         """Sum of squared grads per submodel — [n_submodels] fp32 on the
-        params' device (caller all-reduces across ranks)."""
+        params' device (already global in the replicated-grad DDP path)."""
         if self._use_planned():
             from ..ops.mt_plan import l2norm_planned
 

@@ -147,13 +147,13 @@ def build_training_engine(cfg, params_groups):
     return optimizer, reducer.finalize
 
 
-def build_multi_resolution_data_loader_from_cfg(cfg, model: SSLMetaArch, sampler_advance: int = 0):
+def build_multi_resolution_data_loader_from_cfg(cfg, model: SSLMetaArch, start_iter: int = 0):
     """Multi-resolution crop schedules: crops.global_crops_size may be a list
     of sizes (paired with local sizes + ratios) — one loader per resolution,
     combined by sampling ratio (reference train.py:718-769)."""
     g_sizes = cfg.crops.global_crops_size
     if isinstance(g_sizes, int):
-        return build_data_loader_from_cfg(cfg, model, sampler_advance)
+        return build_data_loader_from_cfg(cfg, model, start_iter)
     from ..data.loaders import CombinedDataLoader
     import copy
 
@@ -164,16 +164,18 @@ def build_multi_resolution_data_loader_from_cfg(cfg, model: SSLMetaArch, sampler
     if isinstance(ratios, (int, float)):
         ratios = [float(ratios)] * len(g_sizes)
     loaders = []
-    for gs, ls in zip(g_sizes, l_sizes):
+    for i, (gs, ls) in enumerate(zip(g_sizes, l_sizes)):
         sub = copy.deepcopy(cfg)
         sub.crops.global_crops_size = gs
         sub.crops.local_crops_size = ls
-        loaders.append(build_data_loader_from_cfg(sub, model, sampler_advance))
+        # decorrelate per-resolution streams (reference train.py:755)
+        sub.train.seed = cfg.train.seed + i + 1
+        loaders.append(build_data_loader_from_cfg(sub, model, start_iter))
     logger.info("multi-resolution loader: sizes %s ratios %s", list(g_sizes), ratios)
     return CombinedDataLoader(loaders, ratios, seed=cfg.train.seed)
 
 
-def build_data_loader_from_cfg(cfg, model: SSLMetaArch, sampler_advance: int = 0):
+def build_data_loader_from_cfg(cfg, model: SSLMetaArch, start_iter: int = 0):
     img_size = cfg.crops.global_crops_size
     patch_size = cfg.student.patch_size
     n_tokens = (img_size // patch_size) ** 2
@@ -210,14 +212,17 @@ def build_data_loader_from_cfg(cfg, model: SSLMetaArch, sampler_advance: int = 0
         transform=transform,
         target_transform=lambda _: (),
     )
+    # Resume must not replay the sequence from iteration 0: the sampler is
+    # advanced by the consumed sample count and the seed offset by start_iter
+    # (reference train.py:838-840).
     return make_data_loader(
         dataset=dataset,
         batch_size=cfg.train.batch_size_per_gpu,
         num_workers=cfg.train.num_workers,
         shuffle=True,
-        seed=cfg.train.seed,
+        seed=cfg.train.seed + start_iter + 1,
         sampler_type=SamplerType.EPOCH,
-        sampler_advance=sampler_advance,
+        sampler_advance=start_iter * cfg.train.batch_size_per_gpu,
         drop_last=True,
         collate_fn=collate_fn,
     )
@@ -257,7 +262,7 @@ def do_train(cfg, model: SSLMetaArch, resume: bool = True, max_iterations: int =
             start_iter = payload["iteration"] + 1
             logger.info("resumed at iteration %d", start_iter)
 
-    data_loader = build_multi_resolution_data_loader_from_cfg(cfg, model)
+    data_loader = build_multi_resolution_data_loader_from_cfg(cfg, model, start_iter)
     total_iterations = schedulers["total_iterations"]
     if max_iterations > 0:
         total_iterations = min(total_iterations, start_iter + max_iterations)
@@ -348,7 +353,9 @@ def do_train(cfg, model: SSLMetaArch, resume: bool = True, max_iterations: int =
         clip_scales = None
         if clip is not None and clip > 0:
             sums = optimizer.grad_norm_sums()  # [n_submodels], stays on device
-            if parallel.get_world_size() > 1:
+            # Only shard-local sums need the cross-rank reduction; in the DDP
+            # path grads are replicated and the sums are already global.
+            if parallel.get_world_size() > 1 and getattr(optimizer, "needs_norm_allreduce", True):
                 import torch.distributed as dist
 
                 dist.all_reduce(sums)
