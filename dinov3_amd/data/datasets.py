@@ -68,11 +68,12 @@ class _SyntheticSplitDataset(ExtendedVisionDataset):
     NAME = "Synthetic"
     NUM_CLASSES = 1000
 
-    def __init__(self, split: Split = Split.TRAIN, root: str = "", transform=None, target_transform=None,
-                 length: Optional[int] = None):
+    def __init__(self, split: Split = Split.TRAIN, root: str = "", extra: str = "",
+                 transform=None, target_transform=None, length: Optional[int] = None):
         super().__init__(transform, target_transform)
         self.split = split
         self.root = root
+        self.extra = extra
         default = _SPLIT_LENGTHS.get(self.NAME, {}).get(split, 10_000)
         self._length = length if length is not None else default
 
@@ -83,14 +84,73 @@ class _SyntheticSplitDataset(ExtendedVisionDataset):
         return self._length
 
 
-class ImageNet(_SyntheticSplitDataset):
+class _RealReaderMixin:
+    """Datasets with a real on-disk reader when root+extra are configured
+    (synthetic decode otherwise — the offline default). `_make_reader` returns
+    an object with get_image_data/get_target/__len__."""
+
+    _reader = None
+
+    def _attach_reader(self, length_override: Optional[int]) -> None:
+        if self.root and self.extra:
+            self._reader = self._make_reader()
+            self._length = length_override if length_override is not None else len(self._reader)
+
+    def get_image_data(self, index: int) -> Optional[bytes]:
+        return self._reader.get_image_data(index) if self._reader is not None else None
+
+    def get_target(self, index: int):
+        if self._reader is not None:
+            return self._reader.get_target(index)
+        return super().get_target(index)
+
+    def get_targets(self) -> Optional[np.ndarray]:
+        return self._reader.get_targets() if self._reader is not None else None
+
+    def decode_image(self, data: Optional[bytes], index: int) -> torch.Tensor:
+        if data is not None:
+            from .readers import decode_image_bytes
+
+            return decode_image_bytes(data)
+        return random_image()
+
+
+class ImageNet(_RealReaderMixin, _SyntheticSplitDataset):
+    """Real mode: npy entries/class-ids index over an image tree (reference
+    dinov3_jax/data/datasets/image_net.py:27-337, I/O un-bypassed here)."""
+
     NAME = "ImageNet"
     NUM_CLASSES = 1000
 
+    def __init__(self, split: Split = Split.TRAIN, root: str = "", extra: str = "",
+                 transform=None, target_transform=None, length: Optional[int] = None):
+        super().__init__(split, root, extra, transform, target_transform, length)
+        self._attach_reader(length)
 
-class ImageNet22k(_SyntheticSplitDataset):
+    def _make_reader(self):
+        from .readers import ImageNetIndexReader
+
+        return ImageNetIndexReader(self.root, self.extra, self.split.value.lower())
+
+
+class ImageNet22k(_RealReaderMixin, _SyntheticSplitDataset):
+    """Real mode: per-class tarballs + block-offset entries.npy (reference
+    image_net_22k.py, mmap cache + gzip members honored)."""
+
     NAME = "ImageNet22k"
     NUM_CLASSES = 21_841
+
+    def __init__(self, split: Split = Split.TRAIN, root: str = "", extra: str = "",
+                 transform=None, target_transform=None, length: Optional[int] = None,
+                 mmap_cache_size: int = 16):
+        super().__init__(split, root, extra, transform, target_transform, length)
+        self._mmap_cache_size = mmap_cache_size
+        self._attach_reader(length)
+
+    def _make_reader(self):
+        from .readers import ImageNet22kTarballReader
+
+        return ImageNet22kTarballReader(self.root, self.extra, self._mmap_cache_size)
 
 
 class ADE20K(_SyntheticSplitDataset):
@@ -111,9 +171,10 @@ class SyntheticDataset(_SyntheticSplitDataset):
 
     NAME = "Synthetic"
 
-    def __init__(self, split: Split = Split.TRAIN, root: str = "", transform=None, target_transform=None,
+    def __init__(self, split: Split = Split.TRAIN, root: str = "", extra: str = "",
+                 transform=None, target_transform=None,
                  length: int = 10_000, height: int = 224, width: int = 224):
-        super().__init__(split, root, transform, target_transform, length=length)
+        super().__init__(split, root, extra, transform, target_transform, length=length)
         self.height = height
         self.width = width
 

@@ -64,7 +64,8 @@ def make_dataset(*, dataset_str: str, transform: Optional[Callable] = None,
 def _make_sampler(*, dataset, type: Optional[SamplerType], shuffle: bool, seed: int, advance: int):
     sample_count = len(dataset)
     if type == SamplerType.EPOCH:
-        return EpochSampler(size=sample_count, sample_count=sample_count, shuffle=shuffle, seed=seed)
+        return EpochSampler(size=sample_count, sample_count=sample_count, shuffle=shuffle, seed=seed,
+                            advance=advance)
     if type == SamplerType.INFINITE:
         return InfiniteSampler(sample_count=sample_count, shuffle=shuffle, seed=seed, advance=advance)
     if type == SamplerType.SHARDED_INFINITE:

@@ -17,8 +17,14 @@ from .. import parallel
 
 
 class EpochSampler(torch.utils.data.Sampler):
+    """Per-epoch seeded permutation, rank-strided. `advance` skips that many
+    already-consumed samples on resume: whole epochs raise the effective epoch
+    index, the remainder is skipped inside the first epoch iterated (the
+    reference forces EPOCH sampling but raises NotImplementedError for
+    advance>0, dinov3_jax/data/loaders.py:133-134 — resume needs it real)."""
+
     def __init__(self, *, size: int, sample_count: int, shuffle: bool = False, seed: int = 0,
-                 start: Optional[int] = None, step: Optional[int] = None):
+                 start: Optional[int] = None, step: Optional[int] = None, advance: int = 0):
         self._size = size
         self._sample_count = sample_count
         self._shuffle = shuffle
@@ -26,6 +32,9 @@ class EpochSampler(torch.utils.data.Sampler):
         self._start = parallel.get_rank() if start is None else start
         self._step = parallel.get_world_size() if step is None else step
         self._epoch = 0
+        per_epoch = max(len(self), 1)
+        self._epoch_offset = advance // per_epoch
+        self._skip_first = advance % per_epoch
 
     def __len__(self) -> int:
         return (self._size - self._start + self._step - 1) // self._step
@@ -34,12 +43,16 @@ class EpochSampler(torch.utils.data.Sampler):
         count = (self._size + self._sample_count - 1) // self._sample_count
         tiled = np.tile(np.arange(self._sample_count), count)[: self._size]
         if self._shuffle:
-            rng = np.random.default_rng(self._seed + self._epoch)
+            rng = np.random.default_rng(self._seed + self._epoch + self._epoch_offset)
             tiled = rng.permutation(tiled)
         return tiled[self._start:: self._step]
 
     def __iter__(self) -> Iterator[int]:
-        yield from map(int, self._iterable())
+        it = map(int, self._iterable())
+        if self._skip_first > 0:
+            it = itertools.islice(it, self._skip_first, None)
+            self._skip_first = 0
+        yield from it
 
     def set_epoch(self, epoch: int) -> None:
         self._epoch = epoch
