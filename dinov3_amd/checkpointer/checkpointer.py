@@ -112,6 +112,11 @@ def load_checkpoint(ckpt_dir: os.PathLike, model: torch.nn.Module,
     if missing or unexpected:
         logger.warning("partial restore: missing=%d unexpected=%d", len(missing), len(unexpected))
     if optimizer is not None and "optimizer" in payload:
-        optimizer.load_state_dict(payload["optimizer"])
+        # Resharded restore: the saved engine layout / world size may differ
+        # from the live one (e.g. an 8-GPU run resuming a 1-GPU checkpoint)
+        from ..train.optim_state import load_optimizer_state
+
+        load_optimizer_state(optimizer, ckpt_dir, payload,
+                             rank=parallel.get_rank(), world=parallel.get_world_size())
     logger.info("loaded checkpoint %s (iteration %d)", ckpt_dir, payload["iteration"])
     return payload

@@ -280,6 +280,8 @@ class ShardedEngine:
             b["grad_flat"].zero_()
             b["grad_shard"].zero_()
 
+    state_format = "shards"
+
     def state_dict(self) -> dict:
         return {
             "step_count": self.step_count,
@@ -291,13 +293,17 @@ class ShardedEngine:
                     "exp_avg": b["exp_avg"],
                     "exp_avg_sq": b["exp_avg_sq"],
                     "master": b["master"],
+                    # unpadded length, for world-independent resharding
+                    "total": sum(p.numel() for p in b["group"]["params"]),
                 }
                 for b in self.buckets
             ],
         }
 
     def load_state_dict(self, state: dict) -> None:
-        assert state["world"] == self.world, "resharded restore not supported yet"
+        assert state["world"] == self.world, (
+            "world size changed — use optim_state.load_optimizer_state for a "
+            "resharded restore")
         self.step_count = state["step_count"]
         by_names = {tuple(s["names"]): s for s in state["shards"]}
         for b in self.buckets:
@@ -309,6 +315,30 @@ class ShardedEngine:
             b["exp_avg_sq"].copy_(s["exp_avg_sq"])
             if b["master"] is not None and s["master"] is not None:
                 b["master"].copy_(s["master"])
+
+    def load_canonical(self, canon: dict) -> None:
+        """Restore from the world-size-independent canonical form
+        (train/optim_state.py): slice each group's flat fp32 vectors to this
+        rank's shard of the padded bucket."""
+        self.step_count = canon["step_count"]
+        for b in self.buckets:
+            key = tuple(b["group"]["names"])
+            gs = canon["groups"].get(key)
+            if gs is None:
+                logger.warning("optimizer group %s missing from canonical state", key[:1])
+                continue
+            padded = b["flat"].numel()
+            shard_n = b["param_shard"].numel()
+            s0 = self.rank * shard_n
+            for field, dst in (("exp_avg", b["exp_avg"]), ("exp_avg_sq", b["exp_avg_sq"]),
+                               ("master", b["master"])):
+                src = gs.get(field)
+                if dst is None or src is None:
+                    continue
+                vec = torch.zeros(padded, dtype=torch.float32)
+                n = min(src.numel(), padded)
+                vec[:n] = src.reshape(-1)[:n]
+                dst.copy_(vec[s0: s0 + shard_n].to(dst.device))
 
     def remove_hooks(self) -> None:
         for h in self._hooks:

@@ -175,6 +175,8 @@ class FusedAdamW:
             torch._foreach_zero_(grads)
 
     # ------------------------------------------------------------------
+    state_format = "groups"
+
     def state_dict(self) -> dict:
         return {
             "step_count": self.step_count,
@@ -204,3 +206,24 @@ class FusedAdamW:
             if g["master"] is not None and gs["master"] is not None:
                 for dst, src in zip(g["master"], gs["master"]):
                     dst.copy_(src)
+
+    def load_canonical(self, canon: dict) -> None:
+        """Restore from the world-size-independent canonical form
+        (optim_state.py): split each group's flat fp32 vectors back into
+        per-param tensors."""
+        self.step_count = canon["step_count"]
+        for g in self.groups:
+            gs = canon["groups"].get(tuple(g["names"]))
+            if gs is None:
+                logger.warning("optimizer group %s missing from canonical state", g["names"][:1])
+                continue
+            for field, dsts in (("exp_avg", g["exp_avg"]), ("exp_avg_sq", g["exp_avg_sq"]),
+                                ("master", g["master"])):
+                src = gs.get(field)
+                if dsts is None or src is None:
+                    continue
+                off = 0
+                for dst in dsts:
+                    n = dst.numel()
+                    dst.copy_(src[off: off + n].view(dst.shape).to(dst.device))
+                    off += n

@@ -385,11 +385,12 @@ def test_fused_residual_block_gpu():
         os.environ["DINOV3_DISABLE_HIP"] = "0"
         os.environ["DINOV3_FUSED_RESIDUAL"] = "1" if fused else "0"
         torch.manual_seed(31)
-        blk = SelfAttentionBlock(dim=64, num_heads=2, qkv_bias=True, drop_path=0.5,
+        # head_dim must be 64 (the FMHA kernel's supported sizes are 64/128)
+        blk = SelfAttentionBlock(dim=128, num_heads=2, qkv_bias=True, drop_path=0.5,
                                  init_values=1e-2).to(DEV).bfloat16()
         blk.train()
         torch.manual_seed(32)
-        x = torch.randn(6, 32, 64, device=DEV).bfloat16()
+        x = torch.randn(6, 32, 128, device=DEV).bfloat16()
         flat, _, _ = cat_keep_shapes([x])
         flat = flat.clone().requires_grad_(True)
         metas = [SelfAttention._meta_for(x, None, 0)]
