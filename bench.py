@@ -136,7 +136,9 @@ def main():
         clip_scales = None
         if clip:
             sums = optimizer.grad_norm_sums()
-            if world > 1:
+            # only shard-local sums need the reduction (DDP grads are already
+            # replicated — reducing again would inflate norms by world_size)
+            if world > 1 and getattr(optimizer, "needs_norm_allreduce", True):
                 import torch.distributed as dist
 
                 dist.all_reduce(sums)
