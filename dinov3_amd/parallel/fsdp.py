@@ -69,7 +69,8 @@ class ShardedEngine:
 
     def __init__(self, param_groups: List[dict], beta1: float = 0.9, beta2: float = 0.999,
                  eps: float = 1e-8, process_group=None, align: int = 64,
-                 comm_dtype: torch.dtype = torch.float32):
+                 comm_dtype: torch.dtype = torch.float32,
+                 coalesce_below: int = 1 << 20):
         assert dist.is_available() and dist.is_initialized(), "ShardedEngine needs torch.distributed"
         self.group = process_group
         self.world = dist.get_world_size(process_group)
@@ -77,6 +78,13 @@ class ShardedEngine:
         self.beta1, self.beta2, self.eps = beta1, beta2, eps
         self.step_count = 0
         self.comm_dtype = comm_dtype
+        # Buckets below this many elements are launch-latency-bound on RCCL
+        # (ViT-L: 29 of 58 fused groups hold just 0.38M of 350M params) —
+        # their reduce-scatters are deferred to ONE coalesced group launch at
+        # finalize instead of 29 separate collectives. xGMI is point-to-point,
+        # so the big per-block buckets (8-33 MB) already saturate the links.
+        self.coalesce_below = coalesce_below
+        self._is_nccl = dist.get_backend(process_group or dist.group.WORLD) == "nccl"
         self.groups = param_groups
         self.submodels = sorted({g["submodel"] for g in param_groups})
         self._sub_idx = {s: i for i, s in enumerate(self.submodels)}
@@ -117,6 +125,7 @@ class ShardedEngine:
                 "pending": 0,
                 "rs_work": None,
                 "rs_buf": None,
+                "deferred": total < coalesce_below,
             }
             self.buckets.append(bucket)
 
@@ -141,7 +150,7 @@ class ShardedEngine:
     def _on_grad(self, p: torch.nn.Parameter) -> None:
         b = self._param_bucket[id(p)]
         b["pending"] -= 1
-        if b["pending"] == 0:
+        if b["pending"] == 0 and not (b["deferred"] and self._is_nccl):
             self._launch_rs(b)
 
     def _launch_rs(self, b: dict) -> None:
@@ -152,15 +161,36 @@ class ShardedEngine:
         b["rs_work"] = _reduce_scatter(b["grad_shard"], full, self.group)
 
     def finalize_backward(self) -> None:
-        """Wait for all grad reductions; average (sum -> mean)."""
+        """Wait for all grad reductions; average (sum -> mean). Deferred
+        (small) buckets go out as one coalesced RCCL group launch."""
+        deferred = []
         for b in self.buckets:
-            if b["pending"] > 0:  # params without grads this step
-                self._launch_rs(b)
-                b["pending"] = 0
+            if b["rs_work"] is None and b["rs_buf"] is None:
+                if b["deferred"] and self._is_nccl:
+                    deferred.append(b)
+                elif b["pending"] > 0:  # params without grads this step
+                    self._launch_rs(b)
+            b["pending"] = 0
+        cm = None
+        if deferred:
+            from torch.distributed.distributed_c10d import _coalescing_manager
+
+            device = deferred[0]["grad_shard"].device
+            with _coalescing_manager(self.group, device, async_ops=True) as cm:
+                for b in deferred:
+                    full = b["grad_flat"]
+                    if full.dtype != self.comm_dtype:
+                        full = full.to(self.comm_dtype)
+                    b["rs_buf"] = full
+                    dist.reduce_scatter_tensor(b["grad_shard"], full,
+                                               op=dist.ReduceOp.SUM, group=self.group)
         for b in self.buckets:
             if b["rs_work"] is not None:
                 b["rs_work"].wait()
                 b["rs_work"] = None
+        if cm is not None:
+            cm.wait()
+        for b in self.buckets:
             b["rs_buf"] = None
             b["grad_shard"].div_(self.world)
         self._reset_pending()
@@ -205,14 +235,23 @@ class ShardedEngine:
             self._step_planned(lr, weight_decay, last_layer_lr, clip_scales)
         else:
             self._step_eager(lr, weight_decay, last_layer_lr, clip_scales)
-        # republish updated param shards (async all-gathers, then wait)
-        works = []
-        for b in self.buckets:
-            w = _all_gather(b["flat"], b["param_shard"], self.group)
-            if w is not None:
-                works.append(w)
-        for w in works:
-            w.wait()
+        # republish updated param shards: ONE coalesced all-gather group
+        # launch on RCCL (the wait is stream-side, not a host sync)
+        if self._is_nccl:
+            from torch.distributed.distributed_c10d import _coalescing_manager
+
+            with _coalescing_manager(self.group, device, async_ops=True) as cm:
+                for b in self.buckets:
+                    dist.all_gather_into_tensor(b["flat"], b["param_shard"], group=self.group)
+            cm.wait()
+        else:
+            works = []
+            for b in self.buckets:
+                w = _all_gather(b["flat"], b["param_shard"], self.group)
+                if w is not None:
+                    works.append(w)
+            for w in works:
+                w.wait()
 
     def _clip_tensor(self, clip_scales, device) -> torch.Tensor:
         if clip_scales is None:
