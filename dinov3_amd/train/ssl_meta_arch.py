@@ -78,6 +78,7 @@ class SSLMetaArch(nn.Module):
             self.koleo_loss = KoLeoLossDistributed(
                 topk=config.dino.koleo_topk,
                 loss_group_size=config.dino.koleo_distributed_loss_group_size,
+                group_data=config.dino.koleo_distributed_loss_group_data,
             )
         else:
             assert config.dino.koleo_topk == 1
