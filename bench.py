@@ -47,6 +47,9 @@ def parse_args():
     p.add_argument("--local-crops", type=int, default=8)
     p.add_argument("--global-size", type=int, default=224)
     p.add_argument("--local-size", type=int, default=96)
+    p.add_argument("--patch-size", type=int, default=16)
+    p.add_argument("--grad-checkpointing", action="store_true",
+                   help="activation checkpointing (config #5: ViT-g sizing)")
     p.add_argument("--dtype", type=str, default="bf16")
     p.add_argument("--profile-tag", type=str, default="", help="label kernels for rocprof runs")
     return p.parse_args()
@@ -57,6 +60,9 @@ def build_cfg(args):
 
     cfg = get_default_config()
     cfg.student.arch = args.arch
+    cfg.student.patch_size = args.patch_size
+    if args.grad_checkpointing:
+        cfg.train.checkpointing = True
     cfg.student.drop_path_rate = 0.3
     cfg.student.layerscale = 1.0e-05
     cfg.train.batch_size_per_gpu = args.batch_size
@@ -173,6 +179,10 @@ def main():
     images = args.batch_size * world * args.steps
     img_per_sec = images / elapsed
     ms_per_step = elapsed / args.steps * 1000.0
+    if use_gpu:
+        peak_gb = torch.cuda.max_memory_allocated(device) / 2**30
+        print(f"[bench] peak memory allocated: {peak_gb:.1f} GiB "
+              f"(rank {rank}, 288 GiB HBM3E)", file=sys.stderr)
 
     if rank == 0:
         parallelism = ("fsdp" if (world > 1 and cfg.compute_precision.sharding_strategy

@@ -735,7 +735,8 @@ torch::Tensor patch_embed_fwd(torch::Tensor x, torch::Tensor w, torch::Tensor bi
   TORCH_CHECK(x.dim() == 4 && x.scalar_type() == at::ScalarType::BFloat16);
   const int B = x.size(0), C = x.size(1), H = x.size(2), W = x.size(3);
   const int D = w.size(0);
-  TORCH_CHECK(patch == 16 && C == 3, "patch_embed_fwd kernel supports P=16, C=3");
+  TORCH_CHECK(C == 3, "patch_embed_fwd kernel supports C=3 (any patch size)");
+  TORCH_CHECK(H % patch == 0 && W % patch == 0, "H/W must be multiples of patch");
   const long rows = (long)B * (H / patch) * (W / patch);
   auto out = torch::empty({(long)B, rows / B, (long)D}, x.options());
   launch_patch_embed_fwd((const __hip_bfloat16*)x.data_ptr(),

@@ -123,6 +123,113 @@ __global__ __launch_bounds__(256) void fwd_kernel(
   }
 }
 
+// Generic-P variant (ViT-g/14 and high-res adapt configs): the 16-element
+// k-runs are no longer dx-aligned pixel runs, so the A staging gathers
+// per-element with a k < K bound (K zero-padded up to the 32-wide step).
+// Same 128x128 MFMA tile structure as the P=16 kernel.
+template <int C>
+__global__ __launch_bounds__(256) void fwd_kernel_anyP(
+    const __hip_bfloat16* __restrict__ x, const __hip_bfloat16* __restrict__ w,
+    const __hip_bfloat16* __restrict__ bias, __hip_bfloat16* __restrict__ out,
+    int Bimg, int H, int W, int P, int D) {
+  const int K = C * P * P;
+  constexpr int BK = 32;
+  constexpr int A_STRIDE = BK + 8;
+  const int wp = W / P;
+  const long rows = (long)Bimg * (H / P) * wp;
+
+  const int row0 = blockIdx.x * 128;
+  const int col0 = blockIdx.y * 128;
+  const int wave = threadIdx.x / 64;
+  const int lane = threadIdx.x & 63;
+  const int hhalf = lane >> 5;
+  const int l31 = lane & 31;
+
+  extern __shared__ __attribute__((aligned(16))) char smem_raw[];
+  __hip_bfloat16* a_lds = reinterpret_cast<__hip_bfloat16*>(smem_raw);
+  __hip_bfloat16* b_lds = a_lds + 128 * A_STRIDE;
+
+  f32x16_pe acc[4];
+#pragma unroll
+  for (int t = 0; t < 4; ++t) acc[t] = {};
+
+  const int s_row = threadIdx.x / 2;
+  const int s_half = (threadIdx.x & 1) * 16;
+
+  // per-row patch coordinates hoisted out of the K loop
+  const long n = (long)row0 + s_row;
+  const bool n_ok = n < rows;
+  int img = 0, ph = 0, pw = 0;
+  if (n_ok) {
+    img = (int)(n / ((long)(H / P) * wp));
+    const int pidx = (int)(n % ((long)(H / P) * wp));
+    ph = pidx / wp;
+    pw = pidx % wp;
+  }
+
+  const int Kpad = (K + BK - 1) / BK * BK;
+  for (int k0 = 0; k0 < Kpad; k0 += BK) {
+    __syncthreads();
+    {
+      __hip_bfloat16 a_stage[16];
+#pragma unroll
+      for (int e = 0; e < 16; ++e) {
+        const int k = k0 + s_half + e;
+        __hip_bfloat16 v = __hip_bfloat16(0.f);
+        if (n_ok && k < K) {
+          const int c = k / (P * P);
+          const int dy = (k / P) % P;
+          const int dx = k % P;
+          v = x[(((long)img * C + c) * H + (ph * P + dy)) * W + pw * P + dx];
+        }
+        a_stage[e] = v;
+      }
+      *reinterpret_cast<bf16x8_pe*>(&a_lds[s_row * A_STRIDE + s_half]) =
+          *reinterpret_cast<bf16x8_pe*>(&a_stage[0]);
+      *reinterpret_cast<bf16x8_pe*>(&a_lds[s_row * A_STRIDE + s_half + 8]) =
+          *reinterpret_cast<bf16x8_pe*>(&a_stage[8]);
+
+      const int d = col0 + s_row;
+      __hip_bfloat16 w_stage[16];
+#pragma unroll
+      for (int e = 0; e < 16; ++e) {
+        const int k = k0 + s_half + e;
+        w_stage[e] = (d < D && k < K) ? w[(long)d * K + k] : __hip_bfloat16(0.f);
+      }
+      *reinterpret_cast<bf16x8_pe*>(&b_lds[s_row * A_STRIDE + s_half]) =
+          *reinterpret_cast<bf16x8_pe*>(&w_stage[0]);
+      *reinterpret_cast<bf16x8_pe*>(&b_lds[s_row * A_STRIDE + s_half + 8]) =
+          *reinterpret_cast<bf16x8_pe*>(&w_stage[8]);
+    }
+    __syncthreads();
+
+#pragma unroll
+    for (int ks = 0; ks < 2; ++ks) {
+      bf16x8_pe af = load8(&a_lds[(wave * 32 + l31) * A_STRIDE + ks * 16 + hhalf * 8]);
+#pragma unroll
+      for (int t = 0; t < 4; ++t) {
+        bf16x8_pe bf = load8(&b_lds[(t * 32 + l31) * A_STRIDE + ks * 16 + hhalf * 8]);
+        acc[t] = PE_MFMA(af, bf, acc[t]);
+      }
+    }
+  }
+
+  {
+#pragma unroll
+    for (int t = 0; t < 4; ++t) {
+#pragma unroll
+      for (int r = 0; r < 16; ++r) {
+        const long rn = (long)row0 + wave * 32 + c_row(r, hhalf);
+        const int d = col0 + t * 32 + l31;
+        if (rn < rows && d < D) {
+          float v = acc[t][r] + bf16_to_f32(*(const short*)(bias + d));
+          *reinterpret_cast<short*>(out + rn * (long)D + d) = f32_to_bf16(v);
+        }
+      }
+    }
+  }
+}
+
 }  // namespace patch_embed
 
 void launch_patch_embed_fwd(const __hip_bfloat16* x, const __hip_bfloat16* w,
@@ -131,10 +238,13 @@ void launch_patch_embed_fwd(const __hip_bfloat16* x, const __hip_bfloat16* w,
   const long rows = (long)Bimg * (H / P) * (W / P);
   dim3 grid((rows + 127) / 128, (D + 127) / 128);
   size_t shmem = 2 * 128 * (32 + 8) * sizeof(__hip_bfloat16);
-  // P must be 16 (k-slices = dx-aligned contiguous pixel runs); other patch
-  // sizes take the patchify+hipBLASLt path in ops/patch_embed_op.py.
   if (P == 16 && C == 3) {
+    // fast path: 16-wide k-slices are dx-aligned contiguous pixel runs
     hipLaunchKernelGGL(HIP_KERNEL_NAME(patch_embed::fwd_kernel<16, 3>), grid, dim3(256),
                        shmem, stream, x, w, bias, out, Bimg, H, W, D);
+  } else if (C == 3) {
+    // generic patch size (14 for ViT-g/14, 8, ...): gather staging
+    hipLaunchKernelGGL(HIP_KERNEL_NAME(patch_embed::fwd_kernel_anyP<3>), grid, dim3(256),
+                       shmem, stream, x, w, bias, out, Bimg, H, W, P, D);
   }
 }

@@ -37,7 +37,9 @@ def patch_embed_gemm(x: torch.Tensor, weight: torch.Tensor, bias: torch.Tensor,
     """x [B,C,H,W] -> [B, N, D]."""
     from . import use_hip
 
-    if (use_hip(x) and x.dtype == torch.bfloat16 and patch == 16 and x.shape[1] == 3
+    # any patch size with C=3: P=16 takes the vector-staged kernel, others
+    # (14 for ViT-g/14, 8, ...) the gather-staged generic variant
+    if (use_hip(x) and x.dtype == torch.bfloat16 and x.shape[1] == 3
             and not x.requires_grad):
         return _PatchEmbedFn.apply(x.contiguous(), weight.contiguous(),
                                    bias.contiguous(), patch, patchify)

@@ -101,6 +101,41 @@ def main():
         timeit(sdpa, it),
         B * N * 3 * Hh * hd * 2 * 2)
 
+    # FMHA hd-128 at the high-res-adapt shape: ViT-7b/16 @768px -> N=2305
+    # (dinov3_vit7b16_high_res_adapt.yaml:162-186), 32 heads x d128
+    B2, N2, Hh2, hd2 = 4, 2305, 32, 128
+    qkv2 = torch.randn(B2 * N2, 3 * Hh2 * hd2, device=dev).bfloat16()
+    Pn2 = N2 - 1
+    a2 = torch.rand(Pn2, hd2 // 2, device=dev)
+    a2 = torch.cat([a2, a2], dim=-1)
+    sin2, cos2 = a2.sin().contiguous(), a2.cos().contiguous()
+    metas2 = [(0, B2, N2, sin2, cos2, 1)]
+    q2 = qkv2.view(B2, N2, 3, Hh2, hd2)[:, :, 0].permute(0, 2, 1, 3)
+
+    def sdpa2():
+        qq = q2.contiguous()
+        return torch.nn.functional.scaled_dot_product_attention(qq, qq, qq)
+
+    row("fmha_rope_fwd [4,2305,32,128]",
+        timeit(lambda: flat_multi_fmha(qkv2, Hh2, metas2), it),
+        timeit(sdpa2, it),
+        B2 * N2 * 3 * Hh2 * hd2 * 2 * 2)
+
+    def fmha2_fwd_bwd():
+        qkv_g = qkv2.detach().requires_grad_(True)
+        out = flat_multi_fmha(qkv_g, Hh2, metas2)
+        out.float().sum().backward()
+
+    def sdpa2_fwd_bwd():
+        qq = q2.detach().contiguous().requires_grad_(True)
+        out = torch.nn.functional.scaled_dot_product_attention(qq, qq, qq)
+        out.float().sum().backward()
+
+    row("fmha_rope_f+b [4,2305,32,128]",
+        timeit(fmha2_fwd_bwd, max(it // 4, 3)),
+        timeit(sdpa2_fwd_bwd, max(it // 4, 3)),
+        B2 * N2 * 3 * Hh2 * hd2 * 2 * 6)
+
     # fused sinkhorn over K=65536 (teacher shapes: M=128 cls rows)
     from dinov3_amd.ops.proto_scores import sinkhorn_knopp_factored
 
