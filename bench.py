@@ -51,6 +51,12 @@ def parse_args():
     p.add_argument("--grad-checkpointing", action="store_true",
                    help="activation checkpointing (config #5: ViT-g sizing)")
     p.add_argument("--dtype", type=str, default="bf16")
+    p.add_argument("--data", type=str, default="synthetic", choices=["synthetic", "loader"],
+                   help="synthetic = pre-staged device batches (headline); loader = "
+                        "real DataLoader path with per-step augment+collate+H2D in the "
+                        "timed region")
+    p.add_argument("--num-workers", type=int, default=12,
+                   help="dataloader workers per rank for --data loader")
     p.add_argument("--profile-tag", type=str, default="", help="label kernels for rocprof runs")
     return p.parse_args()
 
@@ -131,11 +137,33 @@ def main():
     groups = model.get_params_groups()
     optimizer, finalize_backward = build_training_engine(cfg, groups)
 
-    batches = make_synthetic_batch(cfg, device, dtype)
+    if args.data == "loader":
+        # full input path in the timed region: synthetic decode -> multi-crop
+        # augment (workers) -> collate -> pinned H2D each step
+        from dinov3_amd.train.train import batch_to_device, build_data_loader_from_cfg
+
+        cfg.train.num_workers = args.num_workers
+        cfg.train.dataset_path = "Synthetic:split=TRAIN"
+        loader = build_data_loader_from_cfg(cfg, model)
+        loader_it = iter(loader)
+
+        def next_batch(i):
+            nonlocal loader_it
+            try:
+                data = next(loader_it)
+            except StopIteration:
+                loader_it = iter(loader)
+                data = next(loader_it)
+            return batch_to_device(data, device)
+    else:
+        batches = make_synthetic_batch(cfg, device, dtype)
+
+        def next_batch(i):
+            return batches[i % len(batches)]
     clip = cfg.optim.clip_grad
 
     def step(i):
-        data = batches[i % len(batches)]
+        data = next_batch(i)
         loss, _ = model(data, teacher_temp=0.07, iteration=i)
         loss.backward()
         finalize_backward()
@@ -199,7 +227,7 @@ def main():
             "scaling": "weak",
             "vs_baseline": img_per_sec / (BASELINE_IMG_PER_SEC_PER_GPU * world),
             "dtype": args.dtype if use_gpu else "fp32",
-            "data": "synthetic",
+            "data": "synthetic" if args.data == "synthetic" else "synthetic(loader+H2D timed)",
             "config": {
                 "model": args.arch,
                 "global_batch": args.batch_size * world,
