@@ -51,10 +51,12 @@ def parse_args():
     p.add_argument("--grad-checkpointing", action="store_true",
                    help="activation checkpointing (config #5: ViT-g sizing)")
     p.add_argument("--dtype", type=str, default="bf16")
-    p.add_argument("--data", type=str, default="synthetic", choices=["synthetic", "loader"],
+    p.add_argument("--data", type=str, default="synthetic",
+                   choices=["synthetic", "loader", "gpu-aug"],
                    help="synthetic = pre-staged device batches (headline); loader = "
                         "real DataLoader path with per-step augment+collate+H2D in the "
-                        "timed region")
+                        "timed region; gpu-aug = workers decode only, multi-crop "
+                        "augmentation batched on the training GPU")
     p.add_argument("--num-workers", type=int, default=12,
                    help="dataloader workers per rank for --data loader")
     p.add_argument("--profile-tag", type=str, default="", help="label kernels for rocprof runs")
@@ -137,7 +139,22 @@ def main():
     groups = model.get_params_groups()
     optimizer, finalize_backward = build_training_engine(cfg, groups)
 
-    if args.data == "loader":
+    if args.data == "gpu-aug":
+        from dinov3_amd.data.gpu_pipeline import build_gpu_augment_pipeline_from_cfg
+
+        cfg.train.num_workers = args.num_workers
+        cfg.train.dataset_path = "Synthetic:split=TRAIN"
+        pipeline = build_gpu_augment_pipeline_from_cfg(cfg, device, dtype)
+        pipe_it = iter(pipeline)
+
+        def next_batch(i):
+            nonlocal pipe_it
+            try:
+                return next(pipe_it)
+            except StopIteration:
+                pipe_it = iter(pipeline)
+                return next(pipe_it)
+    elif args.data == "loader":
         # full input path in the timed region: synthetic decode -> multi-crop
         # augment (workers) -> collate -> pinned H2D each step
         from dinov3_amd.train.train import batch_to_device, build_data_loader_from_cfg
@@ -227,7 +244,9 @@ def main():
             "scaling": "weak",
             "vs_baseline": img_per_sec / (BASELINE_IMG_PER_SEC_PER_GPU * world),
             "dtype": args.dtype if use_gpu else "fp32",
-            "data": "synthetic" if args.data == "synthetic" else "synthetic(loader+H2D timed)",
+            "data": {"synthetic": "synthetic",
+                     "loader": "synthetic(loader+H2D timed)",
+                     "gpu-aug": "synthetic(gpu-augment timed)"}[args.data],
             "config": {
                 "model": args.arch,
                 "global_batch": args.batch_size * world,

@@ -14,6 +14,50 @@ from typing import Optional
 import torch
 
 
+def build_mask_batch(
+    B: int,
+    n_tokens: int,
+    mask_generator,
+    mask_ratio_tuple,
+    mask_probability: float,
+    random_circular_shift: bool = False,
+) -> dict:
+    """iBOT mask construction for a crop-major batch of B global-crop rows:
+    linspace'd mask ratios over mask_probability*B rows, shuffled, with the
+    derived mask_indices_list / masks_weight / upperbound (reference
+    collate.py:46-80). Host-side (the block-mask generator is sequential)."""
+    N = n_tokens
+    n_samples_masked = int(B * mask_probability)
+    probs = torch.linspace(*mask_ratio_tuple, n_samples_masked + 1)
+    upperbound = 0
+    masks_list = []
+    for i in range(n_samples_masked):
+        prob_max = probs[i + 1]
+        mask = torch.from_numpy(mask_generator(int(N * prob_max)))
+        if random_circular_shift:
+            shift_x = random.randint(0, mask.shape[0] - 1)
+            shift_y = random.randint(0, mask.shape[1] - 1)
+            mask = torch.roll(mask, (shift_x, shift_y), (0, 1))
+        masks_list.append(mask)
+        upperbound += int(N * prob_max)
+    for _ in range(n_samples_masked, B):
+        masks_list.append(torch.from_numpy(mask_generator(0)))
+    random.shuffle(masks_list)
+
+    collated_masks = torch.stack(masks_list).flatten(1)
+    mask_indices_list = collated_masks.flatten().nonzero().flatten()
+    masks_weight = (
+        (1 / collated_masks.sum(-1).clamp(min=1.0)).unsqueeze(-1).expand_as(collated_masks)[collated_masks]
+    )
+    return {
+        "collated_masks": collated_masks,
+        "mask_indices_list": mask_indices_list,
+        "masks_weight": masks_weight,
+        "upperbound": upperbound,
+        "n_masked_patches": torch.full((1,), fill_value=mask_indices_list.shape[0], dtype=torch.long),
+    }
+
+
 def collate_data_and_cast(
     samples_list,
     mask_ratio_tuple,
@@ -42,38 +86,15 @@ def collate_data_and_cast(
         )
 
     B = n_global_crops * local_batch_size if local_batch_size is not None else len(collated_global_crops)
-    N = n_tokens
-    n_samples_masked = int(B * mask_probability)
-    probs = torch.linspace(*mask_ratio_tuple, n_samples_masked + 1)
-    upperbound = 0
-    masks_list = []
-    for i in range(n_samples_masked):
-        prob_max = probs[i + 1]
-        mask = torch.from_numpy(mask_generator(int(N * prob_max)))
-        if random_circular_shift:
-            shift_x = random.randint(0, mask.shape[0] - 1)
-            shift_y = random.randint(0, mask.shape[1] - 1)
-            mask = torch.roll(mask, (shift_x, shift_y), (0, 1))
-        masks_list.append(mask)
-        upperbound += int(N * prob_max)
-    for _ in range(n_samples_masked, B):
-        masks_list.append(torch.from_numpy(mask_generator(0)))
-    random.shuffle(masks_list)
-
-    collated_masks = torch.stack(masks_list).flatten(1)
-    mask_indices_list = collated_masks.flatten().nonzero().flatten()
-    masks_weight = (
-        (1 / collated_masks.sum(-1).clamp(min=1.0)).unsqueeze(-1).expand_as(collated_masks)[collated_masks]
+    mask_batch = build_mask_batch(
+        B, n_tokens, mask_generator, mask_ratio_tuple, mask_probability,
+        random_circular_shift=random_circular_shift,
     )
 
     out = {
         "collated_global_crops": collated_global_crops.to(dtype),
         "collated_local_crops": collated_local_crops.to(dtype),
-        "collated_masks": collated_masks,
-        "mask_indices_list": mask_indices_list,
-        "masks_weight": masks_weight,
-        "upperbound": upperbound,
-        "n_masked_patches": torch.full((1,), fill_value=mask_indices_list.shape[0], dtype=torch.long),
+        **mask_batch,
     }
     if collated_gram_teacher_crops is not None:
         out["collated_gram_teacher_crops"] = collated_gram_teacher_crops.to(dtype)

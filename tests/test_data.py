@@ -192,3 +192,46 @@ class TestGpuAugment:
         assert torch.isfinite(out["global_crops"]).all()
         # normalized output: roughly zero-centred
         assert out["global_crops"].mean().abs() < 2.0
+
+
+def test_gpu_augment_pipeline_contract_cpu():
+    """GpuAugmentPipeline emits collate_data_and_cast-shaped batches (CPU run
+    of the same code path the GPU uses)."""
+    import types
+
+    import torch
+
+    from dinov3_amd.configs import get_default_config
+    from dinov3_amd.data.gpu_pipeline import build_gpu_augment_pipeline_from_cfg
+
+    cfg = get_default_config()
+    cfg.train.batch_size_per_gpu = 3
+    cfg.train.num_workers = 0
+    cfg.train.dataset_path = "Synthetic:split=TRAIN:length=16"
+    cfg.crops.global_crops_size = 64
+    cfg.crops.local_crops_size = 32
+    cfg.crops.local_crops_number = 4
+    pipe = build_gpu_augment_pipeline_from_cfg(cfg, torch.device("cpu"), torch.float32,
+                                               canonical_size=96)
+    batch = next(iter(pipe))
+    B = 3
+    assert batch["collated_global_crops"].shape == (2 * B, 3, 64, 64)
+    assert batch["collated_local_crops"].shape == (4 * B, 3, 32, 32)
+    n_tokens = (64 // cfg.student.patch_size) ** 2
+    assert batch["collated_masks"].shape == (2 * B, n_tokens)
+    assert batch["mask_indices_list"].numel() == int(batch["n_masked_patches"][0])
+    assert batch["masks_weight"].numel() == batch["mask_indices_list"].numel()
+    # normalized outputs: roughly zero-mean after ImageNet normalization
+    assert batch["collated_global_crops"].abs().mean() < 5.0
+
+
+def test_canonical_decode_sizes():
+    import torch
+
+    from dinov3_amd.data.gpu_pipeline import CanonicalDecode
+
+    dec = CanonicalDecode(64)
+    for h, w in ((100, 80), (64, 64), (50, 200)):
+        out = dec(torch.rand(3, h, w))
+        assert out.shape == (3, 64, 64)
+        assert out.dtype == torch.uint8
