@@ -40,8 +40,11 @@ class SmoothedValue:
 
         if not (dist.is_available() and dist.is_initialized()):
             return
+        from .. import parallel
+
+        # meters are per-student under multi-distillation: sync the subgroup
         t = torch.tensor([self.count, self.total], dtype=torch.float64)
-        dist.all_reduce(t)
+        dist.all_reduce(t, group=parallel.subgroup())
         self.count = int(t[0].item())
         self.total = t[1].item()
 

@@ -46,8 +46,10 @@ class DINOLoss(nn.Module):
 
     @torch.no_grad()
     def apply_center_update(self, teacher_output: torch.Tensor) -> None:
+        from .. import parallel
+
         local_center = teacher_output.float().mean(dim=0, keepdim=True)
         if dist.is_available() and dist.is_initialized():
-            dist.all_reduce(local_center)
-            local_center /= dist.get_world_size()
+            dist.all_reduce(local_center, group=parallel.subgroup())
+            local_center /= parallel.subgroup_size()
         self.center.mul_(self.center_momentum).add_(local_center * (1 - self.center_momentum))

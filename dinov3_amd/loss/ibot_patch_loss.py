@@ -44,8 +44,10 @@ class iBOTPatchLoss(nn.Module):
         if B.ndim > 0:
             B = B.sum()
         if dist.is_available() and dist.is_initialized():
+            from .. import parallel
+
             B = B.to(teacher_output.device)
-            dist.all_reduce(B)
+            dist.all_reduce(B, group=parallel.subgroup())
         return sinkhorn_knopp(teacher_output, teacher_temp, total_columns=B, n_iterations=n_iterations)
 
     def forward(self, student_patch_tokens: torch.Tensor, teacher_patch_tokens: torch.Tensor,
@@ -84,8 +86,10 @@ class iBOTPatchLoss(nn.Module):
 
     @torch.no_grad()
     def apply_center_update(self, teacher_output: torch.Tensor) -> None:
+        from .. import parallel
+
         local_center = teacher_output.float().mean(dim=0, keepdim=True)
         if dist.is_available() and dist.is_initialized():
-            dist.all_reduce(local_center)
-            local_center /= dist.get_world_size()
+            dist.all_reduce(local_center, group=parallel.subgroup())
+            local_center /= parallel.subgroup_size()
         self.center.mul_(self.center_momentum).add_(local_center * (1 - self.center_momentum))

@@ -87,11 +87,19 @@ class KoLeoLossDistributed(nn.Module):
         return group, group_rank, ranks_per_group
 
     def forward(self, student_output: torch.Tensor, eps: float = 1e-8) -> torch.Tensor:
+        from .. import parallel
+
         x = student_output.float()
         x = x / (x.norm(p=2, dim=-1, keepdim=True) + eps)
         local_B = x.shape[0]
-        if dist.is_available() and dist.is_initialized() and dist.get_world_size() > 1:
-            group, rank, gw = self._gather_group(local_B)
+        if dist.is_available() and dist.is_initialized() and parallel.subgroup_size() > 1:
+            if parallel.subgroup() is not None:
+                # multi-distillation: gather over the student's subgroup
+                # (loss_group_size sub-partitioning is not composed with it)
+                group, rank, gw = (parallel.subgroup(), parallel.subgroup_rank(),
+                                   parallel.subgroup_size())
+            else:
+                group, rank, gw = self._gather_group(local_B)
             gathered = [torch.empty_like(x) for _ in range(gw)]
             dist.all_gather(gathered, x, group=group)
             gathered[rank] = x  # keep autograd path through the local shard

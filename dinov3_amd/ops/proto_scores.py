@@ -20,7 +20,16 @@ import torch.nn.functional as F
 
 
 def _world() -> int:
-    return dist.get_world_size() if (dist.is_available() and dist.is_initialized()) else 1
+    # training-collective scope: the multi-distillation subgroup when set
+    from .. import parallel
+
+    return parallel.subgroup_size()
+
+
+def _group():
+    from .. import parallel
+
+    return parallel.subgroup()
 
 
 class FactoredProbs:
@@ -75,7 +84,7 @@ def sinkhorn_knopp_factored(teacher_logits: torch.Tensor, teacher_temp: float,
         A = ops.sinkhorn_fact_colsum(x, u if it > 0 else torch.empty(0, device=x.device),
                                      teacher_temp)
         if world > 1:
-            dist.all_reduce(A)
+            dist.all_reduce(A, group=_group())
         v = torch.reciprocal(A * K)
         u = ops.sinkhorn_fact_rowsum(x, v, teacher_temp)
     return FactoredProbs(x, u, v, teacher_temp)
@@ -180,7 +189,7 @@ def _sinkhorn_knopp_hip(teacher_logits: torch.Tensor, teacher_temp: float,
     M, K = teacher_logits.shape
     Q, total = ops.sinkhorn_exp(teacher_logits.contiguous(), teacher_temp)
     if world > 1:
-        dist.all_reduce(total)
+        dist.all_reduce(total, group=_group())
     if total_columns is None:
         B = torch.full((), float(M * world), device=Q.device)
     else:
@@ -189,7 +198,7 @@ def _sinkhorn_knopp_hip(teacher_logits: torch.Tensor, teacher_temp: float,
     for it in range(n_iterations):
         col = ops.sinkhorn_colsum(Q, total if it == 0 else one)
         if world > 1:
-            dist.all_reduce(col)
+            dist.all_reduce(col, group=_group())
         ops.sinkhorn_div_row(Q, col, float(K), B, it == n_iterations - 1)
     return Q
 
@@ -219,13 +228,13 @@ def sinkhorn_knopp(teacher_logits: torch.Tensor, teacher_temp: float,
 
     sum_Q = Q.sum()
     if world > 1:
-        dist.all_reduce(sum_Q)
+        dist.all_reduce(sum_Q, group=_group())
     Q /= sum_Q
 
     for _ in range(n_iterations):
         sum_rows = Q.sum(dim=1, keepdim=True)  # [K, 1]
         if world > 1:
-            dist.all_reduce(sum_rows)
+            dist.all_reduce(sum_rows, group=_group())
         Q /= sum_rows
         Q /= K
         Q /= Q.sum(dim=0, keepdim=True)

@@ -71,6 +71,36 @@ def barrier() -> None:
         dist.barrier()
 
 
+# ---------------------------------------------------------------------------
+# Training-collective subgroup (multi-distillation): when set, every
+# loss/gradient collective (sinkhorn psums, grad reduce-scatter/all-reduce,
+# koleo gather, grad-norm sums) runs over THIS group instead of the world.
+# Data sharding (sampler rank striding) and logging stay world-scoped.
+_SUBGROUP = None
+
+
+def set_subgroup(pg) -> None:
+    global _SUBGROUP
+    _SUBGROUP = pg
+
+
+def subgroup():
+    """Process group for training collectives (None = the world group)."""
+    return _SUBGROUP
+
+
+def subgroup_size() -> int:
+    if not is_enabled():
+        return 1
+    return dist.get_world_size(_SUBGROUP) if _SUBGROUP is not None else dist.get_world_size()
+
+
+def subgroup_rank() -> int:
+    if not is_enabled():
+        return 0
+    return dist.get_rank(_SUBGROUP) if _SUBGROUP is not None else dist.get_rank()
+
+
 def destroy() -> None:
     global _INITIALIZED_HERE
     if _INITIALIZED_HERE and is_enabled():

@@ -25,8 +25,11 @@ logger = logging.getLogger("dinov3")
 class GradReducer:
     def __init__(self, params: List[torch.nn.Parameter], bucket_cap_mb: float = 32.0,
                  reduce_dtype: Optional[torch.dtype] = None):
+        from . import subgroup, subgroup_size
+
         self.params = [p for p in params if p.requires_grad]
-        self.world = dist.get_world_size() if (dist.is_available() and dist.is_initialized()) else 1
+        self.group = subgroup()
+        self.world = subgroup_size()
         self.reduce_dtype = reduce_dtype
         self._works: List[dist.Work] = []
         self._hooks = []
@@ -56,7 +59,7 @@ class GradReducer:
         flat = torch.cat([g.reshape(-1) for g in grads])
         if self.reduce_dtype is not None and flat.dtype != self.reduce_dtype:
             flat = flat.to(self.reduce_dtype)
-        work = dist.all_reduce(flat, op=dist.ReduceOp.SUM, async_op=True)
+        work = dist.all_reduce(flat, op=dist.ReduceOp.SUM, group=self.group, async_op=True)
         self._works.append((work, flat, grads))
 
     def finalize(self) -> None:
@@ -82,10 +85,12 @@ class GradReducer:
 
 def all_reduce_scalar_sums(values: Dict[str, torch.Tensor]) -> Dict[str, torch.Tensor]:
     """All-reduce a dict of scalar tensors (sum). Used for grad-norm sq."""
-    if not (dist.is_available() and dist.is_initialized()) or dist.get_world_size() == 1:
+    from . import subgroup, subgroup_size
+
+    if not (dist.is_available() and dist.is_initialized()) or subgroup_size() == 1:
         return values
     keys = sorted(values.keys())
     device = next(iter(values.values())).device if values else torch.device("cpu")
     buf = torch.stack([values[k].to(device).float().reshape(()) for k in keys])
-    dist.all_reduce(buf)
+    dist.all_reduce(buf, group=subgroup())
     return {k: buf[i] for i, k in enumerate(keys)}

@@ -132,13 +132,16 @@ def build_training_engine(cfg, params_groups):
 
     Returns (optimizer-like, finalize_backward_fn).
     """
-    world = parallel.get_world_size()
+    # grads reduce over the training subgroup (the world unless
+    # multi-distillation set a per-student subgroup)
+    world = parallel.subgroup_size()
     strategy = cfg.compute_precision.sharding_strategy
     if world > 1 and strategy in ("SHARD_GRAD_OP", "FULL_SHARD"):
         from ..parallel.fsdp import ShardedEngine
 
         engine = ShardedEngine(params_groups, beta1=cfg.optim.adamw_beta1,
-                               beta2=cfg.optim.adamw_beta2)
+                               beta2=cfg.optim.adamw_beta2,
+                               process_group=parallel.subgroup())
         return engine, engine.finalize_backward
     optimizer = build_optimizer(cfg, params_groups)
     student_params = [p for g in params_groups for p in g["params"]]
@@ -355,10 +358,10 @@ def do_train(cfg, model: SSLMetaArch, resume: bool = True, max_iterations: int =
             sums = optimizer.grad_norm_sums()  # [n_submodels], stays on device
             # Only shard-local sums need the cross-rank reduction; in the DDP
             # path grads are replicated and the sums are already global.
-            if parallel.get_world_size() > 1 and getattr(optimizer, "needs_norm_allreduce", True):
+            if parallel.subgroup_size() > 1 and getattr(optimizer, "needs_norm_allreduce", True):
                 import torch.distributed as dist
 
-                dist.all_reduce(sums)
+                dist.all_reduce(sums, group=parallel.subgroup())
             clip_scales = optimizer.clip_factors(sums, clip)
             if it % 10 == 0:  # avoid a host sync every step
                 for name, s in zip(optimizer.submodels, sums.tolist()):
@@ -498,9 +501,15 @@ def main(argv=None):
     if meta_arch_cls is None:
         raise ValueError(f"unknown MODEL.META_ARCHITECTURE {cfg.MODEL.META_ARCHITECTURE}")
     if meta_arch_cls is MultiDistillationMetaArch:
-        MultiDistillationMetaArch(cfg)  # validates the subgroup layout
-        raise NotImplementedError("multi-distillation training loop is a stub (reference parity)")
-    model = meta_arch_cls(cfg)
+        # each rank builds only its own student; collectives are scoped to
+        # the student's subgroup inside the wrapper; the per-student merged
+        # config drives the optimizer/schedules/data for this rank
+        model = MultiDistillationMetaArch(cfg)
+        rank_cfg = model.rank_config
+        rank_cfg.train.output_dir = cfg.train.output_dir
+        cfg = rank_cfg
+    else:
+        model = meta_arch_cls(cfg)
     if args.eval_only:
         # load weights for evaluation: latest checkpoint unless MODEL.WEIGHTS
         # names one explicitly (reference train.py:302-309)

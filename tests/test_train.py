@@ -234,16 +234,39 @@ def test_main_eval_only_with_weights(tmp_path):
     assert results and "knn_top1" in results
 
 
-def test_main_multidist_dispatch():
-    """The multi-distillation recipe dispatches to MultiDistillationMetaArch,
-    validates the subgroup layout, and reports the unimplemented loop
-    explicitly (the reference's is an empty stub)."""
-    import pytest as _pytest
-
+def test_main_multidist_dispatch(tmp_path):
+    """The multi-distillation recipe dispatches to MultiDistillationMetaArch
+    and TRAINS this rank's student (single process -> first student) — the
+    reference's is an empty stub. Tiny student/teacher configs keep it
+    CPU-fast; --max-iterations bounds the loop."""
+    student = tmp_path / "stu.yaml"
+    student.write_text(
+        "dino: {head_n_prototypes: 64, head_bottleneck_dim: 32, head_hidden_dim: 64}\n"
+        "ibot: {head_n_prototypes: 64, head_bottleneck_dim: 32, head_hidden_dim: 64}\n"
+        "student: {arch: vit_small, patch_size: 16, drop_path_rate: 0.0, ffn_ratio: 1.0}\n"
+        "compute_precision: {param_dtype: fp32}\n"
+        "train: {batch_size_per_gpu: 2, dataset_path: 'Synthetic:split=TRAIN:length=16',\n"
+        "        num_workers: 0, OFFICIAL_EPOCH_LENGTH: 2}\n"
+        "optim: {epochs: 1}\n"
+        "crops: {local_crops_number: 2, global_crops_size: 112, local_crops_size: 48}\n"
+        "evaluation: {eval_period_iterations: 0}\n"
+        "checkpointing: {period: 0}\n"
+    )
+    recipe = tmp_path / "recipe.yaml"
+    recipe.write_text(
+        "MODEL: {META_ARCHITECTURE: MultiDistillationMetaArch}\n"
+        "multidistillation:\n"
+        "  enabled: true\n"
+        "  global_batch_size: 2\n"
+        "  students:\n"
+        f"  - {{name: only, config_path: {student}, ranks_range: [0, 1]}}\n"
+        f"distillation: {{enabled: true, full_cfg_path: {student}, checkpoint_path: ignore}}\n"
+    )
     from dinov3_amd.train.train import main
 
-    with _pytest.raises((NotImplementedError, AssertionError)):
-        main([
-            "--config-file", "dinov3_amd/configs/train/multi_distillation_test.yaml",
-            "--output-dir", "",
-        ])
+    result = main([
+        "--config-file", str(recipe),
+        "--output-dir", str(tmp_path / "out"),
+        "--no-resume", "--max-iterations", "2",
+    ])
+    assert "total_loss" in result

@@ -14,11 +14,11 @@ echo "data_gpu=$?" >> "$OUT/summary.txt"
 
 # 1. kernel-level diff: base vs DINOV3_FUSED_RESIDUAL=1 (trace/stats only)
 cd /tmp
-timeout 300 rocprofv3 --stats -d "$GRAFT_REPO_ROOT/$OUT/prof_base" -o base --output-format csv -- \
+timeout 300 rocprofv3 --kernel-trace --stats -d "$GRAFT_REPO_ROOT/$OUT/prof_base" -o base --output-format csv -- \
     python "$GRAFT_REPO_ROOT/bench.py" --steps 3 --warmup 1 \
     > "$GRAFT_REPO_ROOT/$OUT/prof_base.log" 2>&1
 echo "prof_base=$?" >> "$GRAFT_REPO_ROOT/$OUT/summary.txt"
-DINOV3_FUSED_RESIDUAL=1 timeout 300 rocprofv3 --stats -d "$GRAFT_REPO_ROOT/$OUT/prof_fused" -o fused --output-format csv -- \
+DINOV3_FUSED_RESIDUAL=1 timeout 300 rocprofv3 --kernel-trace --stats -d "$GRAFT_REPO_ROOT/$OUT/prof_fused" -o fused --output-format csv -- \
     python "$GRAFT_REPO_ROOT/bench.py" --steps 3 --warmup 1 \
     > "$GRAFT_REPO_ROOT/$OUT/prof_fused.log" 2>&1
 echo "prof_fused=$?" >> "$GRAFT_REPO_ROOT/$OUT/summary.txt"
