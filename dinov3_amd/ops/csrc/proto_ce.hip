@@ -213,17 +213,18 @@ __global__ void sinkhorn_div_row_kernel(float* __restrict__ Q,
 // factors cancel in the final row-normalize, so no 1/sum_Q pass is needed.
 
 // A[k] = sum_m exp(x[m,k]*inv_temp) * u[m]   (u == nullptr -> 1)
-// 2D tiling: block (kw, mt) reduces a 256-row m-tile over a 2048-col window
+// 2D tiling: block (kw, mt) reduces an m_tile-row slab over a 2048-col window
 // (contiguous 4 KB per row per block: streams, unlike a per-thread column
 // walk with a 128 KB stride), then one atomicAdd per column per m-tile.
-#define SK_MT 256
+// m_tile is chosen by the launcher so kw*mt fills the 256 CUs even at the
+// DINO shape (M=128, kw=32 -> m_tile 8 gives 512 blocks, not 32).
 __global__ void sinkhorn_fact_colsum_kernel(const __hip_bfloat16* __restrict__ x,
                                             const float* __restrict__ u,
                                             float* __restrict__ A, int M, long K,
-                                            float inv_temp) {
+                                            float inv_temp, int m_tile) {
   const long k0 = (long)blockIdx.x * (blockDim.x * 8);
-  const int m0 = blockIdx.y * SK_MT;
-  const int m1 = min(m0 + SK_MT, M);
+  const int m0 = blockIdx.y * m_tile;
+  const int m1 = min(m0 + m_tile, M);
   float acc[8] = {0.f, 0.f, 0.f, 0.f, 0.f, 0.f, 0.f, 0.f};
   const long kb = k0 + threadIdx.x * 8;
   if (kb >= K) return;
@@ -246,19 +247,33 @@ __global__ void sinkhorn_fact_colsum_kernel(const __hip_bfloat16* __restrict__ x
 }
 
 // u[m] = 1 / sum_k exp(x[m,k]*inv_temp) * v[k]
+// k-split over gridDim.y so small-M launches (DINO: M=128) still fill the
+// chip; partials land in sums[m] via atomicAdd and a finalize kernel inverts.
 __global__ void sinkhorn_fact_rowsum_kernel(const __hip_bfloat16* __restrict__ x,
                                             const float* __restrict__ v,
-                                            float* __restrict__ u, long K,
+                                            float* __restrict__ sums, long K,
                                             float inv_temp) {
   __shared__ float red[16];
   const int m = blockIdx.x;
+  const int split = gridDim.y;
+  const long kchunk = (K + split - 1) / split;
+  const long ks = blockIdx.y * kchunk;
+  const long ke = min(ks + kchunk, K);
   const __hip_bfloat16* xr = x + (long)m * K;
   float acc = 0.f;
-  for (long k = threadIdx.x; k < K; k += blockDim.x) {
+  for (long k = ks + threadIdx.x; k < ke; k += blockDim.x) {
     acc += __expf(bf16_to_f32(*(const short*)(xr + k)) * inv_temp) * v[k];
   }
   acc = block_reduce_sum(acc, red);
-  if (threadIdx.x == 0) u[m] = 1.0f / acc;
+  if (threadIdx.x == 0) {
+    if (split == 1) sums[m] = 1.0f / acc;
+    else atomicAdd(sums + m, acc);
+  }
+}
+
+__global__ void sinkhorn_rowsum_invert_kernel(float* __restrict__ u, int M) {
+  const int m = blockIdx.x * blockDim.x + threadIdx.x;
+  if (m < M) u[m] = 1.0f / u[m];
 }
 
 // ---------------- CE with factored teacher ----------------
@@ -431,16 +446,29 @@ void launch_ibot_ce_bwd(const __hip_bfloat16* x, const float* t, const float* w,
 void launch_sinkhorn_fact_colsum(const __hip_bfloat16* x, const float* u, float* A,
                                  int M, long K, float inv_temp, hipStream_t stream) {
   const int kw = (int)((K + CE_BLOCK * 8 - 1) / (CE_BLOCK * 8));
-  const int mt = (M + SK_MT - 1) / SK_MT;
+  // target >=512 workgroups so the 256-CU chip is filled even at M=128
+  int tiles = 512 / (kw > 0 ? kw : 1);
+  if (tiles < 1) tiles = 1;
+  if (tiles > M) tiles = M;
+  const int m_tile = (M + tiles - 1) / tiles;
+  const int mt = (M + m_tile - 1) / m_tile;
   if (mt > 1) (void)hipMemsetAsync(A, 0, K * sizeof(float), stream);
   hipLaunchKernelGGL(sinkhorn_fact_colsum_kernel, dim3(kw, mt), dim3(CE_BLOCK), 0, stream,
-                     x, u, A, M, K, inv_temp);
+                     x, u, A, M, K, inv_temp, m_tile);
 }
 
 void launch_sinkhorn_fact_rowsum(const __hip_bfloat16* x, const float* v, float* u,
                                  int M, long K, float inv_temp, hipStream_t stream) {
-  hipLaunchKernelGGL(sinkhorn_fact_rowsum_kernel, dim3(M), dim3(CE_BLOCK), 0, stream, x,
-                     v, u, K, inv_temp);
+  int split = 512 / (M > 0 ? M : 1);
+  if (split < 1) split = 1;
+  if (split > 16) split = 16;
+  if (split > 1) (void)hipMemsetAsync(u, 0, M * sizeof(float), stream);
+  hipLaunchKernelGGL(sinkhorn_fact_rowsum_kernel, dim3(M, split), dim3(CE_BLOCK), 0,
+                     stream, x, v, u, K, inv_temp);
+  if (split > 1) {
+    hipLaunchKernelGGL(sinkhorn_rowsum_invert_kernel, dim3((M + 255) / 256), dim3(256),
+                       0, stream, u, M);
+  }
 }
 
 void launch_ibot_ce_fact_fwd(const __hip_bfloat16* x, const __hip_bfloat16* xt,
