@@ -56,7 +56,7 @@ __global__ void bias_gelu_fwd_kernel(const T* __restrict__ x, const T* __restric
   }
 }
 
-template <typename T>
+template <typename T, int RR_ILP = 4>
 __global__ void bias_gelu_bwd_kernel(const T* __restrict__ dy, const T* __restrict__ x,
                                      const T* __restrict__ bias, T* __restrict__ dx,
                                      float* __restrict__ dbias, long rows, int H) {
@@ -66,20 +66,20 @@ __global__ void bias_gelu_bwd_kernel(const T* __restrict__ dy, const T* __restri
   T bb[8];
   Vec8<T>::load(bb, bias + col8);
   float db[8] = {0.f, 0.f, 0.f, 0.f, 0.f, 0.f, 0.f, 0.f};
-  // 4 rows in flight per iteration: the fixed-column walk is a strided
+  // RR_ILP rows in flight per iteration: the fixed-column walk is a strided
   // stream, so batch the loads for ILP/prefetch depth
-  long row = blockIdx.x * 4L;
-  const long rstep = (long)gridDim.x * 4;
-  for (; row + 3 < rows; row += rstep) {
-    T xb[4][8], gb[4][8], ob[4][8];
+  long row = blockIdx.x * (long)RR_ILP;
+  const long rstep = (long)gridDim.x * RR_ILP;
+  for (; row + RR_ILP - 1 < rows; row += rstep) {
+    T xb[RR_ILP][8], gb[RR_ILP][8], ob[RR_ILP][8];
 #pragma unroll
-    for (int rr = 0; rr < 4; ++rr) {
+    for (int rr = 0; rr < RR_ILP; ++rr) {
       const long off = (row + rr) * (long)H + col8;
       Vec8<T>::load(xb[rr], x + off);
       Vec8<T>::load(gb[rr], dy + off);
     }
 #pragma unroll
-    for (int rr = 0; rr < 4; ++rr) {
+    for (int rr = 0; rr < RR_ILP; ++rr) {
 #pragma unroll
       for (int e = 0; e < 8; ++e) {
         float v = ScalarOps<T>::load(xb[rr] + e) + ScalarOps<T>::load(bb + e);
@@ -462,9 +462,19 @@ template <typename T>
 void launch_bias_gelu_bwd(const T* dy, const T* x, const T* bias, T* dx, float* dbias,
                           long rows, int H, hipStream_t stream) {
   const int col_tiles = (H / 8 + EW_BLOCK - 1) / EW_BLOCK;
+  static const int ilp8 = [] {
+    const char* e = getenv("DINOV3_BG_ILP8");
+    return e && e[0] == '1';
+  }();
+  if (ilp8) {
+    int row_grid = (int)min((rows + 7) / 8, (long)(2048 / col_tiles + 1));
+    hipLaunchKernelGGL((bias_gelu_bwd_kernel<T, 8>), dim3(row_grid, col_tiles),
+                       dim3(EW_BLOCK), 0, stream, dy, x, bias, dx, dbias, rows, H);
+    return;
+  }
   int row_grid = (int)min((rows + 3) / 4, (long)(2048 / col_tiles + 1));
-  hipLaunchKernelGGL((bias_gelu_bwd_kernel<T>), dim3(row_grid, col_tiles), dim3(EW_BLOCK),
-                     0, stream, dy, x, bias, dx, dbias, rows, H);
+  hipLaunchKernelGGL((bias_gelu_bwd_kernel<T, 4>), dim3(row_grid, col_tiles),
+                     dim3(EW_BLOCK), 0, stream, dy, x, bias, dx, dbias, rows, H);
 }
 
 template <typename T>
@@ -480,7 +490,9 @@ template <typename T>
 void launch_ls_axpy_bwd(const T* dout, const T* res, const T* gamma, T* dres,
                         float* dgamma, long rows, int D, hipStream_t stream) {
   const int col_tiles = (D / 8 + EW_BLOCK - 1) / EW_BLOCK;
-  int row_grid = (int)min(rows, (long)(2048 / col_tiles + 1));
+  // cap contributing blocks: dgamma's per-column atomic chain depth == the
+  // number of blocks, and atomics to one address serialize at RMW latency
+  int row_grid = (int)min((rows + 7) / 8, (long)(384 / col_tiles + 1));
   hipLaunchKernelGGL((ls_axpy_bwd_kernel<T>), dim3(row_grid, col_tiles), dim3(EW_BLOCK),
                      0, stream, dout, res, gamma, dres, dgamma, rows, D);
 }
@@ -499,7 +511,7 @@ void launch_ls_axpy_bias_bwd(const T* dout, const T* res, const T* gamma, const 
                              T* dres, float* dgamma, float* dbias, long rows, int D,
                              hipStream_t stream) {
   const int col_tiles = (D / 8 + EW_BLOCK - 1) / EW_BLOCK;
-  int row_grid = (int)min(rows, (long)(2048 / col_tiles + 1));
+  int row_grid = (int)min((rows + 7) / 8, (long)(384 / col_tiles + 1));
   hipLaunchKernelGGL((ls_axpy_bias_bwd_kernel<T>), dim3(row_grid, col_tiles),
                      dim3(EW_BLOCK), 0, stream, dout, res, gamma, bias, dres, dgamma,
                      dbias, rows, D);
@@ -521,7 +533,7 @@ void launch_ls_scatter_bwd(const T* dy, const long* idx, const T* src, const T* 
                            const T* bias, const float* scale, T* dres, float* dgamma,
                            float* dbias, long M, int D, hipStream_t stream) {
   const int col_tiles = (D / 8 + EW_BLOCK - 1) / EW_BLOCK;
-  int row_grid = (int)min(M > 0 ? M : 1, (long)(2048 / col_tiles + 1));
+  int row_grid = (int)min(M > 0 ? (M + 7) / 8 : 1, (long)(384 / col_tiles + 1));
   hipLaunchKernelGGL((ls_scatter_bwd_kernel<T>), dim3(row_grid, col_tiles), dim3(EW_BLOCK),
                      0, stream, dy, idx, src, gamma, bias, scale, dres, dgamma, dbias, M,
                      D);
