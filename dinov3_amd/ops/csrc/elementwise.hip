@@ -144,7 +144,29 @@ __global__ void ls_axpy_bwd_kernel(const T* __restrict__ dout, const T* __restri
   T gb[8];
   Vec8<T>::load(gb, gamma + col8);
   float dg[8] = {0.f, 0.f, 0.f, 0.f, 0.f, 0.f, 0.f, 0.f};
-  for (long row = blockIdx.x; row < rows; row += gridDim.x) {
+  // 4 rows in flight: with the atomic-chain-capped grid (~384 blocks) a
+  // one-row walk leaves too few loads outstanding to cover HBM latency
+  long row = blockIdx.x * 4L;
+  const long rstep = (long)gridDim.x * 4;
+  for (; row + 3 < rows; row += rstep) {
+    T db[4][8], rb[4][8], ob[4][8];
+#pragma unroll
+    for (int rr = 0; rr < 4; ++rr) {
+      Vec8<T>::load(db[rr], dout + (row + rr) * (long)D + col8);
+      Vec8<T>::load(rb[rr], res + (row + rr) * (long)D + col8);
+    }
+#pragma unroll
+    for (int rr = 0; rr < 4; ++rr) {
+#pragma unroll
+      for (int e = 0; e < 8; ++e) {
+        const float g = ScalarOps<T>::load(db[rr] + e);
+        dg[e] += g * ScalarOps<T>::load(rb[rr] + e);
+        ScalarOps<T>::store(ob[rr] + e, g * ScalarOps<T>::load(gb + e));
+      }
+      Vec8<T>::store(dres + (row + rr) * (long)D + col8, ob[rr]);
+    }
+  }
+  for (; row < rows; ++row) {
     const long off = row * (long)D + col8;
     T db[8], rb[8], ob[8];
     Vec8<T>::load(db, dout + off);
@@ -203,7 +225,29 @@ __global__ void ls_axpy_bias_bwd_kernel(const T* __restrict__ dout,
   Vec8<T>::load(bb, bias + col8);
   float dg[8] = {0.f, 0.f, 0.f, 0.f, 0.f, 0.f, 0.f, 0.f};
   float db_acc[8] = {0.f, 0.f, 0.f, 0.f, 0.f, 0.f, 0.f, 0.f};
-  for (long row = blockIdx.x; row < rows; row += gridDim.x) {
+  long row = blockIdx.x * 4L;
+  const long rstep = (long)gridDim.x * 4;
+  for (; row + 3 < rows; row += rstep) {
+    T db[4][8], rb[4][8], ob[4][8];
+#pragma unroll
+    for (int rr = 0; rr < 4; ++rr) {
+      Vec8<T>::load(db[rr], dout + (row + rr) * (long)D + col8);
+      Vec8<T>::load(rb[rr], res + (row + rr) * (long)D + col8);
+    }
+#pragma unroll
+    for (int rr = 0; rr < 4; ++rr) {
+#pragma unroll
+      for (int e = 0; e < 8; ++e) {
+        const float g = ScalarOps<T>::load(db[rr] + e);
+        const float gm = ScalarOps<T>::load(gb + e);
+        dg[e] += g * (ScalarOps<T>::load(rb[rr] + e) + ScalarOps<T>::load(bb + e));
+        db_acc[e] += g * gm;
+        ScalarOps<T>::store(ob[rr] + e, g * gm);
+      }
+      Vec8<T>::store(dres + (row + rr) * (long)D + col8, ob[rr]);
+    }
+  }
+  for (; row < rows; ++row) {
     const long off = row * (long)D + col8;
     T db[8], rb[8], ob[8];
     Vec8<T>::load(db, dout + off);
@@ -280,7 +324,34 @@ __global__ void ls_scatter_bwd_kernel(const T* __restrict__ dy, const long* __re
   if (bias != nullptr) Vec8<T>::load(bb, bias + col8);
   float dg[8] = {0.f, 0.f, 0.f, 0.f, 0.f, 0.f, 0.f, 0.f};
   float db_acc[8] = {0.f, 0.f, 0.f, 0.f, 0.f, 0.f, 0.f, 0.f};
-  for (long r = blockIdx.x; r < M; r += gridDim.x) {
+  long r = blockIdx.x * 4L;
+  const long rstep4 = (long)gridDim.x * 4;
+  for (; r + 3 < M; r += rstep4) {
+    long srow[4];
+    float sc[4];
+    T dyb[4][8], rb[4][8], ob[4][8];
+#pragma unroll
+    for (int rr = 0; rr < 4; ++rr) {
+      srow[rr] = idx[r + rr];
+      sc[rr] = scale != nullptr ? scale[r + rr] : 1.0f;
+      Vec8<T>::load(dyb[rr], dy + srow[rr] * (long)D + col8);
+      Vec8<T>::load(rb[rr], src + (r + rr) * (long)D + col8);
+    }
+#pragma unroll
+    for (int rr = 0; rr < 4; ++rr) {
+#pragma unroll
+      for (int e = 0; e < 8; ++e) {
+        const float gv = ScalarOps<T>::load(dyb[rr] + e) * sc[rr];
+        const float gm = gamma != nullptr ? ScalarOps<T>::load(gb + e) : 1.0f;
+        const float bv = bias != nullptr ? ScalarOps<T>::load(bb + e) : 0.0f;
+        dg[e] += gv * (ScalarOps<T>::load(rb[rr] + e) + bv);
+        db_acc[e] += gv * gm;
+        ScalarOps<T>::store(ob[rr] + e, gv * gm);
+      }
+      Vec8<T>::store(dres + (r + rr) * (long)D + col8, ob[rr]);
+    }
+  }
+  for (; r < M; ++r) {
     const long srow = idx[r];
     const float s = scale != nullptr ? scale[r] : 1.0f;
     T dyb[8], rb[8], ob[8];
