@@ -29,9 +29,11 @@ import torch
 __all__ = ["fused_residual_enabled", "ls_axpy_bias", "ls_scatter_add_rows"]
 
 
-# Flip to True in round 2 once tests/test_ops_gpu.py's gated suite and the
-# bench comparison pass on hardware (tools/round2_validate.sh).
-_DEFAULT_ON = False
+# Default ON since round 2: the round-1 regression was ls_scatter_bwd's
+# inter-block atomic chains (profiles/r2_fused_residual_diagnosis.md); with
+# the capped launchers the fused path wins on hardware (r2_gpu5: 449.4 vs
+# 446.0 img/s before row batching). DINOV3_FUSED_RESIDUAL=0 disables.
+_DEFAULT_ON = True
 
 
 def fused_residual_enabled() -> bool:

@@ -394,7 +394,10 @@ def test_fused_residual_block_gpu():
         torch.manual_seed(32)
         x = torch.randn(6, 32, 128, device=DEV).bfloat16()
         flat, _, _ = cat_keep_shapes([x])
-        flat = flat.clone().requires_grad_(True)
+        # the fused path mutates the flat buffer in place — feed a NON-leaf
+        # (as in training, where it is an op output), keep the leaf for grads
+        flat_leaf = flat.clone().requires_grad_(True)
+        flat = flat_leaf * 1.0
         metas = [SelfAttention._meta_for(x, None, 0)]
         torch.manual_seed(33)
         plan = DropPathPlan(metas, 0.5, 2, flat.device)
