@@ -1,0 +1,25 @@
+#!/bin/bash
+# Round-2 GPU call #6: validate the new defaults (fused residual ON +
+# capped/batched LS backward kernels) — suite, bench, fresh kernel profile.
+set -x
+export TMPDIR=/tmp
+cd "$GRAFT_REPO_ROOT" || cd /root/repo
+OUT=gpurun_out/r2_gpu6
+mkdir -p "$OUT"
+
+timeout 480 python -m pytest tests -m gpu -q > "$OUT/pytest.log" 2>&1
+echo "gpu_suite=$?" >> "$OUT/summary.txt"
+
+timeout 200 python bench.py --steps 15 --warmup 4 2>/dev/null | tail -1 > "$OUT/bench_default.json"
+DINOV3_FUSED_RESIDUAL=0 timeout 200 python bench.py --steps 15 --warmup 4 2>/dev/null | tail -1 > "$OUT/bench_nofused.json"
+
+cd /tmp
+timeout 300 rocprofv3 --kernel-trace --stats -d "$GRAFT_REPO_ROOT/$OUT/prof" -o def --output-format csv -- \
+    python "$GRAFT_REPO_ROOT/bench.py" --steps 3 --warmup 1 \
+    > "$GRAFT_REPO_ROOT/$OUT/prof.log" 2>&1
+echo "prof=$?" >> "$GRAFT_REPO_ROOT/$OUT/summary.txt"
+cd "$GRAFT_REPO_ROOT"
+
+cat "$OUT/summary.txt"
+tail -3 "$OUT/pytest.log"
+for f in "$OUT"/bench_*.json; do echo "$f"; cat "$f"; done
