@@ -187,8 +187,10 @@ std::vector<torch::Tensor> layernorm_bwd(torch::Tensor dy, torch::Tensor x, torc
   const int D = x.size(-1);
   const long rows = x.numel() / D;
   auto dx = torch::empty_like(x);
-  auto dw = torch::zeros({D}, x.options().dtype(torch::kFloat));
-  auto db = torch::zeros({D}, x.options().dtype(torch::kFloat));
+  // 8 shadow accumulators (LN_SHADOWS in norms.hip): per-address atomic
+  // chains shrink 8x; the [8, D] partials are summed here
+  auto dw = torch::zeros({8, D}, x.options().dtype(torch::kFloat));
+  auto db = torch::zeros({8, D}, x.options().dtype(torch::kFloat));
   DISPATCH_FLOAT_BF16(x.scalar_type(), "layernorm_bwd", [&] {
     launch_layernorm_bwd<scalar_t>(
         (const scalar_t*)dy.data_ptr(), (const scalar_t*)x.data_ptr(),
@@ -196,7 +198,7 @@ std::vector<torch::Tensor> layernorm_bwd(torch::Tensor dy, torch::Tensor x, torc
         (scalar_t*)dx.data_ptr(), dw.data_ptr<float>(), db.data_ptr<float>(), rows, D,
         current_stream());
   });
-  return {dx, dw.to(x.scalar_type()), db.to(x.scalar_type())};
+  return {dx, dw.sum(0).to(x.scalar_type()), db.sum(0).to(x.scalar_type())};
 }
 
 std::vector<torch::Tensor> rmsnorm_fwd(torch::Tensor x, torch::Tensor w, double eps) {

@@ -9,6 +9,9 @@
 
 #include "common.h"
 
+// power of two: number of dgamma/dbeta shadow accumulators (see layernorm_bwd)
+#define LN_SHADOWS 8
+
 #define NORM_BLOCK 256
 
 // block size matched to the row width so no lanes idle in the 8-wide loops
@@ -145,9 +148,14 @@ __global__ void layernorm_bwd_kernel(
     }
     __syncthreads();
   }
+  // shadow copies: chain depth of the per-address atomic RMW is
+  // gridDim/shadows instead of gridDim (dw/db are [shadows, D], summed by
+  // the caller)
+  float* dw_s = dw + (long)(blockIdx.x & (LN_SHADOWS - 1)) * D;
+  float* db_s = db + (long)(blockIdx.x & (LN_SHADOWS - 1)) * D;
   for (int i = threadIdx.x; i < D; i += blockDim.x) {
-    atomicAdd(dw + i, dw_acc[i]);
-    atomicAdd(db + i, db_acc[i]);
+    atomicAdd(dw_s + i, dw_acc[i]);
+    atomicAdd(db_s + i, db_acc[i]);
   }
 }
 
