@@ -279,14 +279,14 @@ std::vector<torch::Tensor> bias_gelu_bwd(torch::Tensor dy, torch::Tensor x,
   const int H = x.size(-1);
   const long rows = x.numel() / H;
   auto dx = torch::empty_like(x);
-  auto dbias = torch::zeros({H}, x.options().dtype(torch::kFloat));
+  auto dbias = torch::zeros({8, H}, x.options().dtype(torch::kFloat));
   DISPATCH_FLOAT_BF16(x.scalar_type(), "bias_gelu_bwd", [&] {
     launch_bias_gelu_bwd<scalar_t>((const scalar_t*)dy.data_ptr(),
                                    (const scalar_t*)x.data_ptr(),
                                    (const scalar_t*)bias.data_ptr(), (scalar_t*)dx.data_ptr(),
                                    dbias.data_ptr<float>(), rows, H, current_stream());
   });
-  return {dx, dbias.to(x.scalar_type())};
+  return {dx, dbias.sum(0).to(x.scalar_type())};
 }
 
 torch::Tensor ls_axpy_fwd(torch::Tensor x, torch::Tensor res, torch::Tensor gamma) {
@@ -328,7 +328,7 @@ std::vector<torch::Tensor> ls_axpy_bwd(torch::Tensor dout, torch::Tensor res,
   const int D = dout.size(-1);
   const long rows = dout.numel() / D;
   auto dres = torch::empty_like(dout);
-  auto dgamma = torch::zeros({D}, dout.options().dtype(torch::kFloat));
+  auto dgamma = torch::zeros({8, D}, dout.options().dtype(torch::kFloat));
   DISPATCH_FLOAT_BF16(dout.scalar_type(), "ls_axpy_bwd", [&] {
     launch_ls_axpy_bwd<scalar_t>((const scalar_t*)dout.data_ptr(),
                                  (const scalar_t*)res.data_ptr(),
@@ -336,7 +336,7 @@ std::vector<torch::Tensor> ls_axpy_bwd(torch::Tensor dout, torch::Tensor res,
                                  (scalar_t*)dres.data_ptr(), dgamma.data_ptr<float>(),
                                  rows, D, current_stream());
   });
-  return {dres, dgamma.to(dout.scalar_type())};
+  return {dres, dgamma.sum(0).to(dout.scalar_type())};
 }
 
 std::vector<torch::Tensor> ls_axpy_bias_bwd(torch::Tensor dout, torch::Tensor res,
@@ -346,8 +346,8 @@ std::vector<torch::Tensor> ls_axpy_bias_bwd(torch::Tensor dout, torch::Tensor re
   const long rows = dout.numel() / D;
   auto dres = torch::empty_like(dout);
   auto fopt = dout.options().dtype(torch::kFloat);
-  auto dgamma = torch::zeros({D}, fopt);
-  auto dbias = torch::zeros({D}, fopt);
+  auto dgamma = torch::zeros({8, D}, fopt);
+  auto dbias = torch::zeros({8, D}, fopt);
   DISPATCH_FLOAT_BF16(dout.scalar_type(), "ls_axpy_bias_bwd", [&] {
     launch_ls_axpy_bias_bwd<scalar_t>(
         (const scalar_t*)dout.data_ptr(), (const scalar_t*)res.data_ptr(),
@@ -355,7 +355,7 @@ std::vector<torch::Tensor> ls_axpy_bias_bwd(torch::Tensor dout, torch::Tensor re
         (scalar_t*)dres.data_ptr(), dgamma.data_ptr<float>(), dbias.data_ptr<float>(),
         rows, D, current_stream());
   });
-  return {dres, dgamma.to(dout.scalar_type()), dbias.to(dout.scalar_type())};
+  return {dres, dgamma.sum(0).to(dout.scalar_type()), dbias.sum(0).to(dout.scalar_type())};
 }
 
 torch::Tensor row_gather(torch::Tensor src, torch::Tensor idx) {
@@ -425,8 +425,8 @@ std::vector<torch::Tensor> ls_scatter_bwd(torch::Tensor dy, torch::Tensor idx,
   const long M = idx.numel();
   auto dres = torch::empty_like(src);
   auto fopt = dy.options().dtype(torch::kFloat);
-  auto dgamma = gamma.defined() ? torch::zeros({D}, fopt) : torch::Tensor();
-  auto dbias = bias.defined() ? torch::zeros({D}, fopt) : torch::Tensor();
+  auto dgamma = gamma.defined() ? torch::zeros({8, D}, fopt) : torch::Tensor();
+  auto dbias = bias.defined() ? torch::zeros({8, D}, fopt) : torch::Tensor();
   DISPATCH_FLOAT_BF16(dy.scalar_type(), "ls_scatter_bwd", [&] {
     launch_ls_scatter_bwd<scalar_t>(
         (const scalar_t*)dy.data_ptr(), idx.data_ptr<long>(),
@@ -438,8 +438,8 @@ std::vector<torch::Tensor> ls_scatter_bwd(torch::Tensor dy, torch::Tensor idx,
         dgamma.defined() ? dgamma.data_ptr<float>() : nullptr,
         dbias.defined() ? dbias.data_ptr<float>() : nullptr, M, D, current_stream());
   });
-  return {dres, dgamma.defined() ? dgamma.to(dy.scalar_type()) : dgamma,
-          dbias.defined() ? dbias.to(dy.scalar_type()) : dbias};
+  return {dres, dgamma.defined() ? dgamma.sum(0).to(dy.scalar_type()) : dgamma,
+          dbias.defined() ? dbias.sum(0).to(dy.scalar_type()) : dbias};
 }
 
 torch::Tensor swiglu_fwd(torch::Tensor x12) {
