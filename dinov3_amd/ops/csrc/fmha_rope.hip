@@ -100,8 +100,14 @@ struct QkvView {
 // ---------------------------------------------------------------------------
 // Forward
 // ---------------------------------------------------------------------------
-template <int HD, int NT = 256>
-__global__ __launch_bounds__(NT) void fwd_kernel(
+// QLDS/PREFETCH: the hd-128 long-N configuration holds 304 registers/thread
+// with in-register Q fragments + double-buffered K/V staging -> occupancy
+// 1 wave/SIMD and fully exposed HBM latency (252 GB/s effective at N=2305).
+// QLDS parks the rotated Q fragments in LDS in their REGISTER layout (lane-
+// indexed, conflict-free b128), PREFETCH=false drops the register double
+// buffer; together they fit 2 waves/SIMD.
+template <int HD, int NT = 256, bool QLDS = false, bool PREFETCH = true, int MINW = 1>
+__global__ __launch_bounds__(NT, MINW) void fwd_kernel(
     const __hip_bfloat16* __restrict__ qkv, const float* __restrict__ sin_t,
     const float* __restrict__ cos_t, __hip_bfloat16* __restrict__ o,
     float* __restrict__ lse, int B, int H, int N, int P, float scale) {
@@ -131,22 +137,33 @@ __global__ __launch_bounds__(NT) void fwd_kernel(
   extern __shared__ __attribute__((aligned(16))) char smem_raw[];
   __hip_bfloat16* k_lds = reinterpret_cast<__hip_bfloat16*>(smem_raw);
   __hip_bfloat16* vt_lds = k_lds + 2 * KVB * LDS_STRIDE;
+  // per-wave Q fragment store in register layout: index = slice, lane
+  __hip_bfloat16* q_lds = vt_lds + HD * VT_STRIDE + (QLDS ? wave * KSLICES * 64 * 8 : 0);
 
   // Q fragments (+ rope)
-  bf16x8 qf[KSLICES];
+  bf16x8 qf[QLDS ? 1 : KSLICES];
   {
+    bf16x8 qtmp[KSLICES];
     const int qrow = q0 + l31;
     const int safe = qrow < N ? qrow : (N - 1);
 #pragma unroll
-    for (int s = 0; s < KSLICES; ++s) qf[s] = load8(qv.at(safe, 0, s * 16 + hhalf * 8));
+    for (int s = 0; s < KSLICES; ++s) qtmp[s] = load8(qv.at(safe, 0, s * 16 + hhalf * 8));
     const int p = safe - prefix;
     if (use_rope && p >= 0) {
       const float* srow = sin_t + (long)p * HD;
       const float* crow = cos_t + (long)p * HD;
 #pragma unroll
       for (int s2 = 0; s2 < KSLICES / 2; ++s2) {
-        rope_rotate8(qf[s2], qf[s2 + KSLICES / 2], srow, crow, s2 * 16 + hhalf * 8);
+        rope_rotate8(qtmp[s2], qtmp[s2 + KSLICES / 2], srow, crow, s2 * 16 + hhalf * 8);
       }
+    }
+    if constexpr (QLDS) {
+#pragma unroll
+      for (int s = 0; s < KSLICES; ++s)
+        *reinterpret_cast<bf16x8*>(&q_lds[(s * 64 + lane) * 8]) = qtmp[s];
+    } else {
+#pragma unroll
+      for (int s = 0; s < KSLICES; ++s) qf[s] = qtmp[s];
     }
   }
 
@@ -217,12 +234,15 @@ __global__ __launch_bounds__(NT) void fwd_kernel(
   };
 
   const int n_super = (N + 2 * KVB - 1) / (2 * KVB);  // 64 keys per barrier pair
-  issue_loads(0);
+  if constexpr (PREFETCH) issue_loads(0);
   for (int kt = 0; kt < n_super; ++kt) {
     const int kbase0 = kt * 2 * KVB;
     __syncthreads();               // compute of tile kt-1 done reading LDS
+    if constexpr (!PREFETCH) issue_loads(kbase0);
     write_tile(kbase0);            // regs -> LDS (rope applied at write)
-    if (kt + 1 < n_super) issue_loads(kbase0 + 2 * KVB);  // fly during compute
+    if constexpr (PREFETCH) {
+      if (kt + 1 < n_super) issue_loads(kbase0 + 2 * KVB);  // fly during compute
+    }
     __syncthreads();
 
    for (int sub = 0; sub < 2 && kbase0 + sub * KVB < N; ++sub) {
@@ -233,7 +253,10 @@ __global__ __launch_bounds__(NT) void fwd_kernel(
 #pragma unroll
     for (int s = 0; s < KSLICES; ++s) {
       bf16x8 af = load8(&k_lds[(krow_off + l31) * LDS_STRIDE + s * 16 + hhalf * 8]);
-      s_acc = MFMA32(af, qf[s], s_acc);
+      bf16x8 qs;
+      if constexpr (QLDS) qs = load8(&q_lds[(s * 64 + lane) * 8]);
+      else qs = qf[s];
+      s_acc = MFMA32(af, qs, s_acc);
     }
     float sv[16];
 #pragma unroll
@@ -297,8 +320,8 @@ __global__ __launch_bounds__(NT) void fwd_kernel(
 // ---------------------------------------------------------------------------
 // Backward dQ
 // ---------------------------------------------------------------------------
-template <int HD, int NT = 256>
-__global__ __launch_bounds__(NT) void bwd_dq_kernel(
+template <int HD, int NT = 256, int MINW = 1>
+__global__ __launch_bounds__(NT, MINW) void bwd_dq_kernel(
     const __hip_bfloat16* __restrict__ qkv, const __hip_bfloat16* __restrict__ dout,
     const float* __restrict__ sin_t, const float* __restrict__ cos_t,
     const float* __restrict__ lse, const float* __restrict__ D,
@@ -487,8 +510,8 @@ __global__ __launch_bounds__(NT) void bwd_dq_kernel(
 // ---------------------------------------------------------------------------
 // Backward dK/dV
 // ---------------------------------------------------------------------------
-template <int HD, int NT = 256>
-__global__ __launch_bounds__(NT) void bwd_dkv_kernel(
+template <int HD, int NT = 256, int MINW = 1>
+__global__ __launch_bounds__(NT, MINW) void bwd_dkv_kernel(
     const __hip_bfloat16* __restrict__ qkv, const __hip_bfloat16* __restrict__ dout,
     const float* __restrict__ sin_t, const float* __restrict__ cos_t,
     const float* __restrict__ lse, const float* __restrict__ D,
@@ -1015,12 +1038,24 @@ void launch_fmha_rope_fwd(const __hip_bfloat16* qkv, const float* sin_t,
                          shmem, stream, qkv, sin_t, cos_t, o, lse, B, H, N, P, scale);
   } else if (HD == 128) {
     size_t shmem = (64 * 136 + 128 * 72) * sizeof(__hip_bfloat16);
-    if (small)
+    if (small) {
       hipLaunchKernelGGL(HIP_KERNEL_NAME(fmha_rope::fwd_kernel<128, 128>), grid, dim3(128),
                          shmem, stream, qkv, sin_t, cos_t, o, lse, B, H, N, P, scale);
-    else
-      hipLaunchKernelGGL(HIP_KERNEL_NAME(fmha_rope::fwd_kernel<128, 256>), grid, dim3(256),
-                         shmem, stream, qkv, sin_t, cos_t, o, lse, B, H, N, P, scale);
+    } else {
+      // QLDS + no-prefetch: 2 waves/SIMD (the register variant runs at 1)
+      auto* kfn = reinterpret_cast<const void*>(
+          HIP_KERNEL_NAME(fmha_rope::fwd_kernel<128, 256, true, false, 2>));
+      static bool attr_ok = [kfn] {
+        return hipFuncSetAttribute(kfn, hipFuncAttributeMaxDynamicSharedMemorySize,
+                                   160 * 1024) == hipSuccess;
+      }();
+      (void)attr_ok;
+      size_t shmem_q = shmem + (size_t)(256 / 64) * (128 / 16) * 64 * 8 *
+                                   sizeof(__hip_bfloat16);
+      hipLaunchKernelGGL(HIP_KERNEL_NAME(fmha_rope::fwd_kernel<128, 256, true, false, 2>),
+                         grid, dim3(256), shmem_q, stream, qkv, sin_t, cos_t, o, lse, B,
+                         H, N, P, scale);
+    }
   }
 }
 
@@ -1055,7 +1090,8 @@ void launch_fmha_rope_bwd_dq(const __hip_bfloat16* qkv, const __hip_bfloat16* do
                          dim3(128), shmem, stream, qkv, dout, sin_t, cos_t, lse, D, dqkv,
                          B, H, N, P, scale);
     else
-      hipLaunchKernelGGL(HIP_KERNEL_NAME(fmha_rope::bwd_dq_kernel<128, 256>), grid,
+      // MINW=2 packs to 256 regs (20 B/lane cold scratch) for 2 waves/SIMD
+      hipLaunchKernelGGL(HIP_KERNEL_NAME(fmha_rope::bwd_dq_kernel<128, 256, 2>), grid,
                          dim3(256), shmem, stream, qkv, dout, sin_t, cos_t, lse, D, dqkv,
                          B, H, N, P, scale);
   }
