@@ -159,6 +159,87 @@ __global__ void layernorm_bwd_kernel(
   }
 }
 
+// Wave-per-row LayerNorm backward: each 64-lane wave owns a row, the row's
+// dy/x/w values stay register-cached between the reduction and the dx pass
+// (ONE read of dy and x instead of two), and the row loop has no block
+// barriers — wave-level __shfl_xor reductions only. Per-wave dgamma/dbeta
+// partials live in LDS and merge once at block end into the shadowed global
+// accumulators. CHUNKS = ceil(D/512) (1: D<=512, 2: 1024, 3: 1536).
+template <typename T, int CHUNKS>
+__global__ __launch_bounds__(256) void layernorm_bwd_wave_kernel(
+    const T* __restrict__ dy, const T* __restrict__ x, const T* __restrict__ w,
+    const float* __restrict__ mean, const float* __restrict__ rstd,
+    T* __restrict__ dx, float* __restrict__ dw, float* __restrict__ db,
+    long rows, int D) {
+  constexpr int NW = 4;  // waves per 256-thread block
+  const int wid = threadIdx.x / 64;
+  const int lane = threadIdx.x & 63;
+  extern __shared__ float smem[];  // [NW][D] dw partials + [NW][D] db
+  float* dw_acc = smem + (long)wid * D;
+  float* db_acc = smem + (long)(NW + wid) * D;
+  for (int i = lane; i < D; i += 64) {
+    dw_acc[i] = 0.f;
+    db_acc[i] = 0.f;
+  }
+  float g[CHUNKS][8], v[CHUNKS][8], wv[CHUNKS][8];
+  for (long row = (long)blockIdx.x * NW + wid; row < rows; row += (long)gridDim.x * NW) {
+    const T* dyr = dy + row * (long)D;
+    const T* xr = x + row * (long)D;
+    const float m = mean[row], r = rstd[row];
+    float sum_dyw = 0.f, sum_dyw_xhat = 0.f;
+#pragma unroll
+    for (int c = 0; c < CHUNKS; ++c) {
+      const int i = (c * 64 + lane) * 8;
+      if (i < D) {
+        row_load8<T, true>(dyr, i, g[c]);
+        row_load8<T, true>(xr, i, v[c]);
+        row_load8<T, true>(w, i, wv[c]);
+#pragma unroll
+        for (int e = 0; e < 8; ++e) {
+          const float xhat = (v[c][e] - m) * r;
+          const float gw = g[c][e] * wv[c][e];
+          sum_dyw += gw;
+          sum_dyw_xhat += gw * xhat;
+          dw_acc[i + e] += g[c][e] * xhat;
+          db_acc[i + e] += g[c][e];
+        }
+      }
+    }
+#pragma unroll
+    for (int off = 32; off > 0; off >>= 1) {
+      sum_dyw += __shfl_xor(sum_dyw, off, 64);
+      sum_dyw_xhat += __shfl_xor(sum_dyw_xhat, off, 64);
+    }
+    const float inv_d = 1.0f / D;
+    T* dxr = dx + row * (long)D;
+#pragma unroll
+    for (int c = 0; c < CHUNKS; ++c) {
+      const int i = (c * 64 + lane) * 8;
+      if (i < D) {
+        float o[8];
+#pragma unroll
+        for (int e = 0; e < 8; ++e) {
+          const float xhat = (v[c][e] - m) * r;
+          o[e] = (g[c][e] * wv[c][e] - (sum_dyw + xhat * sum_dyw_xhat) * inv_d) * r;
+        }
+        row_store8<T, true>(dxr, i, o);
+      }
+    }
+  }
+  __syncthreads();
+  const long sh = (long)(blockIdx.x & (LN_SHADOWS - 1)) * D;
+  for (int i = threadIdx.x; i < D; i += 256) {
+    float a = 0.f, bsum = 0.f;
+#pragma unroll
+    for (int w2 = 0; w2 < NW; ++w2) {
+      a += smem[(long)w2 * D + i];
+      bsum += smem[(long)(NW + w2) * D + i];
+    }
+    atomicAdd(dw + sh + i, a);
+    atomicAdd(db + sh + i, bsum);
+  }
+}
+
 // ------------------------------ RMSNorm --------------------------------
 
 template <typename T, bool V8>
@@ -328,6 +409,24 @@ template <typename T>
 void launch_layernorm_bwd(const T* dy, const T* x, const T* w, const float* mean,
                           const float* rstd, T* dx, float* dw, float* db, long rows,
                           int D, hipStream_t stream) {
+  if (D % 8 == 0 && D <= 1536) {
+    // wave-per-row: one read of dy/x, no block barriers in the row loop
+    int grid = (int)min((rows + 3) / 4, (long)2048);
+    size_t shmem = 2 * 4 * (size_t)D * sizeof(float);
+    if (D <= 512)
+      hipLaunchKernelGGL(HIP_KERNEL_NAME(layernorm_bwd_wave_kernel<T, 1>), dim3(grid),
+                         dim3(256), shmem, stream, dy, x, w, mean, rstd, dx, dw, db,
+                         rows, D);
+    else if (D <= 1024)
+      hipLaunchKernelGGL(HIP_KERNEL_NAME(layernorm_bwd_wave_kernel<T, 2>), dim3(grid),
+                         dim3(256), shmem, stream, dy, x, w, mean, rstd, dx, dw, db,
+                         rows, D);
+    else
+      hipLaunchKernelGGL(HIP_KERNEL_NAME(layernorm_bwd_wave_kernel<T, 3>), dim3(grid),
+                         dim3(256), shmem, stream, dy, x, w, mean, rstd, dx, dw, db,
+                         rows, D);
+    return;
+  }
   int grid = (int)min(rows, (long)2048);
   size_t shmem = (16 + 2 * (size_t)D) * sizeof(float);
   DISPATCH_V8(D, hipLaunchKernelGGL(HIP_KERNEL_NAME(layernorm_bwd_kernel<T, V8>), dim3(grid),
