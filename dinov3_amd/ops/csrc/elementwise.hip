@@ -568,11 +568,12 @@ void launch_ls_axpy_fwd(const T* x, const T* res, const T* gamma, T* out, long r
 template <typename T>
 void launch_ls_axpy_bwd(const T* dout, const T* res, const T* gamma, T* dres,
                         float* dgamma, long rows, int D, hipStream_t stream) {
-  const int col_tiles = (D / 8 + EW_BLOCK - 1) / EW_BLOCK;
-  // cap contributing blocks: dgamma's per-column atomic chain depth == the
-  // number of blocks, and atomics to one address serialize at RMW latency
+  // block sized to the row width (no idle half-waves at D=1024) and the
+  // grid bounded: dgamma's per-column atomic chain depth == blocks/shadows
+  const int block = D / 8 < EW_BLOCK ? D / 8 : EW_BLOCK;
+  const int col_tiles = (D / 8 + block - 1) / block;
   int row_grid = (int)min((rows + 7) / 8, (long)(1024 / col_tiles + 1));
-  hipLaunchKernelGGL((ls_axpy_bwd_kernel<T>), dim3(row_grid, col_tiles), dim3(EW_BLOCK),
+  hipLaunchKernelGGL((ls_axpy_bwd_kernel<T>), dim3(row_grid, col_tiles), dim3(block),
                      0, stream, dout, res, gamma, dres, dgamma, rows, D);
 }
 
@@ -589,10 +590,11 @@ template <typename T>
 void launch_ls_axpy_bias_bwd(const T* dout, const T* res, const T* gamma, const T* bias,
                              T* dres, float* dgamma, float* dbias, long rows, int D,
                              hipStream_t stream) {
-  const int col_tiles = (D / 8 + EW_BLOCK - 1) / EW_BLOCK;
+  const int block = D / 8 < EW_BLOCK ? D / 8 : EW_BLOCK;
+  const int col_tiles = (D / 8 + block - 1) / block;
   int row_grid = (int)min((rows + 7) / 8, (long)(1024 / col_tiles + 1));
   hipLaunchKernelGGL((ls_axpy_bias_bwd_kernel<T>), dim3(row_grid, col_tiles),
-                     dim3(EW_BLOCK), 0, stream, dout, res, gamma, bias, dres, dgamma,
+                     dim3(block), 0, stream, dout, res, gamma, bias, dres, dgamma,
                      dbias, rows, D);
 }
 
@@ -611,9 +613,10 @@ template <typename T>
 void launch_ls_scatter_bwd(const T* dy, const long* idx, const T* src, const T* gamma,
                            const T* bias, const float* scale, T* dres, float* dgamma,
                            float* dbias, long M, int D, hipStream_t stream) {
-  const int col_tiles = (D / 8 + EW_BLOCK - 1) / EW_BLOCK;
+  const int block = D / 8 < EW_BLOCK ? D / 8 : EW_BLOCK;
+  const int col_tiles = (D / 8 + block - 1) / block;
   int row_grid = (int)min(M > 0 ? (M + 7) / 8 : 1, (long)(1024 / col_tiles + 1));
-  hipLaunchKernelGGL((ls_scatter_bwd_kernel<T>), dim3(row_grid, col_tiles), dim3(EW_BLOCK),
+  hipLaunchKernelGGL((ls_scatter_bwd_kernel<T>), dim3(row_grid, col_tiles), dim3(block),
                      0, stream, dy, idx, src, gamma, bias, scale, dres, dgamma, dbias, M,
                      D);
 }

@@ -106,14 +106,14 @@ struct QkvView {
 // QLDS parks the rotated Q fragments in LDS in their REGISTER layout (lane-
 // indexed, conflict-free b128), PREFETCH=false drops the register double
 // buffer; together they fit 2 waves/SIMD.
-template <int HD, int NT = 256, bool QLDS = false, bool PREFETCH = true, int MINW = 1>
+template <int HD, int NT = 256, bool QLDS = false, bool PREFETCH = true, int MINW = 1,
+          int KVB = 32>
 __global__ __launch_bounds__(NT, MINW) void fwd_kernel(
     const __hip_bfloat16* __restrict__ qkv, const float* __restrict__ sin_t,
     const float* __restrict__ cos_t, __hip_bfloat16* __restrict__ o,
     float* __restrict__ lse, int B, int H, int N, int P, float scale) {
   constexpr int KSLICES = HD / 16;
   constexpr int DTILES = HD / 32;
-  constexpr int KVB = 32;
   constexpr int LDS_STRIDE = HD + 8;
   constexpr int VT_STRIDE = 2 * KVB + 8;
   constexpr int HALF = HD / 2;
@@ -245,9 +245,10 @@ __global__ __launch_bounds__(NT, MINW) void fwd_kernel(
     }
     __syncthreads();
 
-   for (int sub = 0; sub < 2 && kbase0 + sub * KVB < N; ++sub) {
-    const int kbase = kbase0 + sub * KVB;
-    const int krow_off = sub * KVB;
+   constexpr int SUBK = 32;  // keys per MFMA tile (fixed by the 32x32 MFMA)
+   for (int sub = 0; sub < (2 * KVB) / SUBK && kbase0 + sub * SUBK < N; ++sub) {
+    const int kbase = kbase0 + sub * SUBK;
+    const int krow_off = sub * SUBK;
 
     f32x16 s_acc = {};
 #pragma unroll
@@ -444,9 +445,10 @@ __global__ __launch_bounds__(NT, MINW) void bwd_dq_kernel(
     if (kt + 1 < n_super) issue_loads(kbase0 + 2 * KVB);
     __syncthreads();
 
-   for (int sub = 0; sub < 2 && kbase0 + sub * KVB < N; ++sub) {
-    const int kbase = kbase0 + sub * KVB;
-    const int krow_off = sub * KVB;
+   constexpr int SUBK = 32;  // keys per MFMA tile (fixed by the 32x32 MFMA)
+   for (int sub = 0; sub < (2 * KVB) / SUBK && kbase0 + sub * SUBK < N; ++sub) {
+    const int kbase = kbase0 + sub * SUBK;
+    const int krow_off = sub * SUBK;
     f32x16 s_acc = {}, dp_acc = {};
 #pragma unroll
     for (int s = 0; s < KSLICES; ++s) {
@@ -1029,13 +1031,17 @@ void launch_fmha_rope_fwd(const __hip_bfloat16* qkv, const float* sin_t,
   const bool small = N <= 64;
   dim3 grid(B * H, small ? 1 : (N + 127) / 128);
   if (HD == 64) {
-    size_t shmem = (64 * 72 + 64 * 72) * sizeof(__hip_bfloat16);
-    if (small)
+    if (small) {
+      size_t shmem = (64 * 72 + 64 * 72) * sizeof(__hip_bfloat16);
       hipLaunchKernelGGL(HIP_KERNEL_NAME(fmha_rope::fwd_kernel<64, 128>), grid, dim3(128),
                          shmem, stream, qkv, sin_t, cos_t, o, lse, B, H, N, P, scale);
-    else
-      hipLaunchKernelGGL(HIP_KERNEL_NAME(fmha_rope::fwd_kernel<64, 256>), grid, dim3(256),
-                         shmem, stream, qkv, sin_t, cos_t, o, lse, B, H, N, P, scale);
+    } else {
+      // KVB=64: 128 keys per barrier pair; k_lds [128][72], vt_lds [64][136]
+      size_t shmem = (128 * 72 + 64 * 136) * sizeof(__hip_bfloat16);
+      hipLaunchKernelGGL(
+          HIP_KERNEL_NAME(fmha_rope::fwd_kernel<64, 256, false, true, 1, 64>), grid,
+          dim3(256), shmem, stream, qkv, sin_t, cos_t, o, lse, B, H, N, P, scale);
+    }
   } else if (HD == 128) {
     size_t shmem = (64 * 136 + 128 * 72) * sizeof(__hip_bfloat16);
     if (small) {
