@@ -316,7 +316,7 @@ __global__ void ls_scatter_add_kernel(T* __restrict__ dst, const long* __restric
 // backward: dres[r,c] = dy[idx[r],c] * gamma[c] * scale[r];
 //           dgamma[c] += sum_r dy[idx[r],c] * (src[r,c]+bias[c]) * scale[r];
 //           dbias[c]  += sum_r dy[idx[r],c] * gamma[c] * scale[r]
-template <typename T>
+template <typename T, int RR = 4>
 __global__ void ls_scatter_bwd_kernel(const T* __restrict__ dy, const long* __restrict__ idx,
                                       const T* __restrict__ src,
                                       const T* __restrict__ gamma,
@@ -331,21 +331,21 @@ __global__ void ls_scatter_bwd_kernel(const T* __restrict__ dy, const long* __re
   if (bias != nullptr) Vec8<T>::load(bb, bias + col8);
   float dg[8] = {0.f, 0.f, 0.f, 0.f, 0.f, 0.f, 0.f, 0.f};
   float db_acc[8] = {0.f, 0.f, 0.f, 0.f, 0.f, 0.f, 0.f, 0.f};
-  long r = blockIdx.x * 4L;
-  const long rstep4 = (long)gridDim.x * 4;
-  for (; r + 3 < M; r += rstep4) {
-    long srow[4];
-    float sc[4];
-    T dyb[4][8], rb[4][8], ob[4][8];
+  long r = blockIdx.x * (long)RR;
+  const long rstep4 = (long)gridDim.x * RR;
+  for (; r + RR - 1 < M; r += rstep4) {
+    long srow[RR];
+    float sc[RR];
+    T dyb[RR][8], rb[RR][8], ob[RR][8];
 #pragma unroll
-    for (int rr = 0; rr < 4; ++rr) {
+    for (int rr = 0; rr < RR; ++rr) {
       srow[rr] = idx[r + rr];
       sc[rr] = scale != nullptr ? scale[r + rr] : 1.0f;
       Vec8<T>::load(dyb[rr], dy + srow[rr] * (long)D + col8);
       Vec8<T>::load(rb[rr], src + (r + rr) * (long)D + col8);
     }
 #pragma unroll
-    for (int rr = 0; rr < 4; ++rr) {
+    for (int rr = 0; rr < RR; ++rr) {
 #pragma unroll
       for (int e = 0; e < 8; ++e) {
         const float gv = ScalarOps<T>::load(dyb[rr] + e) * sc[rr];
@@ -615,8 +615,10 @@ void launch_ls_scatter_bwd(const T* dy, const long* idx, const T* src, const T* 
                            float* dbias, long M, int D, hipStream_t stream) {
   const int block = D / 8 < EW_BLOCK ? D / 8 : EW_BLOCK;
   const int col_tiles = (D / 8 + block - 1) / block;
-  int row_grid = (int)min(M > 0 ? (M + 7) / 8 : 1, (long)(1024 / col_tiles + 1));
-  hipLaunchKernelGGL((ls_scatter_bwd_kernel<T>), dim3(row_grid, col_tiles), dim3(block),
+  // 8 rows x 2 gathered streams in flight per thread and a deep grid: this
+  // kernel is HBM-latency-limited, not atomic-limited (shadows cover that)
+  int row_grid = (int)min(M > 0 ? (M + 7) / 8 : 1, (long)(2048 / col_tiles + 1));
+  hipLaunchKernelGGL((ls_scatter_bwd_kernel<T, 8>), dim3(row_grid, col_tiles), dim3(block),
                      0, stream, dy, idx, src, gamma, bias, scale, dres, dgamma, dbias, M,
                      D);
 }
