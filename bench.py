@@ -162,16 +162,23 @@ def main():
         cfg.train.num_workers = args.num_workers
         cfg.train.dataset_path = "Synthetic:split=TRAIN"
         loader = build_data_loader_from_cfg(cfg, model)
-        loader_it = iter(loader)
 
-        def next_batch(i):
-            nonlocal loader_it
-            try:
-                data = next(loader_it)
-            except StopIteration:
-                loader_it = iter(loader)
-                data = next(loader_it)
-            return batch_to_device(data, device)
+        def cycle():
+            while True:
+                yield from loader
+
+        if use_gpu:
+            from dinov3_amd.data.prefetch import CudaBatchPrefetcher
+
+            loader_it = CudaBatchPrefetcher(cycle(), device)
+
+            def next_batch(i):
+                return next(loader_it)
+        else:
+            loader_it = cycle()
+
+            def next_batch(i):
+                return batch_to_device(next(loader_it), device)
     else:
         batches = make_synthetic_batch(cfg, device, dtype)
 

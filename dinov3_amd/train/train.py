@@ -311,8 +311,16 @@ def do_train(cfg, model: SSLMetaArch, resume: bool = True, max_iterations: int =
             epoch += 1
 
     clip = cfg.optim.clip_grad
+    batches = infinite_batches()
+    prefetched = device.type == "cuda"
+    if prefetched:
+        # stage the next batch's H2D copies on a side stream, overlapped with
+        # the current step's compute (data/prefetch.py)
+        from ..data.prefetch import CudaBatchPrefetcher
+
+        batches = CudaBatchPrefetcher(batches, device)
     for data in metric_logger.log_every(
-        infinite_batches(), 10, header, n_iterations=total_iterations, start_iteration=start_iter
+        batches, 10, header, n_iterations=total_iterations, start_iteration=start_iter
     ):
         if iteration >= total_iterations:
             break
@@ -323,7 +331,8 @@ def do_train(cfg, model: SSLMetaArch, resume: bool = True, max_iterations: int =
         teacher_temp = schedulers["teacher_temp"][it]
         last_layer_lr = 0.0 if it < schedulers["freeze_last_layer_iterations"] else lr
 
-        data = batch_to_device(data, device)
+        if not prefetched:
+            data = batch_to_device(data, device)
         if profiling and device.type == "cuda":
             torch.cuda.nvtx.range_push(f"step_{it}")  # roctx range on ROCm
         try:
