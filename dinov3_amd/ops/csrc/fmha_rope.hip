@@ -321,7 +321,7 @@ __global__ __launch_bounds__(NT, MINW) void fwd_kernel(
 // ---------------------------------------------------------------------------
 // Backward dQ
 // ---------------------------------------------------------------------------
-template <int HD, int NT = 256, int MINW = 1>
+template <int HD, int NT = 256, int MINW = 1, int KVB = 32>
 __global__ __launch_bounds__(NT, MINW) void bwd_dq_kernel(
     const __hip_bfloat16* __restrict__ qkv, const __hip_bfloat16* __restrict__ dout,
     const float* __restrict__ sin_t, const float* __restrict__ cos_t,
@@ -329,7 +329,6 @@ __global__ __launch_bounds__(NT, MINW) void bwd_dq_kernel(
     __hip_bfloat16* __restrict__ dqkv, int B, int H, int N, int P, float scale) {
   constexpr int KSLICES = HD / 16;
   constexpr int DTILES = HD / 32;
-  constexpr int KVB = 32;
   constexpr int LDS_STRIDE = HD + 8;
   constexpr int KT_STRIDE = 2 * KVB + 8;
   constexpr int HALF = HD / 2;
@@ -1085,10 +1084,13 @@ void launch_fmha_rope_bwd_dq(const __hip_bfloat16* qkv, const __hip_bfloat16* do
       hipLaunchKernelGGL(HIP_KERNEL_NAME(fmha_rope::bwd_dq_kernel<64, 128>), grid,
                          dim3(128), shmem, stream, qkv, dout, sin_t, cos_t, lse, D, dqkv,
                          B, H, N, P, scale);
-    else
-      hipLaunchKernelGGL(HIP_KERNEL_NAME(fmha_rope::bwd_dq_kernel<64, 256>), grid,
-                         dim3(256), shmem, stream, qkv, dout, sin_t, cos_t, lse, D, dqkv,
-                         B, H, N, P, scale);
+    else {
+      // KVB=64: 128 keys per barrier pair (k_lds+v_lds [128][72], kt [64][136])
+      size_t shmem64 = (2 * 128 * 72 + 64 * 136) * sizeof(__hip_bfloat16);
+      hipLaunchKernelGGL(HIP_KERNEL_NAME(fmha_rope::bwd_dq_kernel<64, 256, 2, 64>), grid,
+                         dim3(256), shmem64, stream, qkv, dout, sin_t, cos_t, lse, D,
+                         dqkv, B, H, N, P, scale);
+    }
   } else if (HD == 128) {
     size_t shmem = (2 * 64 * 136 + 128 * 72) * sizeof(__hip_bfloat16);
     if (small)
