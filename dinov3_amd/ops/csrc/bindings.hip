@@ -189,8 +189,8 @@ std::vector<torch::Tensor> layernorm_bwd(torch::Tensor dy, torch::Tensor x, torc
   auto dx = torch::empty_like(x);
   // 8 shadow accumulators (LN_SHADOWS in norms.hip): per-address atomic
   // chains shrink 8x; the [8, D] partials are summed here
-  auto dw = torch::zeros({8, D}, x.options().dtype(torch::kFloat));
-  auto db = torch::zeros({8, D}, x.options().dtype(torch::kFloat));
+  auto dw = torch::zeros({32, D}, x.options().dtype(torch::kFloat));
+  auto db = torch::zeros({32, D}, x.options().dtype(torch::kFloat));
   DISPATCH_FLOAT_BF16(x.scalar_type(), "layernorm_bwd", [&] {
     launch_layernorm_bwd<scalar_t>(
         (const scalar_t*)dy.data_ptr(), (const scalar_t*)x.data_ptr(),
@@ -279,7 +279,7 @@ std::vector<torch::Tensor> bias_gelu_bwd(torch::Tensor dy, torch::Tensor x,
   const int H = x.size(-1);
   const long rows = x.numel() / H;
   auto dx = torch::empty_like(x);
-  auto dbias = torch::zeros({8, H}, x.options().dtype(torch::kFloat));
+  auto dbias = torch::zeros({32, H}, x.options().dtype(torch::kFloat));
   DISPATCH_FLOAT_BF16(x.scalar_type(), "bias_gelu_bwd", [&] {
     launch_bias_gelu_bwd<scalar_t>((const scalar_t*)dy.data_ptr(),
                                    (const scalar_t*)x.data_ptr(),
@@ -328,7 +328,7 @@ std::vector<torch::Tensor> ls_axpy_bwd(torch::Tensor dout, torch::Tensor res,
   const int D = dout.size(-1);
   const long rows = dout.numel() / D;
   auto dres = torch::empty_like(dout);
-  auto dgamma = torch::zeros({8, D}, dout.options().dtype(torch::kFloat));
+  auto dgamma = torch::zeros({32, D}, dout.options().dtype(torch::kFloat));
   DISPATCH_FLOAT_BF16(dout.scalar_type(), "ls_axpy_bwd", [&] {
     launch_ls_axpy_bwd<scalar_t>((const scalar_t*)dout.data_ptr(),
                                  (const scalar_t*)res.data_ptr(),
@@ -346,8 +346,8 @@ std::vector<torch::Tensor> ls_axpy_bias_bwd(torch::Tensor dout, torch::Tensor re
   const long rows = dout.numel() / D;
   auto dres = torch::empty_like(dout);
   auto fopt = dout.options().dtype(torch::kFloat);
-  auto dgamma = torch::zeros({8, D}, fopt);
-  auto dbias = torch::zeros({8, D}, fopt);
+  auto dgamma = torch::zeros({32, D}, fopt);
+  auto dbias = torch::zeros({32, D}, fopt);
   DISPATCH_FLOAT_BF16(dout.scalar_type(), "ls_axpy_bias_bwd", [&] {
     launch_ls_axpy_bias_bwd<scalar_t>(
         (const scalar_t*)dout.data_ptr(), (const scalar_t*)res.data_ptr(),
@@ -425,8 +425,8 @@ std::vector<torch::Tensor> ls_scatter_bwd(torch::Tensor dy, torch::Tensor idx,
   const long M = idx.numel();
   auto dres = torch::empty_like(src);
   auto fopt = dy.options().dtype(torch::kFloat);
-  auto dgamma = gamma.defined() ? torch::zeros({8, D}, fopt) : torch::Tensor();
-  auto dbias = bias.defined() ? torch::zeros({8, D}, fopt) : torch::Tensor();
+  auto dgamma = gamma.defined() ? torch::zeros({32, D}, fopt) : torch::Tensor();
+  auto dbias = bias.defined() ? torch::zeros({32, D}, fopt) : torch::Tensor();
   DISPATCH_FLOAT_BF16(dy.scalar_type(), "ls_scatter_bwd", [&] {
     launch_ls_scatter_bwd<scalar_t>(
         (const scalar_t*)dy.data_ptr(), idx.data_ptr<long>(),

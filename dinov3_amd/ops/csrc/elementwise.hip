@@ -7,7 +7,7 @@
 // shadow accumulators for the column-reduction backward kernels: per-address
 // atomic RMW chains shrink by this factor (outputs are [EW_SHADOWS, D],
 // summed by the binding)
-#define EW_SHADOWS 8
+#define EW_SHADOWS 32
 
 // fast tanh via the hardware exp pipe: tanh(y) = 1 - 2/(1 + exp(2y)).
 // tanhf() is a slow polyline in libm; __expf is one v_exp_f32 — the GELU

@@ -10,7 +10,7 @@
 #include "common.h"
 
 // power of two: number of dgamma/dbeta shadow accumulators (see layernorm_bwd)
-#define LN_SHADOWS 8
+#define LN_SHADOWS 32
 
 #define NORM_BLOCK 256
 
