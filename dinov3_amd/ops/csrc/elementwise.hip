@@ -615,9 +615,9 @@ void launch_ls_scatter_bwd(const T* dy, const long* idx, const T* src, const T* 
                            float* dbias, long M, int D, hipStream_t stream) {
   const int block = D / 8 < EW_BLOCK ? D / 8 : EW_BLOCK;
   const int col_tiles = (D / 8 + block - 1) / block;
-  // 4 rows in flight (8 measured SLOWER: the 3x[RR][8] register arrays halve
-  // occupancy, r2_gpu10) over a deep grid; shadows keep the atomic chains short
-  int row_grid = (int)min(M > 0 ? (M + 7) / 8 : 1, (long)(2048 / col_tiles + 1));
+  // 4 rows in flight (8 halves occupancy, r2_gpu10) over a 1024-block grid
+  // (2048 measured 34% slower, r2_gpu13); shadows keep atomic chains short
+  int row_grid = (int)min(M > 0 ? (M + 7) / 8 : 1, (long)(1024 / col_tiles + 1));
   hipLaunchKernelGGL((ls_scatter_bwd_kernel<T, 4>), dim3(row_grid, col_tiles), dim3(block),
                      0, stream, dy, idx, src, gamma, bias, scale, dres, dgamma, dbias, M,
                      D);
