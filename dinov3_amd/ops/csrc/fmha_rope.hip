@@ -511,7 +511,11 @@ __global__ __launch_bounds__(NT, MINW) void bwd_dq_kernel(
 // ---------------------------------------------------------------------------
 // Backward dK/dV
 // ---------------------------------------------------------------------------
-template <int HD, int NT = 256, int MINW = 1>
+// MODE splits the kernel for register-starved configs (hd-128 holds 366
+// regs with both accumulator sets -> occupancy 1): MODE=1 computes only dV
+// (no vf/dP/dK state), MODE=2 only dK. Each re-streams Q/dO, but at 2
+// waves/SIMD instead of 1 the latency hiding more than pays for it.
+template <int HD, int NT = 256, int MINW = 1, int MODE = 0>
 __global__ __launch_bounds__(NT, MINW) void bwd_dkv_kernel(
     const __hip_bfloat16* __restrict__ qkv, const __hip_bfloat16* __restrict__ dout,
     const float* __restrict__ sin_t, const float* __restrict__ cos_t,
@@ -549,12 +553,12 @@ __global__ __launch_bounds__(NT, MINW) void bwd_dkv_kernel(
 
   const int krow = k0 + l31;
   const int safe = krow < N ? krow : (N - 1);
-  bf16x8 kf[KSLICES], vf[KSLICES];
+  bf16x8 kf[KSLICES], vf[MODE == 1 ? 1 : KSLICES];
   {
 #pragma unroll
     for (int s = 0; s < KSLICES; ++s) {
       kf[s] = load8(qv.at(safe, 1, s * 16 + hhalf * 8));
-      vf[s] = load8(qv.at(safe, 2, s * 16 + hhalf * 8));
+      if constexpr (MODE != 1) vf[s] = load8(qv.at(safe, 2, s * 16 + hhalf * 8));
     }
     const int p = safe - prefix;
     if (use_rope && p >= 0) {
@@ -566,13 +570,13 @@ __global__ __launch_bounds__(NT, MINW) void bwd_dkv_kernel(
     }
   }
 
-  float dk_acc[DTILES][16], dv_acc[DTILES][16];
+  float dk_acc[MODE == 1 ? 1 : DTILES][16], dv_acc[MODE == 2 ? 1 : DTILES][16];
 #pragma unroll
   for (int t = 0; t < DTILES; ++t)
 #pragma unroll
     for (int r = 0; r < 16; ++r) {
-      dk_acc[t][r] = 0.f;
-      dv_acc[t][r] = 0.f;
+      if constexpr (MODE != 1) dk_acc[t][r] = 0.f;
+      if constexpr (MODE != 2) dv_acc[t][r] = 0.f;
     }
 
   constexpr int PAIRS_PER_ROW = HALF / 8;
@@ -666,10 +670,11 @@ __global__ __launch_bounds__(NT, MINW) void bwd_dkv_kernel(
       for (int e = 0; e < 8; ++e) {
         const int d = s * 16 + hhalf * 8 + e;
         reinterpret_cast<__hip_bfloat16*>(&aq)[e] = qt_lds[d * QT_STRIDE + qrow_off + l31];
-        reinterpret_cast<__hip_bfloat16*>(&ad)[e] = dot_lds[d * QT_STRIDE + qrow_off + l31];
+        if constexpr (MODE != 1)
+          reinterpret_cast<__hip_bfloat16*>(&ad)[e] = dot_lds[d * QT_STRIDE + qrow_off + l31];
       }
       s_acc = MFMA32(aq.v8, kf[s], s_acc);
-      dp_acc = MFMA32(ad.v8, vf[s], dp_acc);
+      if constexpr (MODE != 1) dp_acc = MFMA32(ad.v8, vf[s], dp_acc);
     }
     float pv[16], ds[16];
 #pragma unroll
@@ -679,36 +684,49 @@ __global__ __launch_bounds__(NT, MINW) void bwd_dkv_kernel(
       const float l = lse_lds[qrow_off + c_row(r, hhalf)];
       float p = valid ? __expf(s_acc[r] * scale - l) : 0.f;
       pv[r] = p;
-      ds[r] = p * (dp_acc[r] - d_lds[qrow_off + c_row(r, hhalf)]) * scale;
+      if constexpr (MODE != 1)
+        ds[r] = p * (dp_acc[r] - d_lds[qrow_off + c_row(r, hhalf)]) * scale;
     }
-    bf16x8 p0 = pack_fragment(pv, 0);
-    bf16x8 p1 = pack_fragment(pv, 8);
-    bf16x8 s0 = pack_fragment(ds, 0);
-    bf16x8 s1 = pack_fragment(ds, 8);
+    bf16x8 p0, p1, s0, s1;
+    if constexpr (MODE != 2) {
+      p0 = pack_fragment(pv, 0);
+      p1 = pack_fragment(pv, 8);
+    }
+    if constexpr (MODE != 1) {
+      s0 = pack_fragment(ds, 0);
+      s1 = pack_fragment(ds, 8);
+    }
 #pragma unroll
     for (int t = 0; t < DTILES; ++t) {
       union { unsigned u[4]; bf16x8 v8; } ado0, ado1, aq0, aq1;
 #pragma unroll
       for (int e = 0; e < 8; ++e) {
-        reinterpret_cast<__hip_bfloat16*>(&ado0)[e] = dot_lds[(t * 32 + l31) * QT_STRIDE + qrow_off + hhalf * 8 + e];
-        reinterpret_cast<__hip_bfloat16*>(&ado1)[e] = dot_lds[(t * 32 + l31) * QT_STRIDE + qrow_off + 16 + hhalf * 8 + e];
-        reinterpret_cast<__hip_bfloat16*>(&aq0)[e] = qt_lds[(t * 32 + l31) * QT_STRIDE + qrow_off + hhalf * 8 + e];
-        reinterpret_cast<__hip_bfloat16*>(&aq1)[e] = qt_lds[(t * 32 + l31) * QT_STRIDE + qrow_off + 16 + hhalf * 8 + e];
+        if constexpr (MODE != 2) {
+          reinterpret_cast<__hip_bfloat16*>(&ado0)[e] = dot_lds[(t * 32 + l31) * QT_STRIDE + qrow_off + hhalf * 8 + e];
+          reinterpret_cast<__hip_bfloat16*>(&ado1)[e] = dot_lds[(t * 32 + l31) * QT_STRIDE + qrow_off + 16 + hhalf * 8 + e];
+        }
+        if constexpr (MODE != 1) {
+          reinterpret_cast<__hip_bfloat16*>(&aq0)[e] = qt_lds[(t * 32 + l31) * QT_STRIDE + qrow_off + hhalf * 8 + e];
+          reinterpret_cast<__hip_bfloat16*>(&aq1)[e] = qt_lds[(t * 32 + l31) * QT_STRIDE + qrow_off + 16 + hhalf * 8 + e];
+        }
       }
-      f32x16 accv, acck;
+      if constexpr (MODE != 2) {
+        f32x16 accv;
 #pragma unroll
-      for (int r = 0; r < 16; ++r) {
-        accv[r] = dv_acc[t][r];
-        acck[r] = dk_acc[t][r];
+        for (int r = 0; r < 16; ++r) accv[r] = dv_acc[t][r];
+        accv = MFMA32(ado0.v8, p0, accv);
+        accv = MFMA32(ado1.v8, p1, accv);
+#pragma unroll
+        for (int r = 0; r < 16; ++r) dv_acc[t][r] = accv[r];
       }
-      accv = MFMA32(ado0.v8, p0, accv);
-      accv = MFMA32(ado1.v8, p1, accv);
-      acck = MFMA32(aq0.v8, s0, acck);
-      acck = MFMA32(aq1.v8, s1, acck);
+      if constexpr (MODE != 1) {
+        f32x16 acck;
 #pragma unroll
-      for (int r = 0; r < 16; ++r) {
-        dv_acc[t][r] = accv[r];
-        dk_acc[t][r] = acck[r];
+        for (int r = 0; r < 16; ++r) acck[r] = dk_acc[t][r];
+        acck = MFMA32(aq0.v8, s0, acck);
+        acck = MFMA32(aq1.v8, s1, acck);
+#pragma unroll
+        for (int r = 0; r < 16; ++r) dk_acc[t][r] = acck[r];
       }
     }
    }
@@ -716,19 +734,21 @@ __global__ __launch_bounds__(NT, MINW) void bwd_dkv_kernel(
 
   if (krow < N) {
     const int p = krow - prefix;
-    if (use_rope && p >= 0) {
-      const float* srow = sin_t + (long)p * HD;
-      const float* crow = cos_t + (long)p * HD;
+    if constexpr (MODE != 1) {
+      if (use_rope && p >= 0) {
+        const float* srow = sin_t + (long)p * HD;
+        const float* crow = cos_t + (long)p * HD;
 #pragma unroll
-      for (int t = 0; t < DTILES / 2; ++t)
+        for (int t = 0; t < DTILES / 2; ++t)
 #pragma unroll
-        for (int r = 0; r < 16; ++r) {
-          const int d = t * 32 + c_row(r, hhalf);
-          const float c = crow[d], s = srow[d];
-          const float glo = dk_acc[t][r], ghi = dk_acc[t + DTILES / 2][r];
-          dk_acc[t][r] = glo * c + ghi * s;
-          dk_acc[t + DTILES / 2][r] = ghi * c - glo * s;
-        }
+          for (int r = 0; r < 16; ++r) {
+            const int d = t * 32 + c_row(r, hhalf);
+            const float c = crow[d], s = srow[d];
+            const float glo = dk_acc[t][r], ghi = dk_acc[t + DTILES / 2][r];
+            dk_acc[t][r] = glo * c + ghi * s;
+            dk_acc[t + DTILES / 2][r] = ghi * c - glo * s;
+          }
+      }
     }
     __hip_bfloat16* dkp = dqkv + ((((long)b * N + krow) * 3 + 1) * H + h) * HD;
     __hip_bfloat16* dvp = dqkv + ((((long)b * N + krow) * 3 + 2) * H + h) * HD;
@@ -737,8 +757,10 @@ __global__ __launch_bounds__(NT, MINW) void bwd_dkv_kernel(
 #pragma unroll
       for (int r = 0; r < 16; ++r) {
         const int d = t * 32 + c_row(r, hhalf);
-        *reinterpret_cast<short*>(dkp + d) = f32_to_bf16(dk_acc[t][r]);
-        *reinterpret_cast<short*>(dvp + d) = f32_to_bf16(dv_acc[t][r]);
+        if constexpr (MODE != 1)
+          *reinterpret_cast<short*>(dkp + d) = f32_to_bf16(dk_acc[t][r]);
+        if constexpr (MODE != 2)
+          *reinterpret_cast<short*>(dvp + d) = f32_to_bf16(dv_acc[t][r]);
       }
   }
 }
@@ -754,7 +776,7 @@ __global__ __launch_bounds__(NT, MINW) void bwd_dkv_kernel(
 // matching the stride the fwd/dq kernels already use. Validate on hardware
 // (tests/test_fmha_rope_gpu.py::test_dkv64_variant) before enabling.
 // ---------------------------------------------------------------------------
-template <int HD, int NT = 256>
+template <int HD, int NT = 256, int MODE = 0>
 __global__ __launch_bounds__(NT) void bwd_dkv64_kernel(
     const __hip_bfloat16* __restrict__ qkv, const __hip_bfloat16* __restrict__ dout,
     const float* __restrict__ sin_t, const float* __restrict__ cos_t,
@@ -793,12 +815,12 @@ __global__ __launch_bounds__(NT) void bwd_dkv64_kernel(
 
   const int krow = k0 + l31;
   const int safe = krow < N ? krow : (N - 1);
-  bf16x8 kf[KSLICES], vf[KSLICES];
+  bf16x8 kf[KSLICES], vf[MODE == 1 ? 1 : KSLICES];
   {
 #pragma unroll
     for (int s = 0; s < KSLICES; ++s) {
       kf[s] = load8(qv.at(safe, 1, s * 16 + hhalf * 8));
-      vf[s] = load8(qv.at(safe, 2, s * 16 + hhalf * 8));
+      if constexpr (MODE != 1) vf[s] = load8(qv.at(safe, 2, s * 16 + hhalf * 8));
     }
     const int p = safe - prefix;
     if (use_rope && p >= 0) {
@@ -810,13 +832,13 @@ __global__ __launch_bounds__(NT) void bwd_dkv64_kernel(
     }
   }
 
-  float dk_acc[DTILES][16], dv_acc[DTILES][16];
+  float dk_acc[MODE == 1 ? 1 : DTILES][16], dv_acc[MODE == 2 ? 1 : DTILES][16];
 #pragma unroll
   for (int t = 0; t < DTILES; ++t)
 #pragma unroll
     for (int r = 0; r < 16; ++r) {
-      dk_acc[t][r] = 0.f;
-      dv_acc[t][r] = 0.f;
+      if constexpr (MODE != 1) dk_acc[t][r] = 0.f;
+      if constexpr (MODE != 2) dv_acc[t][r] = 0.f;
     }
 
   constexpr int PAIRS_PER_ROW = HALF / 8;
@@ -956,19 +978,21 @@ __global__ __launch_bounds__(NT) void bwd_dkv64_kernel(
 
   if (krow < N) {
     const int p = krow - prefix;
-    if (use_rope && p >= 0) {
-      const float* srow = sin_t + (long)p * HD;
-      const float* crow = cos_t + (long)p * HD;
+    if constexpr (MODE != 1) {
+      if (use_rope && p >= 0) {
+        const float* srow = sin_t + (long)p * HD;
+        const float* crow = cos_t + (long)p * HD;
 #pragma unroll
-      for (int t = 0; t < DTILES / 2; ++t)
+        for (int t = 0; t < DTILES / 2; ++t)
 #pragma unroll
-        for (int r = 0; r < 16; ++r) {
-          const int d = t * 32 + c_row(r, hhalf);
-          const float c = crow[d], s = srow[d];
-          const float glo = dk_acc[t][r], ghi = dk_acc[t + DTILES / 2][r];
-          dk_acc[t][r] = glo * c + ghi * s;
-          dk_acc[t + DTILES / 2][r] = ghi * c - glo * s;
-        }
+          for (int r = 0; r < 16; ++r) {
+            const int d = t * 32 + c_row(r, hhalf);
+            const float c = crow[d], s = srow[d];
+            const float glo = dk_acc[t][r], ghi = dk_acc[t + DTILES / 2][r];
+            dk_acc[t][r] = glo * c + ghi * s;
+            dk_acc[t + DTILES / 2][r] = ghi * c - glo * s;
+          }
+      }
     }
     __hip_bfloat16* dkp = dqkv + ((((long)b * N + krow) * 3 + 1) * H + h) * HD;
     __hip_bfloat16* dvp = dqkv + ((((long)b * N + krow) * 3 + 2) * H + h) * HD;
@@ -977,8 +1001,10 @@ __global__ __launch_bounds__(NT) void bwd_dkv64_kernel(
 #pragma unroll
       for (int r = 0; r < 16; ++r) {
         const int d = t * 32 + c_row(r, hhalf);
-        *reinterpret_cast<short*>(dkp + d) = f32_to_bf16(dk_acc[t][r]);
-        *reinterpret_cast<short*>(dvp + d) = f32_to_bf16(dv_acc[t][r]);
+        if constexpr (MODE != 1)
+          *reinterpret_cast<short*>(dkp + d) = f32_to_bf16(dk_acc[t][r]);
+        if constexpr (MODE != 2)
+          *reinterpret_cast<short*>(dvp + d) = f32_to_bf16(dv_acc[t][r]);
       }
   }
 }
@@ -1143,9 +1169,15 @@ void launch_fmha_rope_bwd_dkv(const __hip_bfloat16* qkv, const __hip_bfloat16* d
       hipLaunchKernelGGL(HIP_KERNEL_NAME(fmha_rope::bwd_dkv64_kernel<128, 256>), grid,
                          dim3(256), shmem, stream, qkv, dout, sin_t, cos_t, lse, D, dqkv,
                          B, H, N, P, scale);
-    else
-      hipLaunchKernelGGL(HIP_KERNEL_NAME(fmha_rope::bwd_dkv_kernel<128, 256>), grid,
-                         dim3(256), shmem, stream, qkv, dout, sin_t, cos_t, lse, D, dqkv,
-                         B, H, N, P, scale);
+    else {
+      // split dV / dK launches: the combined hd-128 kernel holds 366 regs
+      // (occupancy 1); each half fits 2 waves/SIMD and re-streams Q/dO
+      hipLaunchKernelGGL(HIP_KERNEL_NAME(fmha_rope::bwd_dkv_kernel<128, 256, 2, 1>),
+                         grid, dim3(256), shmem, stream, qkv, dout, sin_t, cos_t, lse, D,
+                         dqkv, B, H, N, P, scale);
+      hipLaunchKernelGGL(HIP_KERNEL_NAME(fmha_rope::bwd_dkv_kernel<128, 256, 2, 2>),
+                         grid, dim3(256), shmem, stream, qkv, dout, sin_t, cos_t, lse, D,
+                         dqkv, B, H, N, P, scale);
+    }
   }
 }
