@@ -186,3 +186,43 @@ def test_same_world_fast_path_unchanged(tmp_path):
     for g1, g2 in zip(opt.groups, opt2.groups):
         for a, b in zip(g1["exp_avg"], g2["exp_avg"]):
             assert torch.equal(a, b)
+
+
+def _sharded4_load_worker(rank, port, tmpdir):
+    """4-rank engine resumes the 2-rank checkpoint written by
+    _sharded_save_worker (sharded -> sharded across world sizes)."""
+    os.environ["MASTER_ADDR"] = "127.0.0.1"
+    os.environ["MASTER_PORT"] = str(port)
+    dist.init_process_group("gloo", rank=rank, world_size=4)
+    from dinov3_amd.parallel.fsdp import ShardedEngine
+    from dinov3_amd.train.optim_state import load_optimizer_state, to_canonical
+
+    payload = torch.load(os.path.join(tmpdir, "rank_0.pth"), weights_only=False)
+    model = _make_model()
+    model.load_state_dict(payload["model"])
+    engine = ShardedEngine(_make_groups(model), align=4)
+    load_optimizer_state(engine, tmpdir, payload, rank=rank, world=4)
+    assert engine.step_count == 2
+    gathered = [None] * 4
+    dist.all_gather_object(gathered, engine.state_dict())
+    canon = to_canonical(gathered)
+    src = to_canonical([torch.load(os.path.join(tmpdir, f"rank_{r}.pth"),
+                                   weights_only=False)["optimizer"] for r in range(2)])
+    for key in src["groups"]:
+        for field in ("exp_avg", "exp_avg_sq"):
+            a, b = canon["groups"][key][field], src["groups"][key][field]
+            assert torch.allclose(a, b, atol=2e-6), (key, field)
+    dist.destroy_process_group()
+
+
+def test_world2_checkpoint_restores_into_world4(tmp_path):
+    _run(_sharded_save_worker, 29667, str(tmp_path))
+    ctx = mp.get_context("spawn")
+    procs = [ctx.Process(target=_sharded4_load_worker, args=(r, 29669, str(tmp_path)))
+             for r in range(4)]
+    for p in procs:
+        p.start()
+    for p in procs:
+        p.join(240)
+    for p in procs:
+        assert p.exitcode == 0, f"child exited with {p.exitcode}"
