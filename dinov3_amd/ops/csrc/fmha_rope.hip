@@ -1169,15 +1169,12 @@ void launch_fmha_rope_bwd_dkv(const __hip_bfloat16* qkv, const __hip_bfloat16* d
       hipLaunchKernelGGL(HIP_KERNEL_NAME(fmha_rope::bwd_dkv64_kernel<128, 256>), grid,
                          dim3(256), shmem, stream, qkv, dout, sin_t, cos_t, lse, D, dqkv,
                          B, H, N, P, scale);
-    else {
-      // split dV / dK launches: the combined hd-128 kernel holds 366 regs
-      // (occupancy 1); each half fits 2 waves/SIMD and re-streams Q/dO
-      hipLaunchKernelGGL(HIP_KERNEL_NAME(fmha_rope::bwd_dkv_kernel<128, 256, 2, 1>),
-                         grid, dim3(256), shmem, stream, qkv, dout, sin_t, cos_t, lse, D,
-                         dqkv, B, H, N, P, scale);
-      hipLaunchKernelGGL(HIP_KERNEL_NAME(fmha_rope::bwd_dkv_kernel<128, 256, 2, 2>),
-                         grid, dim3(256), shmem, stream, qkv, dout, sin_t, cos_t, lse, D,
-                         dqkv, B, H, N, P, scale);
-    }
+    else
+      // combined kernel (occupancy 1 from the 366-reg dual accumulators):
+      // the MODE=1/2 dV/dK split reaches 2 waves/SIMD but re-streams Q/dO
+      // and measured ~5% SLOWER end to end (r2_gpu15) — not used
+      hipLaunchKernelGGL(HIP_KERNEL_NAME(fmha_rope::bwd_dkv_kernel<128, 256>), grid,
+                         dim3(256), shmem, stream, qkv, dout, sin_t, cos_t, lse, D, dqkv,
+                         B, H, N, P, scale);
   }
 }
