@@ -113,6 +113,50 @@ __global__ void bias_gelu_bwd_kernel(const T* __restrict__ dy, const T* __restri
   for (int e = 0; e < 8; ++e) atomicAdd(dbias_s + col8 + e, db[e]);
 }
 
+// Contiguous-row bias+GELU backward (A/B variant, DINOV3_BG_ROWS=1): each
+// block walks whole rows (PASSES x 2048-col sweeps, perfectly coalesced
+// 4 KB bursts) instead of a fixed column slice's strided walk; dbias
+// accumulates in registers per (thread, pass) column set and leaves via the
+// shadowed atomics. PASSES = H / (256*8).
+template <typename T, int PASSES>
+__global__ __launch_bounds__(256) void bias_gelu_bwd_rows_kernel(
+    const T* __restrict__ dy, const T* __restrict__ x, const T* __restrict__ bias,
+    T* __restrict__ dx, float* __restrict__ dbias, long rows, int H) {
+  T bb[PASSES][8];
+  float db[PASSES][8];
+#pragma unroll
+  for (int pss = 0; pss < PASSES; ++pss) {
+    Vec8<T>::load(bb[pss], bias + (pss * 256 + threadIdx.x) * 8);
+#pragma unroll
+    for (int e = 0; e < 8; ++e) db[pss][e] = 0.f;
+  }
+  for (long row = blockIdx.x; row < rows; row += gridDim.x) {
+    const long base = row * (long)H;
+#pragma unroll
+    for (int pss = 0; pss < PASSES; ++pss) {
+      const long off = base + (pss * 256 + threadIdx.x) * 8;
+      T xb[8], gb[8], ob[8];
+      Vec8<T>::load(xb, x + off);
+      Vec8<T>::load(gb, dy + off);
+#pragma unroll
+      for (int e = 0; e < 8; ++e) {
+        const float v = ScalarOps<T>::load(xb + e) + ScalarOps<T>::load(bb[pss] + e);
+        const float g = ScalarOps<T>::load(gb + e) * gelu_tanh_grad(v);
+        ScalarOps<T>::store(ob + e, g);
+        db[pss][e] += g;
+      }
+      Vec8<T>::store(dx + off, ob);
+    }
+  }
+  float* dbias_s = dbias + (long)(blockIdx.x & (EW_SHADOWS - 1)) * H;
+#pragma unroll
+  for (int pss = 0; pss < PASSES; ++pss) {
+    const int c8 = (pss * 256 + threadIdx.x) * 8;
+#pragma unroll
+    for (int e = 0; e < 8; ++e) atomicAdd(dbias_s + c8 + e, db[pss][e]);
+  }
+}
+
 // -------------------- LayerScale + residual add (K10) -------------------
 // out = x + gamma * res ; bwd: dx = dout (aliased), dres = dout * gamma,
 // dgamma = colsum(dout * res). 8-wide; same fixed-column-slice register
@@ -541,6 +585,16 @@ template <typename T>
 void launch_bias_gelu_bwd(const T* dy, const T* x, const T* bias, T* dx, float* dbias,
                           long rows, int H, hipStream_t stream) {
   const int col_tiles = (H / 8 + EW_BLOCK - 1) / EW_BLOCK;
+  static const int rowsv = [] {
+    const char* e = getenv("DINOV3_BG_ROWS");
+    return e && e[0] == '1';
+  }();
+  if (rowsv && H == 4096) {
+    int grid = (int)min(rows, (long)2048);
+    hipLaunchKernelGGL((bias_gelu_bwd_rows_kernel<T, 2>), dim3(grid), dim3(256), 0,
+                       stream, dy, x, bias, dx, dbias, rows, H);
+    return;
+  }
   static const int ilp8 = [] {
     const char* e = getenv("DINOV3_BG_ILP8");
     return e && e[0] == '1';
