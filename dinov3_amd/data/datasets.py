@@ -153,16 +153,48 @@ class ImageNet22k(_RealReaderMixin, _SyntheticSplitDataset):
         return ImageNet22kTarballReader(self.root, self.extra, self._mmap_cache_size)
 
 
-class ADE20K(_SyntheticSplitDataset):
+class ADE20K(_RealReaderMixin, _SyntheticSplitDataset):
+    """Real mode: split txt listing + images/ + annotations/ PNG masks
+    (reference ade20k.py:32-102, I/O un-bypassed here)."""
+
     NAME = "ADE20K"
     NUM_CLASSES = 150
 
+    def __init__(self, split: Split = Split.TRAIN, root: str = "", extra: str = "",
+                 transform=None, target_transform=None, length: Optional[int] = None):
+        super().__init__(split, root, extra, transform, target_transform, length)
+        if root:  # ADE20K has no separate "extra" dir — root alone enables it
+            self.extra = root
+        self._attach_reader(length)
 
-class CocoCaptions(_SyntheticSplitDataset):
+    def _make_reader(self):
+        from .readers import ADE20KReader
+
+        return ADE20KReader(self.root, self.split.value.lower())
+
+
+class CocoCaptions(_RealReaderMixin, _SyntheticSplitDataset):
+    """Real mode: COCO caption json + image dirs (reference
+    coco_captions.py:28-102); target = a random caption of the image."""
+
     NAME = "CocoCaptions"
     NUM_CLASSES = 0
 
+    def __init__(self, split: Split = Split.TRAIN, root: str = "", extra: str = "",
+                 transform=None, target_transform=None, length: Optional[int] = None):
+        super().__init__(split, root, extra, transform, target_transform, length)
+        if root:
+            self.extra = root
+        self._attach_reader(length)
+
+    def _make_reader(self):
+        from .readers import CocoCaptionsReader
+
+        return CocoCaptionsReader(self.root, self.split.value.lower())
+
     def get_target(self, index: int) -> str:
+        if self._reader is not None:
+            return self._reader.get_target(index)
         return ""
 
 

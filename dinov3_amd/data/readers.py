@@ -270,3 +270,87 @@ def make_class_tarball(path: str, images: List[Tuple[str, bytes]]) -> None:
             info = tarfile.TarInfo(name=name)
             info.size = len(data)
             tar.addfile(info, io.BytesIO(data))
+
+
+# --------------------------------------------------------------------------
+# ADE20K (split txt listing + images/ + annotations/ PNG masks)
+# --------------------------------------------------------------------------
+
+
+class ADE20KReader:
+    """`root/ADE20K_object150_<split>.txt` lists image file names; images live
+    under `root/images/<name>`, dense labels under
+    `root/annotations/<name>.png` (reference ade20k.py:32-102 contract)."""
+
+    def __init__(self, root: str, split: str):
+        self.root = root
+        list_path = os.path.join(root, f"ADE20K_object150_{split.lower()}.txt")
+        with open(list_path) as f:
+            names = sorted(f.read().strip().split("\n"))
+        self.image_paths = [os.path.join("images", n) for n in names]
+        self.target_paths = [os.path.join("annotations", os.path.splitext(n)[0] + ".png")
+                             for n in names]
+
+    def __len__(self) -> int:
+        return len(self.image_paths)
+
+    def get_image_data(self, index: int) -> bytes:
+        with open(os.path.join(self.root, self.image_paths[index]), "rb") as f:
+            return f.read()
+
+    def get_target(self, index: int) -> torch.Tensor:
+        """Dense segmentation labels as a [H, W] long tensor."""
+        from PIL import Image
+
+        with Image.open(os.path.join(self.root, self.target_paths[index])) as im:
+            return torch.from_numpy(np.array(im, dtype=np.int64))
+
+    def get_targets(self):
+        return None
+
+
+# --------------------------------------------------------------------------
+# COCO captions (annotation json + image dir)
+# --------------------------------------------------------------------------
+
+
+class CocoCaptionsReader:
+    """COCO captions layout (reference coco_captions.py:28-54):
+    train = annotations_trainval2014/annotations/captions_train2014.json +
+    train2014/train2014; val = the 2017 equivalents. Target = a random
+    caption of the image (reference :97-102)."""
+
+    def __init__(self, root: str, split: str, rng=None):
+        import json
+        import random as _random
+
+        self._rng = rng or _random
+        if split.lower() == "train":
+            ann = os.path.join(root, "annotations_trainval2014/annotations/captions_train2014.json")
+            image_dir = os.path.join(root, "train2014/train2014")
+        else:
+            ann = os.path.join(root, "annotations_trainval2017/annotations/captions_train2017.json")
+            image_dir = os.path.join(root, "val2017/val2017")
+        with open(ann) as f:
+            all_annotations = json.load(f)
+        data = {}
+        for item in all_annotations["images"]:
+            data[item["id"]] = {"image": os.path.join(image_dir, item["file_name"]),
+                                "captions": []}
+        for item in all_annotations["annotations"]:
+            data[item["image_id"]]["captions"].append(item["caption"])
+        self.entries = list(data.values())
+
+    def __len__(self) -> int:
+        return len(self.entries)
+
+    def get_image_data(self, index: int) -> bytes:
+        with open(self.entries[index]["image"], "rb") as f:
+            return f.read()
+
+    def get_target(self, index: int) -> str:
+        caps = self.entries[index]["captions"]
+        return self._rng.choice(caps) if caps else ""
+
+    def get_targets(self):
+        return None

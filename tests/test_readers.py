@@ -170,3 +170,71 @@ def test_imagenet22k_with_multicrop_transform(in22k_tree):
     assert len(out["global_crops"]) == 2
     assert out["global_crops"][0].shape == (3, 32, 32)
     assert len(out["local_crops"]) == 2
+
+
+# ------------------------------- ADE20K / Coco -------------------------------
+
+
+def test_ade20k_reader_and_dataset(tmp_path):
+    from PIL import Image
+
+    root = tmp_path / "ade"
+    (root / "images").mkdir(parents=True)
+    (root / "annotations").mkdir()
+    names = ["training/a_0001.jpg", "training/a_0002.jpg"]
+    for i, n in enumerate(names):
+        (root / "images" / n).parent.mkdir(parents=True, exist_ok=True)
+        (root / "images" / n).write_bytes(_jpeg_bytes(COLORS[i]))
+        mask = Image.fromarray(np.full((32, 32), i + 1, dtype=np.uint8))
+        seg = root / "annotations" / (os.path.splitext(n)[0] + ".png")
+        seg.parent.mkdir(parents=True, exist_ok=True)
+        mask.save(seg)
+    (root / "ADE20K_object150_train.txt").write_text("\n".join(names))
+
+    from dinov3_amd.data.readers import ADE20KReader
+
+    r = ADE20KReader(str(root), "train")
+    assert len(r) == 2
+    img = decode_image_bytes(r.get_image_data(0))
+    _check_color(img, COLORS[0])
+    t = r.get_target(1)
+    assert t.shape == (32, 32) and int(t[0, 0]) == 2
+
+    ds = make_dataset(dataset_str=f"ADE20K:split=TRAIN:root={root}")
+    img2, t2 = ds[0]
+    assert img2.shape == (3, 32, 32)
+    assert t2.shape == (32, 32)
+
+
+def test_coco_captions_reader_and_dataset(tmp_path):
+    import json
+
+    root = tmp_path / "coco"
+    img_dir = root / "train2014" / "train2014"
+    img_dir.mkdir(parents=True)
+    ann_dir = root / "annotations_trainval2014" / "annotations"
+    ann_dir.mkdir(parents=True)
+    (img_dir / "img1.jpg").write_bytes(_jpeg_bytes(COLORS[0]))
+    (img_dir / "img2.jpg").write_bytes(_jpeg_bytes(COLORS[1]))
+    ann = {
+        "images": [{"id": 10, "file_name": "img1.jpg"}, {"id": 20, "file_name": "img2.jpg"}],
+        "annotations": [
+            {"image_id": 10, "caption": "a red square"},
+            {"image_id": 10, "caption": "very red"},
+            {"image_id": 20, "caption": "a green square"},
+        ],
+    }
+    (ann_dir / "captions_train2014.json").write_text(json.dumps(ann))
+
+    from dinov3_amd.data.readers import CocoCaptionsReader
+
+    r = CocoCaptionsReader(str(root), "train")
+    assert len(r) == 2
+    _check_color(decode_image_bytes(r.get_image_data(0)), COLORS[0])
+    assert "red" in r.get_target(0)
+    assert r.get_target(1) == "a green square"
+
+    ds = make_dataset(dataset_str=f"CocoCaptions:split=TRAIN:root={root}")
+    img, cap = ds[1]
+    assert img.shape == (3, 32, 32)
+    assert isinstance(cap, str) and "green" in cap
