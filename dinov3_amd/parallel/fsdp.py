@@ -172,18 +172,27 @@ class ShardedEngine:
                     self._launch_rs(b)
             b["pending"] = 0
         cm = None
-        if deferred:
-            from torch.distributed.distributed_c10d import _coalescing_manager
+        if deferred and getattr(self, "_coalesce_ok", True):
+            try:
+                from torch.distributed.distributed_c10d import _coalescing_manager
 
-            device = deferred[0]["grad_shard"].device
-            with _coalescing_manager(self.group, device, async_ops=True) as cm:
-                for b in deferred:
-                    full = b["grad_flat"]
-                    if full.dtype != self.comm_dtype:
-                        full = full.to(self.comm_dtype)
-                    b["rs_buf"] = full
-                    dist.reduce_scatter_tensor(b["grad_shard"], full,
-                                               op=dist.ReduceOp.SUM, group=self.group)
+                device = deferred[0]["grad_shard"].device
+                with _coalescing_manager(self.group, device, async_ops=True) as cm:
+                    for b in deferred:
+                        full = b["grad_flat"]
+                        if full.dtype != self.comm_dtype:
+                            full = full.to(self.comm_dtype)
+                        b["rs_buf"] = full
+                        dist.reduce_scatter_tensor(b["grad_shard"], full,
+                                                   op=dist.ReduceOp.SUM, group=self.group)
+                deferred = []
+            except Exception as e:  # never let a grouped-launch quirk kill the run
+                logger.warning("coalesced reduce-scatter failed (%s); "
+                               "falling back to per-bucket launches", e)
+                self._coalesce_ok = False
+                cm = None
+        for b in deferred:  # non-coalescing fallback
+            self._launch_rs(b)
         for b in self.buckets:
             if b["rs_work"] is not None:
                 b["rs_work"].wait()
@@ -237,21 +246,26 @@ class ShardedEngine:
             self._step_eager(lr, weight_decay, last_layer_lr, clip_scales)
         # republish updated param shards: ONE coalesced all-gather group
         # launch on RCCL (the wait is stream-side, not a host sync)
-        if self._is_nccl:
-            from torch.distributed.distributed_c10d import _coalescing_manager
+        if self._is_nccl and getattr(self, "_coalesce_ok", True):
+            try:
+                from torch.distributed.distributed_c10d import _coalescing_manager
 
-            with _coalescing_manager(self.group, device, async_ops=True) as cm:
-                for b in self.buckets:
-                    dist.all_gather_into_tensor(b["flat"], b["param_shard"], group=self.group)
-            cm.wait()
-        else:
-            works = []
-            for b in self.buckets:
-                w = _all_gather(b["flat"], b["param_shard"], self.group)
-                if w is not None:
-                    works.append(w)
-            for w in works:
-                w.wait()
+                with _coalescing_manager(self.group, device, async_ops=True) as cm:
+                    for b in self.buckets:
+                        dist.all_gather_into_tensor(b["flat"], b["param_shard"], group=self.group)
+                cm.wait()
+                return
+            except Exception as e:
+                logger.warning("coalesced all-gather failed (%s); "
+                               "falling back to per-bucket launches", e)
+                self._coalesce_ok = False
+        works = []
+        for b in self.buckets:
+            w = _all_gather(b["flat"], b["param_shard"], self.group)
+            if w is not None:
+                works.append(w)
+        for w in works:
+            w.wait()
 
     def _clip_tensor(self, clip_scales, device) -> torch.Tensor:
         if clip_scales is None:
