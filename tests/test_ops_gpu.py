@@ -36,7 +36,7 @@ def test_mfma_layout_probe(ops):
     _assert_close(C, ref, atol=0.05, what="mfma probe")
 
 
-@pytest.mark.parametrize("rows,D", [(64, 1024), (197, 384), (1000, 4096), (33, 256)])
+@pytest.mark.parametrize("rows,D", [(64, 1024), (197, 384), (1000, 4096), (33, 256), (130, 1536)])
 def test_layernorm_fwd_bwd(ops, rows, D):
     torch.manual_seed(0)
     x = torch.randn(rows, D, device=DEV).bfloat16().requires_grad_(True)
