@@ -29,6 +29,12 @@ from ..layers import (
 
 logger = logging.getLogger("dinov3")
 
+# DINOV3_NAN_CHECK=1: per-block non-finite activation sanitizer (debug mode;
+# the SURVEY §5 compute-sanitizer analogue — costs a device sync per block)
+import os as _os
+
+_NAN_CHECK = _os.environ.get("DINOV3_NAN_CHECK", "0") == "1"
+
 
 class DinoVisionTransformer(nn.Module):
     def __init__(
@@ -206,6 +212,10 @@ class DinoVisionTransformer(nn.Module):
         else:
             for i, block in enumerate(self.blocks):
                 flat = block.forward_flat(flat, metas, plan, i)
+                if _NAN_CHECK and not torch.isfinite(flat).all():
+                    raise FloatingPointError(
+                        f"non-finite activations after block {i} "
+                        f"(DINOV3_NAN_CHECK sanitizer)")
         tokens = uncat_with_shapes(flat, shapes, counts)
 
         output = []

@@ -157,3 +157,28 @@ def test_gram_ema_teacher_mode(smoke_cfg):
     batch = _synthetic_batch(cfg)
     loss, metrics = model(batch, teacher_temp=0.07, iteration=0)
     assert "gram_loss" in metrics and torch.isfinite(loss)
+
+
+def test_nan_check_sanitizer(monkeypatch):
+    """DINOV3_NAN_CHECK=1 raises at the first block producing non-finites."""
+    import importlib
+
+    import torch
+
+    import dinov3_amd.models.vision_transformer as vt
+
+    monkeypatch.setenv("DINOV3_NAN_CHECK", "1")
+    importlib.reload(vt)
+    try:
+        model = vt.vit_small(patch_size=16)
+        model.eval()
+        with torch.no_grad():
+            # healthy input passes
+            model([torch.randn(1, 3, 64, 64)], [None])
+            # poison a weight -> the sanitizer must fire with the block index
+            model.blocks[2].mlp.fc1.weight.data.fill_(float("inf"))
+            with pytest.raises(FloatingPointError, match="after block 2"):
+                model([torch.randn(1, 3, 64, 64)], [None])
+    finally:
+        monkeypatch.delenv("DINOV3_NAN_CHECK")
+        importlib.reload(vt)
