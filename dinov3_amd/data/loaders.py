@@ -107,6 +107,12 @@ class CombinedDataLoader:
     def __len__(self):
         return sum(len(dl) for dl in self.loaders)
 
+    def set_epoch(self, epoch: int) -> None:
+        for dl in self.loaders:
+            sampler = getattr(dl, "sampler", None)
+            if sampler is not None and hasattr(sampler, "set_epoch"):
+                sampler.set_epoch(epoch)
+
 
 def make_data_loader(
     *,

@@ -305,8 +305,11 @@ def do_train(cfg, model: SSLMetaArch, resume: bool = True, max_iterations: int =
     def infinite_batches():
         epoch = 0
         while True:
-            if hasattr(data_loader.sampler, "set_epoch"):
-                data_loader.sampler.set_epoch(epoch)
+            sampler = getattr(data_loader, "sampler", None)
+            if sampler is not None and hasattr(sampler, "set_epoch"):
+                sampler.set_epoch(epoch)
+            elif hasattr(data_loader, "set_epoch"):  # CombinedDataLoader
+                data_loader.set_epoch(epoch)
             yield from data_loader
             epoch += 1
 

@@ -270,3 +270,30 @@ def test_main_multidist_dispatch(tmp_path):
         "--no-resume", "--max-iterations", "2",
     ])
     assert "total_loss" in result
+
+
+def test_do_train_multi_resolution_schedule(tmp_path):
+    """End-to-end: crops.global_crops_size as a LIST engages the
+    CombinedDataLoader (high-res-adapt mechanism, reference train.py:718-769)
+    and steps batches of different resolutions through the same model."""
+    import types
+
+    from dinov3_amd.configs import setup_config
+    from dinov3_amd.train.ssl_meta_arch import SSLMetaArch
+    from dinov3_amd.train.train import do_train
+
+    args = types.SimpleNamespace(
+        config_file="dinov3_amd/configs/train/vits_smoke.yaml",
+        opts=[], output_dir="")
+    cfg = setup_config(args, apply_scaling=False)
+    cfg.train.output_dir = str(tmp_path)
+    cfg.crops.global_crops_size = [112, 64]
+    cfg.crops.local_crops_size = [48, 32]
+    cfg.crops.global_local_crop_pairs_ratios = [0.5, 0.5]
+    cfg.train.batch_size_per_gpu = 2
+    cfg.crops.local_crops_number = 2
+    cfg.checkpointing.period = 0
+    metrics = do_train(cfg, SSLMetaArch(cfg), resume=False, max_iterations=4)
+    import math
+
+    assert math.isfinite(metrics["total_loss"])
