@@ -297,3 +297,59 @@ def test_do_train_multi_resolution_schedule(tmp_path):
     import math
 
     assert math.isfinite(metrics["total_loss"])
+
+
+def test_do_train_gram_refresh_cadence(tmp_path, smoke_cfg):
+    """End-to-end: gram loss enabled with a 2-iteration refresh cadence —
+    the trainer's refresh/ema-load logic runs, loss stays finite."""
+    import copy
+    import math
+
+    from dinov3_amd.train.ssl_meta_arch import SSLMetaArch
+    from dinov3_amd.train.train import do_train
+
+    cfg = copy.deepcopy(smoke_cfg)
+    cfg.train.output_dir = str(tmp_path)
+    cfg.checkpointing.period = 0
+    cfg.gram.use_loss = True
+    cfg.gram.remove_neg = True
+    cfg.gram.ema_teacher = False
+    cfg.gram.it_load_ema_teacher = 0  # gram backbone seeded from EMA teacher
+    cfg.gram.rep_update = True
+    cfg.gram.update_frequency = 2
+    cfg.gram.it_first_update = 1
+    cfg.train.batch_size_per_gpu = 2
+    cfg.crops.local_crops_number = 2
+    metrics = do_train(cfg, SSLMetaArch(cfg), resume=False, max_iterations=4)
+    assert math.isfinite(metrics["total_loss"])
+    assert any(k.startswith("gram") for k in metrics), metrics.keys()
+
+
+def test_do_train_7b_flavored_options(tmp_path):
+    """End-to-end train with the ViT-7b option set on a tiny model: RMSNorm,
+    SwiGLU (aligned-64), masked k-bias, storage tokens, untied norms."""
+    import math
+    import types
+
+    from dinov3_amd.configs import setup_config
+    from dinov3_amd.train.ssl_meta_arch import SSLMetaArch
+    from dinov3_amd.train.train import do_train
+
+    args = types.SimpleNamespace(
+        config_file="dinov3_amd/configs/train/vits_smoke.yaml",
+        opts=[
+            "student.norm_layer=rmsnorm",
+            "student.ffn_layer=swiglu64",
+            "student.mask_k_bias=true",
+            "student.n_storage_tokens=4",
+            "student.untie_cls_and_patch_norms=true",
+            "student.untie_global_and_local_cls_norm=true",
+            "train.batch_size_per_gpu=2",
+            "crops.local_crops_number=2",
+            "checkpointing.period=0",
+        ],
+        output_dir="")
+    cfg = setup_config(args, apply_scaling=False)
+    cfg.train.output_dir = str(tmp_path)
+    metrics = do_train(cfg, SSLMetaArch(cfg), resume=False, max_iterations=3)
+    assert math.isfinite(metrics["total_loss"])
