@@ -159,10 +159,77 @@ def normalize(img: torch.Tensor, mean: Sequence[float], std: Sequence[float]) ->
 
 
 def make_eval_transform(resize_size: int = 256, crop_size: int = 224,
-                        mean=(0.485, 0.456, 0.406), std=(0.229, 0.224, 0.225)):
+                        mean=(0.485, 0.456, 0.406), std=(0.229, 0.224, 0.225),
+                        resize_square: bool = False, resize_large_side: bool = False,
+                        mode: str = "bicubic"):
+    """Eval preset (reference transforms.py:106-131): resize (short side,
+    square, or long side) -> optional center crop -> normalize."""
+    assert not (resize_square and resize_large_side)
+
     def _t(img: torch.Tensor) -> torch.Tensor:
-        img = resize(img, resize_size)
-        img = center_crop(img, crop_size)
+        if resize_square:
+            img = torch.nn.functional.interpolate(
+                img.unsqueeze(0), size=(resize_size, resize_size), mode=mode,
+                align_corners=False, antialias=True).squeeze(0).clamp(0, 1)
+        elif resize_large_side:
+            c, h, w = img.shape
+            s = resize_size / max(h, w)
+            nh, nw = max(1, round(h * s)), max(1, round(w * s))
+            img = torch.nn.functional.interpolate(
+                img.unsqueeze(0), size=(nh, nw), mode=mode,
+                align_corners=False, antialias=True).squeeze(0).clamp(0, 1)
+        else:
+            img = resize(img, resize_size, mode)
+        if crop_size:
+            img = center_crop(img, crop_size)
         return normalize(img, mean, std)
 
     return _t
+
+
+def make_classification_eval_transform(resize_size: int = 256, crop_size: int = 224,
+                                       mean=(0.485, 0.456, 0.406),
+                                       std=(0.229, 0.224, 0.225)):
+    """Torchvision-style classification eval preset (reference
+    transforms.py:134-150)."""
+    return make_eval_transform(resize_size, crop_size, mean, std)
+
+
+def make_classification_train_transform(crop_size: int = 224, hflip_prob: float = 0.5,
+                                        mean=(0.485, 0.456, 0.406),
+                                        std=(0.229, 0.224, 0.225)):
+    """Torchvision-style classification train preset (reference
+    transforms.py:66-80): RandomResizedCrop -> flip -> normalize."""
+    import random as _random
+
+    def _t(img: torch.Tensor) -> torch.Tensor:
+        img = random_resized_crop(img, crop_size, (0.08, 1.0))
+        if hflip_prob > 0 and _random.random() < hflip_prob:
+            img = hflip(img)
+        return normalize(img, mean, std)
+
+    return _t
+
+
+def voc2007_classification_target_transform(label, n_categories: int = 20) -> torch.Tensor:
+    """Multi-label one-hot from a VOC-style label with .instances
+    (reference transforms.py:153-157)."""
+    one_hot = torch.zeros(n_categories, dtype=torch.long)
+    for instance in label.instances:
+        one_hot[instance.category_id] = 1
+    return one_hot
+
+
+def imaterialist_classification_target_transform(label, n_categories: int = 294) -> torch.Tensor:
+    one_hot = torch.zeros(n_categories, dtype=torch.long)
+    one_hot[label.attributes] = 1
+    return one_hot
+
+
+def get_target_transform(dataset_str: str):
+    """Dataset-string -> target transform (reference transforms.py:166-170)."""
+    if "VOC2007" in dataset_str:
+        return voc2007_classification_target_transform
+    if "IMaterialist" in dataset_str:
+        return imaterialist_classification_target_transform
+    return None

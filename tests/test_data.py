@@ -235,3 +235,41 @@ def test_canonical_decode_sizes():
         out = dec(torch.rand(3, h, w))
         assert out.shape == (3, 64, 64)
         assert out.dtype == torch.uint8
+
+
+def test_transform_presets():
+    import torch
+
+    from dinov3_amd.data.transforms import (
+        get_target_transform,
+        imaterialist_classification_target_transform,
+        make_classification_eval_transform,
+        make_classification_train_transform,
+        make_eval_transform,
+        voc2007_classification_target_transform,
+    )
+
+    img = torch.rand(3, 100, 80)
+    assert make_classification_eval_transform()(img).shape == (3, 224, 224)
+    assert make_classification_train_transform(crop_size=64)(img).shape == (3, 64, 64)
+    assert make_eval_transform(resize_size=64, crop_size=0,
+                               resize_square=True)(img).shape == (3, 64, 64)
+    big = make_eval_transform(resize_size=64, crop_size=0, resize_large_side=True)(img)
+    assert max(big.shape[1:]) == 64
+
+    class _Inst:
+        def __init__(self, cid):
+            self.category_id = cid
+
+    class _Label:
+        instances = [_Inst(3), _Inst(7)]
+
+    one_hot = voc2007_classification_target_transform(_Label())
+    assert one_hot.sum() == 2 and one_hot[3] == 1 and one_hot[7] == 1
+
+    class _IMat:
+        attributes = [1, 5]
+
+    assert imaterialist_classification_target_transform(_IMat()).sum() == 2
+    assert get_target_transform("VOC2007:split=TRAIN") is voc2007_classification_target_transform
+    assert get_target_transform("ImageNet:split=TRAIN") is None
