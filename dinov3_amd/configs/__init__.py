@@ -5,6 +5,7 @@ from .config import (
     load_yaml,
     setup_config,
     setup_job,
+    exit_job,
     write_config,
 )
 
@@ -14,6 +15,7 @@ __all__ = [
     "load_yaml",
     "setup_config",
     "setup_job",
+    "exit_job",
     "apply_scaling_rules_to_cfg",
     "write_config",
 ]

@@ -171,6 +171,20 @@ def setup_config(args, strict_cfg: bool = True, apply_scaling: bool = True) -> D
     return cfg
 
 
+def exit_job(distributed_enabled: bool = True, logging_enabled: bool = True) -> None:
+    """Teardown counterpart of setup_job (reference configs parity: the
+    run/init.py job_context calls exit_job on the way out)."""
+    import logging as _logging
+
+    from .. import parallel
+
+    if distributed_enabled:
+        parallel.destroy()
+    if logging_enabled:
+        for h in _logging.getLogger("dinov3").handlers:
+            h.flush()
+
+
 def setup_job(output_dir: Optional[str] = None, seed: int = 0, distributed_enabled: bool = True,
               logging_enabled: bool = True) -> None:
     """Job context: logging + RNG seeding (+ torch.distributed if launched via torchrun)."""
