@@ -71,18 +71,6 @@ def _subset_rows(metas: List[GroupMeta], keep_ratio: float, device) -> Tuple[tor
     return torch.cat(rows), new_metas, torch.cat(scales)
 
 
-def _norm_gathered(norm, flat, rows):
-    """norm(flat[rows]) — the gather fuses into the LayerNorm kernel on GPU
-    (ops.layer_norm_gather); plain gather+norm for RMSNorm/CPU."""
-    from .norms import LayerNorm as _LN
-    from ..ops import layer_norm_gather
-    from ..ops.row_ops import gather_rows
-
-    if isinstance(norm, _LN):
-        return layer_norm_gather(flat, rows, norm.weight, norm.bias, norm.eps)
-    return norm(gather_rows(flat, rows))
-
-
 class SelfAttentionBlock(nn.Module):
     def __init__(
         self,
@@ -168,27 +156,27 @@ class SelfAttentionBlock(nn.Module):
             rows1, metas1, scale1 = plan.take(2 * block_idx)
         else:
             rows1, metas1, scale1 = _subset_rows(metas, keep_ratio, flat.device)
-        normed = _norm_gathered(self.norm1, flat, rows1)
+        sub = gather_rows(flat, rows1)
         if fa:
             if not inplace_ok:
                 flat = flat.clone()
-            res = self.attn.forward_flat(normed, metas1, skip_proj_bias=True)
+            res = self.attn.forward_flat(self.norm1(sub), metas1, skip_proj_bias=True)
             flat = ls_scatter_add_rows(flat, rows1, res, self.ls1.gamma,
                                        self.attn.proj.bias, scale1)
         else:
-            res = self.ls1(self.attn.forward_flat(normed, metas1))
+            res = self.ls1(self.attn.forward_flat(self.norm1(sub), metas1))
             flat = scatter_add_rows(flat, rows1, res, scale1)
 
         if plan is not None:
             rows2, _, scale2 = plan.take(2 * block_idx + 1)
         else:
             rows2, _, scale2 = _subset_rows(metas, keep_ratio, flat.device)
-        normed = _norm_gathered(self.norm2, flat, rows2)
+        sub = gather_rows(flat, rows2)
         if fm:
-            res = self.mlp(normed, skip_out_bias=True)
+            res = self.mlp(self.norm2(sub), skip_out_bias=True)
             return ls_scatter_add_rows(flat, rows2, res, self.ls2.gamma,
                                        self.mlp.fc2.bias, scale2)
-        res = self.ls2(self.mlp(normed))
+        res = self.ls2(self.mlp(self.norm2(sub)))
         return scatter_add_rows(flat, rows2, res, scale2)
 
     # ------------------------------------------------------------------

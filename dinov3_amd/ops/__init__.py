@@ -19,7 +19,7 @@ import os
 
 import torch
 
-from .layernorm import layer_norm, layer_norm_gather, rms_norm
+from .layernorm import layer_norm, rms_norm
 from .fmha import fmha
 from .rope import rope_apply
 from .l2norm import l2_normalize
@@ -32,7 +32,6 @@ from .flat_attention import flat_multi_fmha
 
 __all__ = [
     "layer_norm",
-    "layer_norm_gather",
     "rms_norm",
     "fmha",
     "rope_apply",

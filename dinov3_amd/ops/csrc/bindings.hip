@@ -14,10 +14,10 @@
 // ---- extern launchers (defined in the .hip kernel TUs) ----
 template <typename T>
 void launch_layernorm_fwd(const T*, const T*, const T*, T*, float*, float*, long, int,
-                          float, hipStream_t, const long* = nullptr);
+                          float, hipStream_t);
 template <typename T>
 void launch_layernorm_bwd(const T*, const T*, const T*, const float*, const float*, T*,
-                          float*, float*, long, int, hipStream_t, const long* = nullptr);
+                          float*, float*, long, int, hipStream_t);
 template <typename T>
 void launch_rmsnorm_fwd(const T*, const T*, T*, float*, long, int, float, hipStream_t);
 template <typename T>
@@ -199,53 +199,6 @@ std::vector<torch::Tensor> layernorm_bwd(torch::Tensor dy, torch::Tensor x, torc
         current_stream());
   });
   return {dx, dw.sum(0).to(x.scalar_type()), db.sum(0).to(x.scalar_type())};
-}
-
-// Fused drop-path gather + LayerNorm: y[r] = LN(flat[idx[r]]).
-std::vector<torch::Tensor> layernorm_gather_fwd(torch::Tensor flat, torch::Tensor idx,
-                                                torch::Tensor w, torch::Tensor b,
-                                                double eps) {
-  CHECK_INPUT(flat);
-  CHECK_INPUT(idx);
-  const int D = flat.size(-1);
-  const long M = idx.numel();
-  TORCH_CHECK(idx.scalar_type() == at::ScalarType::Long, "idx must be int64");
-  auto y = torch::empty({M, (long)D}, flat.options());
-  auto mean = torch::empty({M}, flat.options().dtype(torch::kFloat));
-  auto rstd = torch::empty({M}, flat.options().dtype(torch::kFloat));
-  DISPATCH_FLOAT_BF16(flat.scalar_type(), "layernorm_gather_fwd", [&] {
-    launch_layernorm_fwd<scalar_t>(
-        (const scalar_t*)flat.data_ptr(), (const scalar_t*)w.data_ptr(),
-        (const scalar_t*)b.data_ptr(), (scalar_t*)y.data_ptr(), mean.data_ptr<float>(),
-        rstd.data_ptr<float>(), M, D, (float)eps, current_stream(),
-        idx.data_ptr<long>());
-  });
-  return {y, mean, rstd};
-}
-
-// Backward of the fused gather-LN: dflat is a zeros [R, D] with LN-dx
-// scattered to the idx rows (disjoint), dgamma/dbeta as usual.
-std::vector<torch::Tensor> layernorm_scatter_bwd(torch::Tensor dy, torch::Tensor flat,
-                                                 torch::Tensor idx, torch::Tensor w,
-                                                 torch::Tensor mean, torch::Tensor rstd) {
-  CHECK_INPUT(dy);
-  CHECK_INPUT(flat);
-  const int D = flat.size(-1);
-  const long M = idx.numel();
-  TORCH_CHECK(D % 8 == 0 && D <= 1536,
-              "layernorm_scatter_bwd needs the wave kernel (D%8==0, D<=1536)");
-  auto dflat = torch::zeros_like(flat);
-  // LN_SHADOWS=32 shadow accumulators (must match norms.hip)
-  auto dw = torch::zeros({32, D}, flat.options().dtype(torch::kFloat));
-  auto db = torch::zeros({32, D}, flat.options().dtype(torch::kFloat));
-  DISPATCH_FLOAT_BF16(flat.scalar_type(), "layernorm_scatter_bwd", [&] {
-    launch_layernorm_bwd<scalar_t>(
-        (const scalar_t*)dy.data_ptr(), (const scalar_t*)flat.data_ptr(),
-        (const scalar_t*)w.data_ptr(), mean.data_ptr<float>(), rstd.data_ptr<float>(),
-        (scalar_t*)dflat.data_ptr(), dw.data_ptr<float>(), db.data_ptr<float>(), M, D,
-        current_stream(), idx.data_ptr<long>());
-  });
-  return {dflat, dw.sum(0).to(flat.scalar_type()), db.sum(0).to(flat.scalar_type())};
 }
 
 std::vector<torch::Tensor> rmsnorm_fwd(torch::Tensor x, torch::Tensor w, double eps) {
@@ -1091,8 +1044,6 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, mod) {
   mod.def("blaslt_gemm_bias", &blaslt_gemm_bias);
   mod.def("blaslt_probe_epilogue", &blaslt_probe_epilogue);
   mod.def("layernorm_fwd", &layernorm_fwd);
-  mod.def("layernorm_gather_fwd", &layernorm_gather_fwd);
-  mod.def("layernorm_scatter_bwd", &layernorm_scatter_bwd);
   mod.def("layernorm_bwd", &layernorm_bwd);
   mod.def("rmsnorm_fwd", &rmsnorm_fwd);
   mod.def("rmsnorm_bwd", &rmsnorm_bwd);

@@ -54,11 +54,11 @@ template <typename T, bool V8>
 __global__ void layernorm_fwd_kernel(
     const T* __restrict__ x, const T* __restrict__ w, const T* __restrict__ b,
     T* __restrict__ y, float* __restrict__ mean_out, float* __restrict__ rstd_out,
-    long rows, int D, float eps, const long* __restrict__ idx = nullptr) {
+    long rows, int D, float eps) {
   constexpr int EW = V8 ? 8 : 1;
   __shared__ float red[16];
   for (long row = blockIdx.x; row < rows; row += gridDim.x) {
-    const T* xr = x + (idx != nullptr ? idx[row] : row) * (long)D;
+    const T* xr = x + row * (long)D;
     T* yr = y + row * (long)D;
     float s = 0.f, s2 = 0.f;
     ROW_LOOP(i) {
@@ -170,7 +170,7 @@ __global__ __launch_bounds__(256) void layernorm_bwd_wave_kernel(
     const T* __restrict__ dy, const T* __restrict__ x, const T* __restrict__ w,
     const float* __restrict__ mean, const float* __restrict__ rstd,
     T* __restrict__ dx, float* __restrict__ dw, float* __restrict__ db,
-    long rows, int D, const long* __restrict__ idx = nullptr) {
+    long rows, int D) {
   constexpr int NW = 4;  // waves per 256-thread block
   const int wid = threadIdx.x / 64;
   const int lane = threadIdx.x & 63;
@@ -183,9 +183,8 @@ __global__ __launch_bounds__(256) void layernorm_bwd_wave_kernel(
   }
   float g[CHUNKS][8], v[CHUNKS][8], wv[CHUNKS][8];
   for (long row = (long)blockIdx.x * NW + wid; row < rows; row += (long)gridDim.x * NW) {
-    const long srow = idx != nullptr ? idx[row] : row;
     const T* dyr = dy + row * (long)D;
-    const T* xr = x + srow * (long)D;
+    const T* xr = x + row * (long)D;
     const float m = mean[row], r = rstd[row];
     float sum_dyw = 0.f, sum_dyw_xhat = 0.f;
 #pragma unroll
@@ -212,7 +211,7 @@ __global__ __launch_bounds__(256) void layernorm_bwd_wave_kernel(
       sum_dyw_xhat += __shfl_xor(sum_dyw_xhat, off, 64);
     }
     const float inv_d = 1.0f / D;
-    T* dxr = dx + srow * (long)D;  // scatter when idx is set (rows disjoint)
+    T* dxr = dx + row * (long)D;
 #pragma unroll
     for (int c = 0; c < CHUNKS; ++c) {
       const int i = (c * 64 + lane) * 8;
@@ -399,18 +398,17 @@ __global__ void l2norm_bwd_kernel(
 
 template <typename T>
 void launch_layernorm_fwd(const T* x, const T* w, const T* b, T* y, float* mean,
-                          float* rstd, long rows, int D, float eps, hipStream_t stream,
-                          const long* idx = nullptr) {
+                          float* rstd, long rows, int D, float eps, hipStream_t stream) {
   int grid = (int)min(rows, (long)8192);
   DISPATCH_V8(D, hipLaunchKernelGGL(HIP_KERNEL_NAME(layernorm_fwd_kernel<T, V8>), dim3(grid),
                                      dim3(norm_block_for(D)), 0, stream, x, w, b, y, mean, rstd,
-                                     rows, D, eps, idx));
+                                     rows, D, eps));
 }
 
 template <typename T>
 void launch_layernorm_bwd(const T* dy, const T* x, const T* w, const float* mean,
                           const float* rstd, T* dx, float* dw, float* db, long rows,
-                          int D, hipStream_t stream, const long* idx = nullptr) {
+                          int D, hipStream_t stream) {
   if (D % 8 == 0 && D <= 1536) {
     // wave-per-row: one read of dy/x, no block barriers in the row loop
     int grid = (int)min((rows + 3) / 4, (long)2048);
@@ -418,18 +416,17 @@ void launch_layernorm_bwd(const T* dy, const T* x, const T* w, const float* mean
     if (D <= 512)
       hipLaunchKernelGGL(HIP_KERNEL_NAME(layernorm_bwd_wave_kernel<T, 1>), dim3(grid),
                          dim3(256), shmem, stream, dy, x, w, mean, rstd, dx, dw, db,
-                         rows, D, idx);
+                         rows, D);
     else if (D <= 1024)
       hipLaunchKernelGGL(HIP_KERNEL_NAME(layernorm_bwd_wave_kernel<T, 2>), dim3(grid),
                          dim3(256), shmem, stream, dy, x, w, mean, rstd, dx, dw, db,
-                         rows, D, idx);
+                         rows, D);
     else
       hipLaunchKernelGGL(HIP_KERNEL_NAME(layernorm_bwd_wave_kernel<T, 3>), dim3(grid),
                          dim3(256), shmem, stream, dy, x, w, mean, rstd, dx, dw, db,
-                         rows, D, idx);
+                         rows, D);
     return;
   }
-  (void)idx;  // the block-per-row fallback never runs the fused path
   int grid = (int)min(rows, (long)2048);
   size_t shmem = (16 + 2 * (size_t)D) * sizeof(float);
   DISPATCH_V8(D, hipLaunchKernelGGL(HIP_KERNEL_NAME(layernorm_bwd_kernel<T, V8>), dim3(grid),
@@ -476,10 +473,10 @@ void launch_l2norm_bwd(const T* dy, const T* y, const float* s, T* dx, long rows
 // explicit instantiations used by bindings.cpp
 #define INSTANTIATE_NORMS(T)                                                              \
   template void launch_layernorm_fwd<T>(const T*, const T*, const T*, T*, float*, float*, \
-                                        long, int, float, hipStream_t, const long*);      \
+                                        long, int, float, hipStream_t);                   \
   template void launch_layernorm_bwd<T>(const T*, const T*, const T*, const float*,       \
                                         const float*, T*, float*, float*, long, int,      \
-                                        hipStream_t, const long*);                        \
+                                        hipStream_t);                                     \
   template void launch_rmsnorm_fwd<T>(const T*, const T*, T*, float*, long, int, float,   \
                                       hipStream_t);                                       \
   template void launch_rmsnorm_bwd<T>(const T*, const T*, const T*, const float*, T*,     \

@@ -52,43 +52,6 @@ class _RMSNormFn(torch.autograd.Function):
         return dx, dw, None
 
 
-class _LayerNormGatherFn(torch.autograd.Function):
-    """Fused drop-path gather + LayerNorm: y = LN(flat[idx]) without
-    materializing the gathered rows; backward scatters LN-dx into a zeroed
-    [R, D] gradient (rows disjoint — no atomics)."""
-
-    @staticmethod
-    def forward(ctx, flat, idx, weight, bias, eps):
-        from . import hip_ops
-
-        y, mean, rstd = hip_ops().layernorm_gather_fwd(flat, idx, weight, bias, eps)
-        ctx.save_for_backward(flat, idx, weight, mean, rstd)
-        return y
-
-    @staticmethod
-    def backward(ctx, dy):
-        from . import hip_ops
-
-        flat, idx, weight, mean, rstd = ctx.saved_tensors
-        dflat, dw, db = hip_ops().layernorm_scatter_bwd(
-            dy.contiguous(), flat, idx, weight, mean, rstd)
-        return dflat, None, dw, db, None
-
-
-def layer_norm_gather(flat: torch.Tensor, idx: torch.Tensor, weight: torch.Tensor,
-                      bias: Optional[torch.Tensor], eps: float = 1e-6) -> torch.Tensor:
-    """LN(flat[idx]) — fused on GPU for D%8==0, D<=1536 (the wave-kernel
-    envelope); plain gather+layer_norm otherwise and on CPU."""
-    from . import use_hip
-
-    D = flat.shape[-1]
-    if use_hip(flat) and D % 8 == 0 and D <= 1536:
-        if bias is None:
-            bias = torch.zeros_like(weight)
-        return _LayerNormGatherFn.apply(flat.contiguous(), idx, weight, bias, eps)
-    return layer_norm(flat[idx], weight, bias, eps)
-
-
 def layer_norm(x: torch.Tensor, weight: torch.Tensor, bias: Optional[torch.Tensor], eps: float = 1e-6) -> torch.Tensor:
     from . import use_hip
 

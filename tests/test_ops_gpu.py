@@ -412,34 +412,3 @@ def test_fused_residual_block_gpu():
     for n in grads_a:
         scale = grads_a[n].abs().max().item()
         _assert_close(grads_b[n], grads_a[n], atol=0.05 + 0.03 * scale, what=f"grad {n}")
-
-
-@pytest.mark.parametrize("R,M,D", [(500, 350, 1024), (64, 64, 384), (200, 120, 1536)])
-def test_layernorm_gather_fused_matches_composed(ops, R, M, D):
-    """Fused drop-path gather+LN == gather followed by LN, fwd and bwd
-    (including the scattered dflat and dgamma/dbeta)."""
-    torch.manual_seed(17)
-    from dinov3_amd.ops import layer_norm, layer_norm_gather
-
-    flat = torch.randn(R, D, device=DEV).bfloat16().requires_grad_(True)
-    idx = torch.randperm(R, device=DEV)[:M].sort().values
-    w = torch.randn(D, device=DEV).bfloat16().requires_grad_(True)
-    b = torch.randn(D, device=DEV).bfloat16().requires_grad_(True)
-
-    y = layer_norm_gather(flat, idx, w, b, eps=1e-6)
-    flat2 = flat.detach().clone().requires_grad_(True)
-    w2 = w.detach().clone().requires_grad_(True)
-    b2 = b.detach().clone().requires_grad_(True)
-    y2 = layer_norm(flat2[idx], w2, b2, eps=1e-6)
-    _assert_close(y, y2, atol=0.03, what="ln_gather fwd")
-
-    dy = torch.randn_like(y)
-    y.backward(dy)
-    y2.backward(dy)
-    _assert_close(flat.grad, flat2.grad, atol=0.05, what="ln_gather dflat")
-    # untouched rows must stay exactly zero
-    mask = torch.ones(R, dtype=torch.bool, device=DEV)
-    mask[idx] = False
-    assert flat.grad[mask].abs().max().item() == 0.0
-    _assert_close(w.grad, w2.grad, atol=0.3, rtol=2e-2, what="ln_gather dw")
-    _assert_close(b.grad, b2.grad, atol=0.3, rtol=2e-2, what="ln_gather db")
