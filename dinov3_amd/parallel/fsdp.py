@@ -160,6 +160,19 @@ class ShardedEngine:
         b["rs_buf"] = full  # keep alive until work completes
         b["rs_work"] = _reduce_scatter(b["grad_shard"], full, self.group)
 
+    def _end_coalescing_if_stuck(self, device) -> None:
+        """torch's _coalescing_manager is a bare contextmanager: an exception
+        inside its body skips the group-end epilogue and leaves the process
+        group stuck in coalescing mode (every later collective returns
+        IllegalWork). Manually end it so the per-bucket fallback works."""
+        try:
+            from torch.distributed.distributed_c10d import _get_default_group
+
+            pg = self.group if self.group is not None else _get_default_group()
+            pg._end_coalescing(torch.device(device))
+        except Exception:
+            pass
+
     def finalize_backward(self) -> None:
         """Wait for all grad reductions; average (sum -> mean). Deferred
         (small) buckets go out as one coalesced RCCL group launch."""
@@ -191,6 +204,7 @@ class ShardedEngine:
                                "falling back to per-bucket launches", e)
                 self._coalesce_ok = False
                 cm = None
+                self._end_coalescing_if_stuck(device)
         for b in deferred:  # non-coalescing fallback
             self._launch_rs(b)
         for b in self.buckets:
@@ -259,6 +273,7 @@ class ShardedEngine:
                 logger.warning("coalesced all-gather failed (%s); "
                                "falling back to per-bucket launches", e)
                 self._coalesce_ok = False
+                self._end_coalescing_if_stuck(device)
         works = []
         for b in self.buckets:
             w = _all_gather(b["flat"], b["param_shard"], self.group)

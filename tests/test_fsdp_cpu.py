@@ -196,3 +196,41 @@ def test_full_meta_arch_sharded_step_fused_residual():
     fallback math): grad hooks must still fire for proj/fc2 biases whose
     grads now come from the residual op."""
     _run(_fused_meta_arch_worker, 29619)
+
+
+def _fallback_worker(rank, port):
+    """Force the coalesced-launch path on gloo: the grouped collectives are
+    expected to fail there, the engine must log, flip _coalesce_ok and
+    produce correct gradients via the per-bucket fallback."""
+    _init(rank, port)
+    from dinov3_amd.parallel.fsdp import ShardedEngine
+
+    model = _make_model()
+    engine = ShardedEngine(_make_groups(model), align=4, coalesce_below=1 << 30)
+    engine._is_nccl = True  # pretend RCCL so every bucket defers to coalescing
+    # simulate the realistic failure (private torch API unavailable/renamed):
+    # the coalesced launch must fail BEFORE entering the manager
+    import torch.distributed.distributed_c10d as c10d
+
+    def _boom(*a, **k):
+        raise RuntimeError("simulated coalescing unavailability")
+
+    c10d._coalescing_manager = _boom
+    torch.manual_seed(rank)
+    x = torch.randn(4, 16)
+    (model(x) ** 2).mean().backward()
+    engine.finalize_backward()  # grouped launch fails on gloo -> fallback
+    assert engine._coalesce_ok is False
+    # grads were still reduced: shards sum to the cross-rank mean
+    total = sum(float(b["grad_shard"].abs().sum()) for b in engine.buckets)
+    assert total > 0
+    engine.step(lr=0.01, weight_decay=0.0)
+    flat = torch.cat([p.detach().reshape(-1) for p in model.parameters()])
+    gathered = [torch.empty_like(flat) for _ in range(WORLD)]
+    dist.all_gather(gathered, flat)
+    assert torch.equal(gathered[0], gathered[1])
+    dist.destroy_process_group()
+
+
+def test_coalescing_failure_falls_back():
+    _run(_fallback_worker, 29621)
