@@ -350,17 +350,21 @@ class SSLMetaArch(nn.Module):
         reg = out["x_storage_tokens"]         # [2B, R, D]
         ibot_patch = out["x_norm_patchtokens"]  # [2B, P, D]
 
-        buffer = ibot_patch.reshape(-1, ibot_patch.shape[-1])[mask_indices_list]
-        masked_patch_after_head = self.teacher_ibot_head(buffer)
         cls_after_head = self.teacher_dino_head(cls)
-
         cls_centered = self.dino_loss.sinkhorn_knopp_teacher(
             cls_after_head, teacher_temp=teacher_temp,
         ).reshape(n_global_crops, B, -1)
-        masked_patch_centered = self.ibot_patch_loss.sinkhorn_knopp_teacher(
-            masked_patch_after_head, teacher_temp=teacher_temp,
-            n_masked_patches_tensor=n_masked_patches_tensor,
-        )
+        # iBOT disabled (loss_weight 0, e.g. a ConvNeXt student whose conv
+        # grid has no mask-token substitution point): skip the masked-patch
+        # gather/head/sinkhorn entirely
+        masked_patch_centered = None
+        if self.ibot_loss_weight != 0:
+            buffer = ibot_patch.reshape(-1, ibot_patch.shape[-1])[mask_indices_list]
+            masked_patch_after_head = self.teacher_ibot_head(buffer)
+            masked_patch_centered = self.ibot_patch_loss.sinkhorn_knopp_teacher(
+                masked_patch_after_head, teacher_temp=teacher_temp,
+                n_masked_patches_tensor=n_masked_patches_tensor,
+            )
         D = cls.shape[-1]
         return {
             "cls_pre_head": cls.reshape(n_global_crops, B, D),
@@ -384,8 +388,11 @@ class SSLMetaArch(nn.Module):
         l_reg = local_out["x_storage_tokens"]
         l_patch = local_out["x_norm_patchtokens"]
 
-        masked_patches_pre_head = g_patch.reshape(-1, g_patch.shape[-1])[mask_indices_list]
-        global_masked_patch_after_head = self.student_ibot_head(masked_patches_pre_head)
+        masked_patches_pre_head = None
+        global_masked_patch_after_head = None
+        if self.ibot_loss_weight != 0:
+            masked_patches_pre_head = g_patch.reshape(-1, g_patch.shape[-1])[mask_indices_list]
+            global_masked_patch_after_head = self.student_ibot_head(masked_patches_pre_head)
 
         # one DINO-head pass over [global cls | local cls]
         split = g_cls.shape[0]
@@ -491,15 +498,16 @@ class SSLMetaArch(nn.Module):
         loss_dict["koleo_loss"] = koleo_loss
         loss_accumulator = loss_accumulator + self.dino_koleo_loss_weight * koleo_scale * koleo_loss
 
-        ibot_loss = self.ibot_patch_loss.forward_masked(
-            student_global["masked_patch_after_head"],
-            teacher_global["masked_patch_centered"],
-            student_masks_flat=masks,
-            n_masked_patches=mask_indices_list.shape[0],
-            masks_weight=masks_weight,
-        )
-        loss_dict["ibot_loss"] = ibot_loss
-        loss_accumulator = loss_accumulator + self.ibot_loss_weight * ibot_loss
+        if self.ibot_loss_weight != 0:
+            ibot_loss = self.ibot_patch_loss.forward_masked(
+                student_global["masked_patch_after_head"],
+                teacher_global["masked_patch_centered"],
+                student_masks_flat=masks,
+                n_masked_patches=mask_indices_list.shape[0],
+                masks_weight=masks_weight,
+            )
+            loss_dict["ibot_loss"] = ibot_loss
+            loss_accumulator = loss_accumulator + self.ibot_loss_weight * ibot_loss
 
         if self.gram_use_loss and gram_global:
             s_patches = gram_global["student_patches"]
