@@ -273,3 +273,31 @@ def test_transform_presets():
     assert imaterialist_classification_target_transform(_IMat()).sum() == 2
     assert get_target_transform("VOC2007:split=TRAIN") is voc2007_classification_target_transform
     assert get_target_transform("ImageNet:split=TRAIN") is None
+
+
+def test_augmentation_option_matrix():
+    """Exercise the augmentation options the trainer can reach from config:
+    local-crops-as-subwindows, shared color jitter, no flips, gram crops
+    without distortions."""
+    import torch
+
+    from dinov3_amd.data import DataAugmentationDINO
+
+    img = torch.rand(3, 96, 96)
+    for kwargs in (
+        dict(local_crops_subset_of_global_crops=True, patch_size=16),
+        dict(share_color_jitter=True),
+        dict(horizontal_flips=False),
+        dict(gram_teacher_crops_size=64, gram_teacher_no_distortions=True),
+    ):
+        aug = DataAugmentationDINO((0.32, 1.0), (0.05, 0.32), 3,
+                                   global_crops_size=64, local_crops_size=32, **kwargs)
+        out = aug(img)
+        assert len(out["global_crops"]) == 2
+        assert all(c.shape == (3, 64, 64) for c in out["global_crops"])
+        assert len(out["local_crops"]) == 3
+        assert all(c.shape == (3, 32, 32) for c in out["local_crops"])
+        assert all(torch.isfinite(c).all() for c in out["global_crops"] + out["local_crops"])
+        if "gram_teacher_crops_size" in kwargs:
+            assert len(out["gram_teacher_crops"]) == 2
+            assert out["gram_teacher_crops"][0].shape == (3, 64, 64)
