@@ -53,7 +53,12 @@ class FusedAdamW:
         lr_mult, wd_mult, is_last, sub_id = [], [], [], []
         for g in self.groups:
             for i, p in enumerate(g["params"]):
-                assert p.grad is not None, "plan build requires all grads present"
+                if p.grad is None:
+                    # params a config never trains (e.g. the ibot heads with
+                    # ibot.loss_weight=0) get a persistent zero grad so the
+                    # single-launch plan holds a pointer for them; autograd
+                    # accumulates into it if they ever receive gradients
+                    p.grad = torch.zeros_like(p)
                 p_list.append(p)
                 g_list.append(p.grad)
                 m_list.append(g["exp_avg"][i])
